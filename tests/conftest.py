@@ -1,0 +1,19 @@
+import pytest
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: requires an AMD GPU (MI355X) to run")
+
+
+def pytest_collection_modifyitems(config, items):
+    try:
+        import torch
+        has_gpu = torch.cuda.is_available()
+    except Exception:
+        has_gpu = False
+    if has_gpu:
+        return
+    skip = pytest.mark.skip(reason="no GPU available")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
