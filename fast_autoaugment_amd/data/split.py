@@ -1,0 +1,40 @@
+"""Stratified splits + CV folds (reference data.py:117-203).
+
+Uses sklearn StratifiedShuffleSplit with random_state=0 like the reference
+so reduced_* membership and the 5 CV folds match the published pipeline
+when run on the real datasets.
+"""
+from __future__ import annotations
+
+from typing import List, Tuple
+
+import numpy as np
+from sklearn.model_selection import StratifiedShuffleSplit
+
+
+def stratified_split(labels: np.ndarray, test_size: int | float,
+                     random_state: int = 0) -> Tuple[np.ndarray, np.ndarray]:
+    """One stratified (train_idx, rest_idx) split."""
+    sss = StratifiedShuffleSplit(n_splits=1, test_size=test_size, random_state=random_state)
+    train_idx, rest_idx = next(sss.split(np.zeros(len(labels)), labels))
+    return train_idx, rest_idx
+
+
+def cv_split(labels: np.ndarray, split: float, split_idx: int,
+             random_state: int = 0) -> Tuple[np.ndarray, np.ndarray]:
+    """The reference's K-fold scheme (data.py:192-203): 5 stratified shuffle
+    splits with test_size=split; fold k uses the k-th draw."""
+    sss = StratifiedShuffleSplit(n_splits=5, test_size=split, random_state=random_state)
+    it = sss.split(np.zeros(len(labels)), labels)
+    train_idx, valid_idx = None, None
+    for _ in range(split_idx + 1):
+        train_idx, valid_idx = next(it)
+    return train_idx, valid_idx
+
+
+def reduce_dataset(labels: np.ndarray, keep: int, random_state: int = 0) -> np.ndarray:
+    """reduced_cifar10 (4k) / reduced_svhn (1k) index selection
+    (reference data.py:117-144): stratified keep-subset."""
+    train_idx, _ = stratified_split(labels, test_size=len(labels) - keep,
+                                    random_state=random_state)
+    return train_idx

@@ -1,0 +1,22 @@
+"""Small fused building blocks shared by the model zoo.
+
+``bn_relu(x, bn)`` is the pre-activation hot path of WideResNet/PyramidNet:
+on GPU it dispatches to the fused NHWC BatchNorm+ReLU HIP kernel
+(fwd + bwd as one custom Function); on CPU it composes torch ops. The BN
+module keeps standard nn.BatchNorm2d parameters/buffers so state_dicts stay
+checkpoint-compatible with the reference layout.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from . import has_ext
+
+
+def bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
+    """BatchNorm2d followed by ReLU, fused on GPU."""
+    if x.is_cuda and has_ext():
+        from .bnrelu import fused_bn_relu
+        return fused_bn_relu(x, bn)
+    return F.relu(bn(x))
