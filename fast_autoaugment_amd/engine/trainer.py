@@ -80,6 +80,8 @@ def run_epoch(model, loader, loss_fn, optimizer, desc_default="", epoch=0,
 
         if optimizer:
             loss.backward()
+            if hasattr(model, "finish_gradient_sync"):
+                model.finish_gradient_sync()
             if wd > 0.0:
                 _apply_manual_wd(decay_params, wd)
             grad_clip = conf["optimizer"].get("clip", 5.0)
