@@ -34,8 +34,9 @@ logger = get_logger("faa_amd.trainer")
 
 def _decay_params(model: nn.Module):
     """Non-BN parameters for the manual weight-decay term (train.py:40)."""
-    return [p for name, p in model.named_parameters()
-            if not ("_bn" in name or ".bn" in name or name.startswith("bn"))]
+    from ..parallel.flat import bn_param_names
+    bn = bn_param_names(model)
+    return [p for name, p in model.named_parameters() if name not in bn]
 
 
 def _apply_manual_wd(params, wd: float):

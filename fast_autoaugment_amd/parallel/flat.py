@@ -18,11 +18,17 @@ import torch
 from torch import nn
 
 
-def _is_no_decay(name: str) -> bool:
-    """BN params are excluded from the manual WD (reference train.py:40:
-    names containing '_bn' or '.bn'; our models also use bare 'bn'/'_bn0'
-    prefixes and BN modules named bn*)."""
-    return "_bn" in name or ".bn" in name or name.startswith("bn")
+def bn_param_names(model: nn.Module) -> set:
+    """Parameters belonging to normalization modules — excluded from the
+    manual WD (the reference's name heuristic '_bn'/'.bn' at train.py:40
+    resolves to exactly these on its model zoo; module-type detection is
+    equivalent there and robust to anonymous module names)."""
+    names = set()
+    for mod_name, mod in model.named_modules():
+        if isinstance(mod, (nn.modules.batchnorm._BatchNorm, nn.GroupNorm, nn.LayerNorm)):
+            for pn, _ in mod.named_parameters(recurse=False):
+                names.add(f"{mod_name}.{pn}" if mod_name else pn)
+    return names
 
 
 class FlatParams:
@@ -45,8 +51,9 @@ def flatten_module(model: nn.Module, align: int = 64) -> FlatParams:
     vector-load friendly in the fused kernels.
     """
     named = list(model.named_parameters())
-    decay = [(n, p) for n, p in named if not _is_no_decay(n)]
-    nodecay = [(n, p) for n, p in named if _is_no_decay(n)]
+    bn_names = bn_param_names(model)
+    decay = [(n, p) for n, p in named if n not in bn_names]
+    nodecay = [(n, p) for n, p in named if n in bn_names]
     ordered = decay + nodecay
 
     def padded(n):

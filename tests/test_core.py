@@ -1,0 +1,154 @@
+"""CPU unit tests: policies codec, schedules, optimizers, EMA, metrics, flat params."""
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from fast_autoaugment_amd import policies
+from fast_autoaugment_amd.common import EMA
+from fast_autoaugment_amd.lr_scheduler import build_scheduler
+from fast_autoaugment_amd.metrics import CrossEntropyLabelSmooth, accuracy
+from fast_autoaugment_amd.optim import RMSpropTF
+from fast_autoaugment_amd.parallel.flat import flatten_module
+
+
+def test_policy_archives_load():
+    for name in ["fa_reduced_cifar10", "fa_reduced_svhn", "fa_resnet50_rimagenet"]:
+        pol = policies.get_archive(name)
+        assert len(pol) > 400
+        for sub in pol:
+            for (op, pr, lvl) in sub:
+                assert op in policies.ALL_OPS
+                assert 0.0 <= pr <= 1.0
+
+
+def test_policy_codec_roundtrip():
+    pol = policies.get_archive("fa_reduced_cifar10")[:5]
+    cfg = policies.policy_encoder(pol)
+    back = policies.policy_decoder(cfg, 5, 2)
+    assert [[tuple(op) for op in sub] for sub in back] == \
+           [[tuple(op) for op in sub] for sub in pol]
+
+
+def test_remove_duplicates():
+    p = [[("Rotate", 0.5, 0.5), ("Color", 0.1, 0.2)],
+         [("Rotate", 0.9, 0.1), ("Color", 0.3, 0.4)],
+         [("Color", 0.1, 0.2), ("Rotate", 0.5, 0.5)]]
+    out = policies.remove_duplicates(p)
+    assert len(out) == 2
+
+
+def _sched_conf(stype, epochs, warm=5):
+    return {"epoch": epochs,
+            "lr_schedule": {"type": stype, "warmup": {"multiplier": 1, "epoch": warm}}}
+
+
+def test_cosine_warmup_schedule():
+    opt = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=0.1)
+    s = build_scheduler(_sched_conf("cosine", 200), opt, 0.1)
+    s.step(0.0)
+    assert opt.param_groups[0]["lr"] == 0.0          # warmup start
+    s.step(5.0)
+    assert abs(opt.param_groups[0]["lr"] - 0.1) < 1e-9
+    s.step(205.0)
+    assert opt.param_groups[0]["lr"] < 1e-4          # cosine tail ~0
+
+
+def test_resnet_schedule():
+    opt = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=0.1)
+    s = build_scheduler(_sched_conf("resnet", 270), opt, 0.4)
+    s.step(50.0)
+    assert abs(opt.param_groups[0]["lr"] - 0.4) < 1e-9
+    s.step(100.0)
+    assert abs(opt.param_groups[0]["lr"] - 0.04) < 1e-9
+    s.step(250.0)
+    assert abs(opt.param_groups[0]["lr"] - 0.0004) < 1e-9
+
+
+def test_rmsprop_tf_semantics():
+    """TF semantics: ms init ones, eps inside sqrt (reference rmsprop.py:80-97)."""
+    p = torch.nn.Parameter(torch.tensor([1.0]))
+    opt = RMSpropTF([p], lr=0.1, alpha=0.9, momentum=0.9, eps=0.001)
+    p.grad = torch.tensor([0.5])
+    opt.step()
+    # ms = 1 + (0.25-1)*0.1 = 0.925 ; mom = 0.1*0.5/sqrt(0.925+0.001)
+    expect_mom = 0.1 * 0.5 / math.sqrt(0.925 + 0.001)
+    assert abs(p.item() - (1.0 - expect_mom)) < 1e-6
+
+
+def test_ema_warmup_and_lerp():
+    m = torch.nn.Linear(4, 4)
+    ema = EMA(0.999)
+    ema(m, step=0)       # mu = min(.999, 1/10) -> initial copy
+    w0 = ema.shadow["weight"].clone()
+    with torch.no_grad():
+        m.weight.add_(1.0)
+    ema(m, step=1)       # mu = 2/11
+    mu = 2.0 / 11
+    expect = (1 - mu) * m.weight + mu * w0
+    assert torch.allclose(ema.shadow["weight"], expect, atol=1e-6)
+
+
+def test_label_smooth_ce_matches_manual():
+    torch.manual_seed(0)
+    logits = torch.randn(8, 10)
+    target = torch.randint(0, 10, (8,))
+    crit = CrossEntropyLabelSmooth(10, 0.1)
+    loss = crit(logits, target)
+    # manual
+    logp = torch.log_softmax(logits, 1)
+    t = torch.full_like(logp, 0.1 / 10)
+    t.scatter_(1, target[:, None], 1 - 0.1 + 0.1 / 10)
+    manual = -(t * logp).sum(1).mean()
+    assert torch.allclose(loss, manual, atol=1e-6)
+    # epsilon=0 equals plain CE
+    crit0 = CrossEntropyLabelSmooth(10, 0.0)
+    assert torch.allclose(crit0(logits, target),
+                          torch.nn.functional.cross_entropy(logits, target), atol=1e-6)
+
+
+def test_accuracy_topk():
+    logits = torch.tensor([[0.9, 0.1, 0.0], [0.1, 0.8, 0.1], [0.5, 0.4, 0.1]])
+    target = torch.tensor([0, 1, 2])
+    top1, top2 = accuracy(logits, target, (1, 2))
+    assert abs(top1.item() - 2 / 3) < 1e-6
+    assert abs(top2.item() - 2 / 3) < 1e-6
+
+
+def test_flatten_module_views_and_decay_order():
+    m = torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3), torch.nn.BatchNorm2d(8),
+                            torch.nn.Conv2d(8, 4, 1))
+    flat = flatten_module(m)
+    # all params are views of the flat buffer
+    for p in m.parameters():
+        assert p.data_ptr() >= flat.flat_param.data_ptr()
+        assert p.grad is not None
+    # BN params must sit in the no-decay tail
+    bn_w = dict(m.named_parameters())["1.weight"]
+    off = (bn_w.data_ptr() - flat.flat_param.data_ptr()) // 4
+    assert off >= flat.n_decay
+    # training still works through the views
+    x = torch.randn(2, 3, 8, 8)
+    y = m(x).sum()
+    y.backward()
+    assert flat.flat_grad.abs().sum() > 0
+
+
+def test_flat_optimizer_equivalence():
+    """A step through flat views must equal a step on a regular clone."""
+    torch.manual_seed(0)
+    m1 = torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 2))
+    m2 = torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 2))
+    m2.load_state_dict(m1.state_dict())
+    flatten_module(m1)
+    o1 = torch.optim.SGD(m1.parameters(), lr=0.1, momentum=0.9, nesterov=True)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.1, momentum=0.9, nesterov=True)
+    x = torch.randn(4, 8)
+    for _ in range(3):
+        for m, o in [(m1, o1), (m2, o2)]:
+            o.zero_grad(set_to_none=False)
+            m(x).sum().backward()
+            o.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-6)
