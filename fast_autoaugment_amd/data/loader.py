@@ -57,7 +57,7 @@ class AugLoader:
                  indices: Optional[np.ndarray] = None,
                  shuffle: Optional[bool] = None, drop_last: Optional[bool] = None,
                  rank: int = 0, world_size: int = 1, seed: int = 0,
-                 out_dtype: torch.dtype = torch.float32):
+                 out_dtype: torch.dtype = torch.float32, prefetch: int = 2):
         self.store = store
         self.batch = batch
         self.policy = policy or []
@@ -74,6 +74,7 @@ class AugLoader:
         self.seed = seed
         self.epoch = 0
         self.out_dtype = out_dtype
+        self.prefetch = prefetch
         self._mean_t = None
         self._std_t = None
 
@@ -98,21 +99,45 @@ class AugLoader:
         n = len(self._epoch_indices())
         return n // self.batch if self.drop_last else int(np.ceil(n / self.batch))
 
+    def _gen_host(self, idx: np.ndarray, rng: np.random.Generator, b: int):
+        sel = idx[b * self.batch:(b + 1) * self.batch]
+        H, W = self.store.hw
+        if self.train:
+            prog = aug_ops.compile_program(self.policy, len(sel), W, H, rng)
+            post = aug_ops.compile_post(len(sel), W, H, rng, pad=self.pad,
+                                        cutout_len=self.cutout, train=True)
+        else:
+            prog = np.zeros((len(sel), aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH), np.float32)
+            post = np.zeros((len(sel), 6), np.float32)
+        return sel, prog, post
+
     def __iter__(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
         idx = self._epoch_indices()
         nb = len(self)
         rng = np.random.default_rng((self.seed * 7919 + self.epoch * 13 + self.rank) & 0x7FFFFFFF)
-        H, W = self.store.hw
-        for b in range(nb):
-            sel = idx[b * self.batch:(b + 1) * self.batch]
-            if self.train:
-                prog = aug_ops.compile_program(self.policy, len(sel), W, H, rng)
-                post = aug_ops.compile_post(len(sel), W, H, rng, pad=self.pad,
-                                            cutout_len=self.cutout, train=True)
-            else:
-                prog = np.zeros((len(sel), aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH), np.float32)
-                post = np.zeros((len(sel), 6), np.float32)
-            yield self._make_batch(sel, prog, post)
+        if self.prefetch > 0 and nb > 1:
+            # overlap host-side RNG/program compilation with GPU compute
+            # (the reference's analog: 8 DataLoader worker processes)
+            import queue as _q
+            import threading
+            q: "_q.Queue" = _q.Queue(maxsize=self.prefetch)
+
+            def producer():
+                for b in range(nb):
+                    q.put(self._gen_host(idx, rng, b))
+                q.put(None)
+
+            t = threading.Thread(target=producer, daemon=True)
+            t.start()
+            while True:
+                item = q.get()
+                if item is None:
+                    break
+                yield self._make_batch(*item)
+            t.join()
+        else:
+            for b in range(nb):
+                yield self._make_batch(*self._gen_host(idx, rng, b))
 
     def _make_batch(self, sel: np.ndarray, prog: np.ndarray, post: np.ndarray):
         if self.store.device.type == "cuda":

@@ -1,0 +1,45 @@
+// Python bindings for the fast_autoaugment_amd CDNA4 kernel library.
+#include <torch/extension.h>
+
+torch::Tensor scale_bcast(torch::Tensor x, torch::Tensor s);
+torch::Tensor scale_lerp(torch::Tensor x1, torch::Tensor x2, torch::Tensor a);
+torch::Tensor swish_fwd(torch::Tensor x);
+torch::Tensor swish_bwd(torch::Tensor g, torch::Tensor x);
+torch::Tensor mixup_fwd(torch::Tensor x, torch::Tensor perm, float lam);
+torch::Tensor pad_add(torch::Tensor x, torch::Tensor shortcut);
+std::vector<torch::Tensor> label_smooth_ce_fwd(torch::Tensor logits, torch::Tensor target, double eps);
+torch::Tensor label_smooth_ce_bwd(torch::Tensor softmax, torch::Tensor target,
+                                  torch::Tensor grad_loss, double eps);
+void sgd_fused_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
+                    torch::Tensor normsq, int64_t n_decay, double wd, double clip,
+                    double lr, double momentum, int64_t nesterov);
+void ema_lerp_(torch::Tensor shadow, torch::Tensor x, double mu);
+torch::Tensor aug_pipeline(torch::Tensor images, torch::Tensor sel, torch::Tensor prog,
+                           torch::Tensor post, torch::Tensor mean, torch::Tensor std,
+                           bool bf16_out);
+std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
+                                       torch::Tensor beta, torch::Tensor running_mean,
+                                       torch::Tensor running_var, bool training,
+                                       double momentum, double eps);
+std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor out, torch::Tensor mean,
+                                       torch::Tensor invstd, torch::Tensor gamma,
+                                       bool training);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("scale_bcast", &scale_bcast, "out = x * s[b] (per-sample broadcast)");
+  m.def("scale_lerp", &scale_lerp, "out = a[b]*x1 + (1-a[b])*x2");
+  m.def("swish_fwd", &swish_fwd);
+  m.def("swish_bwd", &swish_bwd);
+  m.def("mixup_fwd", &mixup_fwd);
+  m.def("pad_add", &pad_add, "out = x + zero-channel-padded shortcut (NHWC)");
+  m.def("label_smooth_ce_fwd", &label_smooth_ce_fwd);
+  m.def("label_smooth_ce_bwd", &label_smooth_ce_bwd);
+  m.def("sgd_fused_step", &sgd_fused_step,
+        "fused manual-WD + global-clip + nesterov SGD on flat buffers");
+  m.def("ema_lerp_", &ema_lerp_);
+  m.def("aug_pipeline", &aug_pipeline,
+        "batched augmentation program executor (uint8 NHWC -> normalized bf16/f32)");
+  m.def("bn_relu_fwd", &bn_relu_fwd);
+  m.def("bn_relu_bwd", &bn_relu_bwd);
+}

@@ -1,0 +1,143 @@
+// Fused flat-buffer optimizer step (reference update chain train.py:61-70):
+//   1. wd_norm_kernel: g[:n_decay] += wd * p[:n_decay]   (manual non-BN L2)
+//                      + block-partial ||g||^2 accumulated to normsq[0]
+//   2. sgd_step_kernel: coef = min(1, clip/(sqrt(normsq)+1e-6)) read ON
+//      DEVICE (no host sync -> hipGraph-capturable), nesterov momentum
+//      update, p -= lr*(g*coef + mu*buf')  [torch SGD semantics]
+//   3. ema_lerp_kernel: shadow = (1-mu)*x + mu*shadow on a flat buffer
+//
+// All buffers are fp32 and contiguous (parallel/flat.py), so both kernels
+// stream at HBM bandwidth with float4 vector access.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "faa_common.h"
+
+namespace {
+
+__global__ void zero1_kernel(float* p) { if (threadIdx.x == 0) p[0] = 0.0f; }
+
+__global__ void wd_norm_kernel(float* __restrict__ g, const float* __restrict__ p,
+                               float* __restrict__ normsq, int64_t n, int64_t n_decay,
+                               float wd) {
+  __shared__ float lds[8];
+  int64_t i0 = (int64_t)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  float acc = 0.0f;
+  for (int64_t i = i0; i < n; i += stride) {
+    if (i + 3 < n) {
+      float4 gv = *reinterpret_cast<float4*>(g + i);
+      if (wd != 0.0f && i + 3 < n_decay) {
+        const float4 pv = *reinterpret_cast<const float4*>(p + i);
+        gv.x += wd * pv.x; gv.y += wd * pv.y; gv.z += wd * pv.z; gv.w += wd * pv.w;
+        *reinterpret_cast<float4*>(g + i) = gv;
+      } else if (wd != 0.0f && i < n_decay) {
+        // straddles the decay boundary: scalar tail
+        for (int v = 0; v < 4; ++v)
+          if (i + v < n_decay) g[i + v] += wd * p[i + v];
+        gv = *reinterpret_cast<float4*>(g + i);
+      }
+      acc += gv.x * gv.x + gv.y * gv.y + gv.z * gv.z + gv.w * gv.w;
+    } else {
+      for (int64_t j = i; j < n; ++j) {
+        if (wd != 0.0f && j < n_decay) g[j] += wd * p[j];
+        float gv = g[j];
+        acc += gv * gv;
+      }
+    }
+  }
+  acc = faa_block_reduce_sum(acc, lds);
+  if (threadIdx.x == 0) atomicAdd(normsq, acc);
+}
+
+__global__ void sgd_step_kernel(float* __restrict__ p, float* __restrict__ g,
+                                float* __restrict__ buf, const float* __restrict__ normsq,
+                                int64_t n, float clip, float lr, float mu, int nesterov) {
+  float coef = 1.0f;
+  if (clip > 0.0f) {
+    float norm = sqrtf(normsq[0]);
+    coef = fminf(1.0f, clip / (norm + 1e-6f));
+  }
+  int64_t i0 = (int64_t)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i = i0; i < n; i += stride) {
+    if (i + 3 < n) {
+      float4 gv = *reinterpret_cast<float4*>(g + i);
+      float4 bv = *reinterpret_cast<float4*>(buf + i);
+      float4 pv = *reinterpret_cast<float4*>(p + i);
+      #pragma unroll
+      for (int v = 0; v < 4; ++v) {
+        float* gp = &gv.x + v; float* bp = &bv.x + v; float* pp = &pv.x + v;
+        float gg = *gp * coef;
+        float bb = mu * *bp + gg;
+        float upd = nesterov ? (gg + mu * bb) : bb;
+        *bp = bb;
+        *pp = *pp - lr * upd;
+      }
+      *reinterpret_cast<float4*>(buf + i) = bv;
+      *reinterpret_cast<float4*>(p + i) = pv;
+      // also store clipped grad so later consumers (clip-aware logs) see it
+      gv.x *= coef; gv.y *= coef; gv.z *= coef; gv.w *= coef;
+      *reinterpret_cast<float4*>(g + i) = gv;
+    } else {
+      for (int64_t j = i; j < n; ++j) {
+        float gg = g[j] * coef;
+        float bb = mu * buf[j] + gg;
+        float upd = nesterov ? (gg + mu * bb) : bb;
+        buf[j] = bb;
+        p[j] -= lr * upd;
+        g[j] = gg;
+      }
+    }
+  }
+}
+
+__global__ void ema_lerp_kernel(float* __restrict__ shadow, const float* __restrict__ x,
+                                int64_t n, float mu) {
+  int64_t i0 = (int64_t)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i = i0; i < n; i += stride) {
+    if (i + 3 < n) {
+      float4 sv = *reinterpret_cast<float4*>(shadow + i);
+      const float4 xv = *reinterpret_cast<const float4*>(x + i);
+      sv.x = (1.0f - mu) * xv.x + mu * sv.x;
+      sv.y = (1.0f - mu) * xv.y + mu * sv.y;
+      sv.z = (1.0f - mu) * xv.z + mu * sv.z;
+      sv.w = (1.0f - mu) * xv.w + mu * sv.w;
+      *reinterpret_cast<float4*>(shadow + i) = sv;
+    } else {
+      for (int64_t j = i; j < n; ++j)
+        shadow[j] = (1.0f - mu) * x[j] + mu * shadow[j];
+    }
+  }
+}
+
+}  // namespace
+
+void sgd_fused_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
+                    torch::Tensor normsq, int64_t n_decay, double wd, double clip,
+                    double lr, double momentum, int64_t nesterov) {
+  TORCH_CHECK(p.is_cuda() && p.dtype() == torch::kFloat32 && p.is_contiguous());
+  int64_t n = p.numel();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const int block = 256;
+  int grid = faa_grid(n / 4 + 1, block);
+  hipLaunchKernelGGL(zero1_kernel, dim3(1), dim3(64), 0, stream, normsq.data_ptr<float>());
+  hipLaunchKernelGGL(wd_norm_kernel, dim3(grid), dim3(block), 0, stream,
+                     g.data_ptr<float>(), p.data_ptr<float>(), normsq.data_ptr<float>(),
+                     n, n_decay, (float)wd);
+  hipLaunchKernelGGL(sgd_step_kernel, dim3(grid), dim3(block), 0, stream,
+                     p.data_ptr<float>(), g.data_ptr<float>(), buf.data_ptr<float>(),
+                     normsq.data_ptr<float>(), n, (float)clip, (float)lr,
+                     (float)momentum, (int)nesterov);
+}
+
+void ema_lerp_(torch::Tensor shadow, torch::Tensor x, double mu) {
+  TORCH_CHECK(shadow.is_cuda() && shadow.is_contiguous() && x.is_contiguous());
+  int64_t n = shadow.numel();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const int block = 256;
+  int grid = faa_grid(n / 4 + 1, block);
+  hipLaunchKernelGGL(ema_lerp_kernel, dim3(grid), dim3(block), 0, stream,
+                     shadow.data_ptr<float>(), x.data_ptr<float>(), n, (float)mu);
+}
