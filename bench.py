@@ -31,7 +31,8 @@ def parse_args():
     p.add_argument("--model", type=str, default="wresnet40_2")
     p.add_argument("--dataset", type=str, default="cifar10")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
-    p.add_argument("--graphs", action="store_true", help="capture the step in a hipGraph")
+    p.add_argument("--graphs", type=int, default=1,
+                   help="1: capture the train step in a hipGraph (default), 0: eager")
     return p.parse_args()
 
 
@@ -94,34 +95,99 @@ def main():
     amp_dtype = torch.bfloat16 if args.dtype == "bf16" else None
 
     model.train()
-    it = iter(loader)
-    steps_per_epoch = len(loader)
-
-    def next_batch():
-        nonlocal it
-        try:
-            return next(it)
-        except StopIteration:
-            loader.set_epoch(loader.epoch + 1)
-            it = iter(loader)
-            return next(it)
-
+    steps_per_epoch = max(len(loader), 1)
     step_idx = 0
 
-    def one_step():
-        nonlocal step_idx
-        data, label = next_batch()
+    use_graphs = bool(args.graphs) and not distributed
+    from fast_autoaugment_amd.aug import ops as aug_ops
+    from fast_autoaugment_amd.ops import ext
+    CX = ext()
+
+    # host-side program generation, prefetched on a thread (the analog of the
+    # reference's 8 DataLoader workers): the GPU never waits for host RNG
+    import queue as _q
+    import threading
+    rng = np.random.default_rng(1000 + rank)
+    policy = resolve_aug("fa_reduced_cifar10")
+    n_imgs = len(store)
+
+    def gen_host():
+        sel = rng.integers(0, n_imgs, size=args.batch)
+        prog = aug_ops.compile_program(policy, args.batch, 32, 32, rng)
+        post = aug_ops.compile_post(args.batch, 32, 32, rng, pad=4,
+                                    cutout_len=16, train=True)
+        return sel, prog, post
+
+    host_q: "_q.Queue" = _q.Queue(maxsize=6)
+
+    def producer():
+        while True:
+            host_q.put(gen_host())
+
+    threading.Thread(target=producer, daemon=True).start()
+
+    mean_t = torch.from_numpy(mean).to(dev)
+    std_t = torch.from_numpy(std).to(dev)
+    bf16 = out_dtype == torch.bfloat16
+
+    # static device-side inputs (updated in place each step; graph-replayable)
+    sel_s = torch.zeros(args.batch, dtype=torch.int64, device=dev)
+    prog_s = torch.zeros((args.batch, aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH),
+                         dtype=torch.float32, device=dev)
+    post_s = torch.zeros((args.batch, 6), dtype=torch.float32, device=dev)
+    sel_h = torch.zeros_like(sel_s, device="cpu", pin_memory=True)
+    prog_h = torch.zeros_like(prog_s, device="cpu", pin_memory=True)
+    post_h = torch.zeros_like(post_s, device="cpu", pin_memory=True)
+
+    def upload_next():
+        sel, prog, post = host_q.get()
+        sel_h.copy_(torch.from_numpy(sel))
+        prog_h.copy_(torch.from_numpy(prog))
+        post_h.copy_(torch.from_numpy(post))
+        sel_s.copy_(sel_h, non_blocking=True)
+        prog_s.copy_(prog_h, non_blocking=True)
+        post_s.copy_(post_h, non_blocking=True)
+
+    def step_body():
+        opt.zero_grad()
+        data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s, mean_t, std_t, bf16)
+        label = store.labels.index_select(0, sel_s)
         with torch.autocast("cuda", dtype=amp_dtype, enabled=amp_dtype is not None):
             preds = model(data)
             loss = crit(preds, label)
         loss.backward()
         if distributed:
             model.finish_gradient_sync()
-        sched.step(step_idx / steps_per_epoch)
         opt.step()
-        opt.zero_grad()
-        step_idx += 1
         return loss
+
+    graph = None
+    if use_graphs:
+        opt.sync_lr()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                upload_next()
+                step_body()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        upload_next()
+        with torch.cuda.graph(graph):
+            step_body()
+
+    def one_step():
+        nonlocal step_idx
+        sched.step(step_idx / steps_per_epoch)
+        if graph is not None:
+            opt.sync_lr()
+            upload_next()
+            graph.replay()
+        else:
+            upload_next()
+            step_body()
+        step_idx += 1
 
     # ---- warmup ----
     for _ in range(args.warmup):

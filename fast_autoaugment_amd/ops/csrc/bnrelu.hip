@@ -2,14 +2,17 @@
 //
 // The pre-activation BN->ReLU pair is the second-hottest op family of the
 // CIFAR nets after conv (reference wideresnet.py:37-38, pyramidnet.py:83-96).
-// MIOpen runs BN and ReLU as separate kernels over NCHW; here both fuse into
-// one NHWC pass with fp32 stats: reduce (sum/sumsq per channel, wave-aligned
-// 64-channel strips) -> finalize (mean/invstd + running-stat update)
-// -> apply (normalize+scale+relu). Backward reduces dgamma/dbeta with the
-// relu mask folded in, then applies dx in one pass.
+// MIOpen runs BN and ReLU as separate NCHW kernels; here both fuse into one
+// NHWC pass with fp32 stats.
 //
-// Layout: x is [N,C,H,W] channels_last => storage [rows=N*H*W][C]; lanes map
-// to consecutive channels (coalesced), rows stride across the grid.
+// Reduction scheme (HBM-bound, per G13): the NHWC storage [rows][C] is read
+// LINEARLY in 8-element bf16 vectors. With C a multiple of 8 and the thread
+// stride (blockDim*8) a multiple of C, every thread touches a FIXED set of 8
+// channels -> 2x8 fp32 accumulators live in registers, no LDS/global atomics
+// in the main loop. Per-block partials land in scratch[block][2C] (no
+// zero-fill needed); a small finalize kernel folds partials + running-stat
+// update. C not divisible by 8 (e.g. stem C=3 is never BN'd here) falls back
+// to a strided scalar path.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -17,43 +20,107 @@
 
 namespace {
 
+typedef __attribute__((ext_vector_type(4))) short short4v;
+
+__device__ __forceinline__ float bf16_bits_to_f(short u) {
+  union { float f; uint32_t i; } c;
+  c.i = ((uint32_t)(uint16_t)u) << 16;
+  return c.f;
+}
+
+// ---------------------------------------------------------------- fwd reduce
+// block: 256 threads; each thread owns channels [c0, c0+8) with
+// c0 = (tid*8) % C. partials: scratch[blockIdx.x*2C + {c, C+c}]
 template <typename T>
-__global__ void bn_reduce_kernel(const T* __restrict__ x, float* __restrict__ scratch,
-                                 int64_t rows, int C) {
-  // block: 64 x 4 (channel x row-group); grid: (row_chunks, ceil(C/64))
-  int c = blockIdx.y * 64 + (threadIdx.x & 63);
-  int rg = threadIdx.x >> 6;      // 0..3
-  float s = 0.0f, ss = 0.0f;
-  if (c < C) {
-    for (int64_t r = (int64_t)blockIdx.x * 4 + rg; r < rows; r += (int64_t)gridDim.x * 4) {
-      float v = faa_to_float(x[r * C + c]);
-      s += v; ss += v * v;
+__global__ void bn_reduce_vec_kernel(const T* __restrict__ x, float* __restrict__ scratch,
+                                     int64_t total, int C) {
+  // host guarantees (gridDim*blockDim*8) % C == 0, so each thread's channel
+  // octet c0 is FIXED across its grid-stride loop.
+  float s[8] = {0}, ss[8] = {0};
+  const int c0 = (int)((((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8) % C);
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  if constexpr (sizeof(T) == 2) {
+    for (int64_t i = i0; i < total; i += stride) {
+      short4v v0 = *reinterpret_cast<const short4v*>(reinterpret_cast<const short*>(x) + i);
+      short4v v1 = *reinterpret_cast<const short4v*>(reinterpret_cast<const short*>(x) + i + 4);
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float f0 = bf16_bits_to_f(v0[k]);
+        float f1 = bf16_bits_to_f(v1[k]);
+        s[k] += f0; ss[k] += f0 * f0;
+        s[k + 4] += f1; ss[k + 4] += f1 * f1;
+      }
+    }
+  } else {
+    for (int64_t i = i0; i < total; i += stride) {
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float f = faa_to_float(x[i + k]);
+        s[k] += f; ss[k] += f * f;
+      }
     }
   }
-  __shared__ float lds_s[4][64];
-  __shared__ float lds_ss[4][64];
-  lds_s[rg][threadIdx.x & 63] = s;
-  lds_ss[rg][threadIdx.x & 63] = ss;
+  // fold threads with the same channel octet through LDS (no atomics).
+  // thread t owns octet ((bid*blockDim + t) % groups); for channel c the
+  // owner threads are t = ((c/8 - bid*blockDim) mod groups) + k*groups.
+  __shared__ float lds[256 * 8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = s[k];
   __syncthreads();
-  if (rg == 0 && c < C) {
-    float ts = 0, tss = 0;
-    #pragma unroll
-    for (int k = 0; k < 4; ++k) { ts += lds_s[k][threadIdx.x & 63]; tss += lds_ss[k][threadIdx.x & 63]; }
-    atomicAdd(&scratch[c], ts);
-    atomicAdd(&scratch[C + c], tss);
+  float* out = scratch + (int64_t)blockIdx.x * 2 * C;
+  const int groups = C / 8;
+  const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % groups);
+  if ((int)threadIdx.x < C) {
+    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups)
+      acc += lds[t * 8 + lane];
+    out[threadIdx.x] = acc;
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = ss[k];
+  __syncthreads();
+  if ((int)threadIdx.x < C) {
+    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups)
+      acc += lds[t * 8 + lane];
+    out[C + threadIdx.x] = acc;
   }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ scratch,
+// scalar fallback for C % 8 != 0 (rare; uses global atomics on zeroed scratch)
+template <typename T>
+__global__ void bn_reduce_scalar_kernel(const T* __restrict__ x, float* __restrict__ scratch,
+                                        int64_t total, int C) {
+  int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = i0; i < total; i += stride) {
+    float f = faa_to_float(x[i]);
+    int c = (int)(i % C);
+    atomicAdd(&scratch[c], f);
+    atomicAdd(&scratch[C + c], f * f);
+  }
+}
+
+__global__ void bn_finalize_kernel(const float* __restrict__ scratch, int nblocks,
                                    float* __restrict__ mean, float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
                                    int C, int64_t count, float eps, float momentum) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float m = scratch[c] / count;
-  float var = scratch[C + c] / count - m * m;
-  var = fmaxf(var, 0.0f);
+  float s = 0, ss = 0;
+  for (int b = 0; b < nblocks; ++b) {
+    s += scratch[(int64_t)b * 2 * C + c];
+    ss += scratch[(int64_t)b * 2 * C + C + c];
+  }
+  float m = s / count;
+  float var = fmaxf(ss / count - m * m, 0.0f);
   mean[c] = m;
   invstd[c] = rsqrtf(var + eps);
   if (running_mean != nullptr) {
@@ -63,6 +130,7 @@ __global__ void bn_finalize_kernel(const float* __restrict__ scratch,
   }
 }
 
+// ---------------------------------------------------------------- fwd apply
 template <typename T>
 __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
                                 const float* __restrict__ mean,
@@ -70,47 +138,121 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
                                 int64_t total, int C) {
-  int64_t i0 = blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  // precompute scale/shift per channel would need C regs; just read 4 floats
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (int64_t i = i0; i < total; i += stride) {
-    int c = (int)(i % C);
-    float v = (faa_to_float(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
-    faa_from_float(fmaxf(v, 0.0f), &out[i]);
+    int c0 = (int)(i % C);
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      int64_t j = i + k;
+      if (j < total) {
+        int c = c0 + k; while (c >= C) c -= C;
+        float v = (faa_to_float(x[j]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+        faa_from_float(fmaxf(v, 0.0f), &out[j]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- bwd reduce
+template <typename T>
+__global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __restrict__ out,
+                                         const T* __restrict__ dy,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ invstd,
+                                         float* __restrict__ scratch,
+                                         int64_t total, int C) {
+  float sdy[8] = {0}, sdyx[8] = {0};
+  const int c0 = (int)((((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8) % C);
+  float m[8], is[8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) { int c = (c0 + k) % C; m[k] = mean[c]; is[k] = invstd[c]; }
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  if constexpr (sizeof(T) == 2) {
+    const short* xs = reinterpret_cast<const short*>(x);
+    const short* os = reinterpret_cast<const short*>(out);
+    const short* ds = reinterpret_cast<const short*>(dy);
+    for (int64_t i = i0; i < total; i += stride) {
+      short4v xv0 = *reinterpret_cast<const short4v*>(xs + i);
+      short4v xv1 = *reinterpret_cast<const short4v*>(xs + i + 4);
+      short4v ov0 = *reinterpret_cast<const short4v*>(os + i);
+      short4v ov1 = *reinterpret_cast<const short4v*>(os + i + 4);
+      short4v dv0 = *reinterpret_cast<const short4v*>(ds + i);
+      short4v dv1 = *reinterpret_cast<const short4v*>(ds + i + 4);
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float g0 = bf16_bits_to_f(ov0[k]) > 0.0f ? bf16_bits_to_f(dv0[k]) : 0.0f;
+        float g1 = bf16_bits_to_f(ov1[k]) > 0.0f ? bf16_bits_to_f(dv1[k]) : 0.0f;
+        sdy[k] += g0;
+        sdyx[k] += g0 * (bf16_bits_to_f(xv0[k]) - m[k]) * is[k];
+        sdy[k + 4] += g1;
+        sdyx[k + 4] += g1 * (bf16_bits_to_f(xv1[k]) - m[k + 4]) * is[k + 4];
+      }
+    }
+  } else {
+    for (int64_t i = i0; i < total; i += stride) {
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = faa_to_float(out[i + k]) > 0.0f ? faa_to_float(dy[i + k]) : 0.0f;
+        sdy[k] += g;
+        sdyx[k] += g * (faa_to_float(x[i + k]) - m[k]) * is[k];
+      }
+    }
+  }
+  __shared__ float lds[256 * 8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = sdy[k];
+  __syncthreads();
+  float* outp = scratch + (int64_t)blockIdx.x * 2 * C;
+  const int groups = C / 8;
+  const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % groups);
+  if ((int)threadIdx.x < C) {
+    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups) acc += lds[t * 8 + lane];
+    outp[threadIdx.x] = acc;
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = sdyx[k];
+  __syncthreads();
+  if ((int)threadIdx.x < C) {
+    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups) acc += lds[t * 8 + lane];
+    outp[C + threadIdx.x] = acc;
   }
 }
 
 template <typename T>
-__global__ void bn_bwd_reduce_kernel(const T* __restrict__ x, const T* __restrict__ out,
-                                     const T* __restrict__ dy,
-                                     const float* __restrict__ mean,
-                                     const float* __restrict__ invstd,
-                                     float* __restrict__ scratch,   // [2C]: sum_dy, sum_dy_xhat
-                                     int64_t rows, int C) {
-  int c = blockIdx.y * 64 + (threadIdx.x & 63);
-  int rg = threadIdx.x >> 6;
-  float sdy = 0.0f, sdyx = 0.0f;
-  if (c < C) {
-    float m = mean[c], is = invstd[c];
-    for (int64_t r = (int64_t)blockIdx.x * 4 + rg; r < rows; r += (int64_t)gridDim.x * 4) {
-      int64_t i = r * C + c;
-      float mask = faa_to_float(out[i]) > 0.0f ? 1.0f : 0.0f;
-      float g = faa_to_float(dy[i]) * mask;
-      sdy += g;
-      sdyx += g * (faa_to_float(x[i]) - m) * is;
-    }
+__global__ void bn_bwd_reduce_scalar_kernel(const T* __restrict__ x, const T* __restrict__ out,
+                                            const T* __restrict__ dy,
+                                            const float* __restrict__ mean,
+                                            const float* __restrict__ invstd,
+                                            float* __restrict__ scratch,
+                                            int64_t total, int C) {
+  int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = i0; i < total; i += stride) {
+    int c = (int)(i % C);
+    float g = faa_to_float(out[i]) > 0.0f ? faa_to_float(dy[i]) : 0.0f;
+    atomicAdd(&scratch[c], g);
+    atomicAdd(&scratch[C + c], g * (faa_to_float(x[i]) - mean[c]) * invstd[c]);
   }
-  __shared__ float lds_a[4][64];
-  __shared__ float lds_b[4][64];
-  lds_a[rg][threadIdx.x & 63] = sdy;
-  lds_b[rg][threadIdx.x & 63] = sdyx;
-  __syncthreads();
-  if (rg == 0 && c < C) {
-    float ta = 0, tb = 0;
-    #pragma unroll
-    for (int k = 0; k < 4; ++k) { ta += lds_a[k][threadIdx.x & 63]; tb += lds_b[k][threadIdx.x & 63]; }
-    atomicAdd(&scratch[c], ta);
-    atomicAdd(&scratch[C + c], tb);
-  }
+}
+
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ scratch, int nblocks,
+                                       float* __restrict__ sums, int C) {
+  // sums: [2C] = {sum_dy, sum_dy_xhat}
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= 2 * C) return;
+  float s = 0;
+  for (int b = 0; b < nblocks; ++b) s += scratch[(int64_t)b * 2 * C + c];
+  sums[c] = s;
 }
 
 template <typename T>
@@ -119,25 +261,31 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     const float* __restrict__ gamma,
-                                    const float* __restrict__ scratch,
+                                    const float* __restrict__ sums,
                                     int64_t total, int C, int64_t count, int training) {
-  int64_t i0 = blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   float inv_count = 1.0f / (float)count;
   for (int64_t i = i0; i < total; i += stride) {
-    int c = (int)(i % C);
-    float mask = faa_to_float(out[i]) > 0.0f ? 1.0f : 0.0f;
-    float g = faa_to_float(dy[i]) * mask;
-    float is = invstd[c];
-    float res;
-    if (training) {
-      float xhat = (faa_to_float(x[i]) - mean[c]) * is;
-      float t = g - scratch[c] * inv_count - xhat * scratch[C + c] * inv_count;
-      res = gamma[c] * is * t;
-    } else {
-      res = gamma[c] * is * g;
+    int c0 = (int)(i % C);
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      int64_t j = i + k;
+      if (j < total) {
+        int c = c0 + k; while (c >= C) c -= C;
+        float mask = faa_to_float(out[j]) > 0.0f ? 1.0f : 0.0f;
+        float g = faa_to_float(dy[j]) * mask;
+        float is = invstd[c];
+        float res;
+        if (training) {
+          float xhat = (faa_to_float(x[j]) - mean[c]) * is;
+          res = gamma[c] * is * (g - sums[c] * inv_count - xhat * sums[C + c] * inv_count);
+        } else {
+          res = gamma[c] * is * g;
+        }
+        faa_from_float(res, &dx[j]);
+      }
     }
-    faa_from_float(res, &dx[i]);
   }
 }
 
@@ -150,6 +298,14 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
     else { TORCH_CHECK(false, NAME ": unsupported dtype"); }                  \
   }()
 
+// partial-sum block count: the grid stride (nblocks*256*8) must be a
+// multiple of C so each thread's channel octet is loop-invariant.
+static int bn_nblocks(int C) {
+  auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
+  int q = C / gcd(C, 2048);
+  return ((64 + q - 1) / q) * q;
+}
+
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
@@ -157,7 +313,8 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   TORCH_CHECK(x.dim() == 4, "bn_relu: 4-D input expected");
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   int C = xc.size(1);
-  int64_t rows = xc.numel() / C;
+  int64_t total = xc.numel();
+  int64_t rows = total / C;
   auto out = torch::empty_like(xc);
   auto f32 = xc.options().dtype(torch::kFloat32);
   auto mean = torch::empty({C}, f32);
@@ -168,30 +325,43 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   auto bta = beta.to(torch::kFloat32).contiguous();
 
   if (training) {
-    auto scratch = torch::zeros({2 * C}, f32);
-    dim3 block(256);
-    int row_chunks = (int)std::min<int64_t>((rows + 3) / 4, 1024);
-    dim3 grid(row_chunks, (C + 63) / 64);
-    DISPATCH_FB(xc.scalar_type(), "bn_reduce", [&] {
-      hipLaunchKernelGGL((bn_reduce_kernel<scalar_t>), grid, block, 0, stream,
-                         (const scalar_t*)xc.data_ptr(), scratch.data_ptr<float>(), rows, C);
-    });
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
-                       scratch.data_ptr<float>(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(),
-                       running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
-                       running_var.defined() ? running_var.data_ptr<float>() : nullptr,
-                       C, rows, (float)eps, (float)momentum);
+    bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
+    if (vec) {
+      int nb = bn_nblocks(C);
+      auto scratch = torch::empty({nb, 2 * C}, f32);
+      DISPATCH_FB(xc.scalar_type(), "bn_reduce", [&] {
+        hipLaunchKernelGGL((bn_reduce_vec_kernel<scalar_t>), dim3(nb), dim3(256), 0,
+                           stream, (const scalar_t*)xc.data_ptr(),
+                           scratch.data_ptr<float>(), total, C);
+      });
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
+                         scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                         C, rows, (float)eps, (float)momentum);
+    } else {
+      auto scratch = torch::zeros({2 * C}, f32);
+      int grid = faa_grid(total, 256, 1024);
+      DISPATCH_FB(xc.scalar_type(), "bn_reduce_s", [&] {
+        hipLaunchKernelGGL((bn_reduce_scalar_kernel<scalar_t>), dim3(grid), dim3(256), 0,
+                           stream, (const scalar_t*)xc.data_ptr(),
+                           scratch.data_ptr<float>(), total, C);
+      });
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
+                         scratch.data_ptr<float>(), 1, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                         C, rows, (float)eps, (float)momentum);
+    }
   } else {
-    // eval: mean/invstd from running stats (computed on device, no sync)
     mean.copy_(running_mean);
     invstd.copy_(torch::rsqrt(running_var + eps));
   }
-  int64_t total = xc.numel();
-  int block1 = 256;
-  int grid1 = faa_grid(total, block1);
+  int grid1 = faa_grid(total / 8 + 1, 256);
   DISPATCH_FB(xc.scalar_type(), "bn_apply", [&] {
-    hipLaunchKernelGGL((bn_apply_kernel<scalar_t>), dim3(grid1), dim3(block1), 0, stream,
+    hipLaunchKernelGGL((bn_apply_kernel<scalar_t>), dim3(grid1), dim3(256), 0, stream,
                        (const scalar_t*)xc.data_ptr(), (scalar_t*)out.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        g.data_ptr<float>(), bta.data_ptr<float>(), total, C);
@@ -207,35 +377,47 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
   auto oc = out.contiguous(torch::MemoryFormat::ChannelsLast);
   auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
   int C = xc.size(1);
-  int64_t rows = xc.numel() / C;
   int64_t total = xc.numel();
+  int64_t rows = total / C;
   auto f32 = xc.options().dtype(torch::kFloat32);
-  auto scratch = torch::zeros({2 * C}, f32);
   auto dx = torch::empty_like(xc);
   auto g = gamma.to(torch::kFloat32).contiguous();
+  auto sums = torch::empty({2 * C}, f32);
   auto stream = at::hip::getCurrentHIPStream().stream();
 
-  dim3 block(256);
-  int row_chunks = (int)std::min<int64_t>((rows + 3) / 4, 1024);
-  dim3 grid(row_chunks, (C + 63) / 64);
-  DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce", [&] {
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t>), grid, block, 0, stream,
-                       (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
-                       (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), scratch.data_ptr<float>(), rows, C);
-  });
-  int block1 = 256;
-  int grid1 = faa_grid(total, block1);
+  bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
+  if (vec) {
+    int nb = bn_nblocks(C);
+    auto scratch = torch::empty({nb, 2 * C}, f32);
+    DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce", [&] {
+      hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<scalar_t>), dim3(nb), dim3(256), 0,
+                         stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
+                         (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
+    });
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((2 * C + 255) / 256), dim3(256), 0,
+                       stream, scratch.data_ptr<float>(), nb,
+                       sums.data_ptr<float>(), C);
+  } else {
+    sums.zero_();
+    int grid = faa_grid(total, 256, 1024);
+    DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce_s", [&] {
+      hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<scalar_t>), dim3(grid), dim3(256), 0,
+                         stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
+                         (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), sums.data_ptr<float>(), total, C);
+    });
+  }
+  int grid1 = faa_grid(total / 8 + 1, 256);
   DISPATCH_FB(xc.scalar_type(), "bn_bwd_apply", [&] {
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<scalar_t>), dim3(grid1), dim3(block1), 0, stream,
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<scalar_t>), dim3(grid1), dim3(256), 0, stream,
                        (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
                        (const scalar_t*)dyc.data_ptr(), (scalar_t*)dx.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       g.data_ptr<float>(), scratch.data_ptr<float>(),
+                       g.data_ptr<float>(), sums.data_ptr<float>(),
                        total, C, rows, training ? 1 : 0);
   });
-  // dgamma = scratch[C:2C], dbeta = scratch[0:C]
-  auto dbeta = scratch.narrow(0, 0, C).clone();
-  auto dgamma = scratch.narrow(0, C, C).clone();
+  auto dbeta = sums.narrow(0, 0, C).clone();
+  auto dgamma = sums.narrow(0, C, C).clone();
   return {dx, dgamma, dbeta};
 }

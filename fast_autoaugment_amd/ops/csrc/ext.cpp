@@ -11,8 +11,8 @@ std::vector<torch::Tensor> label_smooth_ce_fwd(torch::Tensor logits, torch::Tens
 torch::Tensor label_smooth_ce_bwd(torch::Tensor softmax, torch::Tensor target,
                                   torch::Tensor grad_loss, double eps);
 void sgd_fused_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
-                    torch::Tensor normsq, int64_t n_decay, double wd, double clip,
-                    double lr, double momentum, int64_t nesterov);
+                    torch::Tensor normsq, torch::Tensor lr_t, int64_t n_decay,
+                    double wd, double clip, double momentum, int64_t nesterov);
 void ema_lerp_(torch::Tensor shadow, torch::Tensor x, double mu);
 torch::Tensor aug_pipeline(torch::Tensor images, torch::Tensor sel, torch::Tensor prog,
                            torch::Tensor post, torch::Tensor mean, torch::Tensor std,

@@ -52,7 +52,9 @@ __global__ void wd_norm_kernel(float* __restrict__ g, const float* __restrict__ 
 
 __global__ void sgd_step_kernel(float* __restrict__ p, float* __restrict__ g,
                                 float* __restrict__ buf, const float* __restrict__ normsq,
-                                int64_t n, float clip, float lr, float mu, int nesterov) {
+                                const float* __restrict__ lr_p,
+                                int64_t n, float clip, float mu, int nesterov) {
+  float lr = lr_p[0];     // device-resident lr -> hipGraph-replayable
   float coef = 1.0f;
   if (clip > 0.0f) {
     float norm = sqrtf(normsq[0]);
@@ -115,8 +117,8 @@ __global__ void ema_lerp_kernel(float* __restrict__ shadow, const float* __restr
 }  // namespace
 
 void sgd_fused_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
-                    torch::Tensor normsq, int64_t n_decay, double wd, double clip,
-                    double lr, double momentum, int64_t nesterov) {
+                    torch::Tensor normsq, torch::Tensor lr_t, int64_t n_decay,
+                    double wd, double clip, double momentum, int64_t nesterov) {
   TORCH_CHECK(p.is_cuda() && p.dtype() == torch::kFloat32 && p.is_contiguous());
   int64_t n = p.numel();
   auto stream = at::hip::getCurrentHIPStream().stream();
@@ -128,7 +130,7 @@ void sgd_fused_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
                      n, n_decay, (float)wd);
   hipLaunchKernelGGL(sgd_step_kernel, dim3(grid), dim3(block), 0, stream,
                      p.data_ptr<float>(), g.data_ptr<float>(), buf.data_ptr<float>(),
-                     normsq.data_ptr<float>(), n, (float)clip, (float)lr,
+                     normsq.data_ptr<float>(), lr_t.data_ptr<float>(), n, (float)clip,
                      (float)momentum, (int)nesterov);
 }
 

@@ -79,8 +79,23 @@ class FusedSGD:
         self.grad_clip = grad_clip
         self.momentum_buf = torch.zeros_like(flat.flat_param)
         self._normsq = torch.zeros(1, device=flat.flat_param.device, dtype=torch.float32)
+        # lr lives on-device so the step kernels are hipGraph-replayable;
+        # the host writes it through a pinned staging scalar before replay
+        dev = flat.flat_param.device
+        self._lr_t = torch.full((1,), lr, device=dev, dtype=torch.float32)
+        self._lr_host = (torch.zeros(1, dtype=torch.float32, pin_memory=True)
+                         if dev.type == "cuda" else torch.zeros(1))
+        self._last_lr = lr
         # mirror of torch.optim param_groups API surface used by the trainer
         self.param_groups = [{"lr": lr}]
+
+    def sync_lr(self):
+        """Push param_groups[0]['lr'] to the device scalar (call pre-replay)."""
+        lr = self.param_groups[0]["lr"]
+        if lr != self._last_lr:
+            self._lr_host[0] = lr
+            self._lr_t.copy_(self._lr_host, non_blocking=True)
+            self._last_lr = lr
 
     def zero_grad(self, set_to_none: bool = False):
         self.flat.flat_grad.zero_()
@@ -94,8 +109,9 @@ class FusedSGD:
         if p.is_cuda:
             from .ops import ext
             C = ext()
-            C.sgd_fused_step(p, g, self.momentum_buf, self._normsq, nd,
-                             self.weight_decay, self.grad_clip, self.lr,
+            self.sync_lr()
+            C.sgd_fused_step(p, g, self.momentum_buf, self._normsq, self._lr_t, nd,
+                             self.weight_decay, self.grad_clip,
                              self.momentum, 1 if self.nesterov else 0)
             return
         # CPU reference path
