@@ -81,18 +81,21 @@ def main():
                        seed=rank, out_dtype=out_dtype, prefetch=4)
 
     model = build_model(conf["model"], nc).to(dev).to(memory_format=torch.channels_last)
+    # pure-bf16 compute: params+grads are bf16 flat views (no autocast cast
+    # kernels), fp32 master lives in the fused optimizer
+    work_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     if distributed:
         from fast_autoaugment_amd.parallel.ddp import FlatDDP
-        model = FlatDDP(model)
+        model = FlatDDP(model, work_dtype=work_dtype)
         flat = model.flat
     else:
-        flat = flatten_module(model)
+        flat = flatten_module(model, work_dtype=work_dtype)
     lr0 = conf["lr"] * world_size
     opt = FusedSGD(flat, lr=lr0, momentum=0.9, nesterov=True,
                    weight_decay=conf["optimizer"]["decay"], grad_clip=5.0)
     sched = build_scheduler(conf, opt, lr0)
     crit = CrossEntropyLabelSmooth(nc, 0.0)
-    amp_dtype = torch.bfloat16 if args.dtype == "bf16" else None
+    amp_dtype = None   # bf16 runs natively through the flat bf16 weights
 
     model.train()
     steps_per_epoch = max(len(loader), 1)
@@ -152,9 +155,8 @@ def main():
         opt.zero_grad()
         data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s, mean_t, std_t, bf16)
         label = store.labels.index_select(0, sel_s)
-        with torch.autocast("cuda", dtype=amp_dtype, enabled=amp_dtype is not None):
-            preds = model(data)
-            loss = crit(preds, label)
+        preds = model(data)
+        loss = crit(preds, label)
         loss.backward()
         if distributed:
             model.finish_gradient_sync()

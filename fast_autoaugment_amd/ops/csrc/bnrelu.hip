@@ -107,18 +107,23 @@ __global__ void bn_reduce_scalar_kernel(const T* __restrict__ x, float* __restri
   }
 }
 
+// one 64-lane wave per channel: lanes stride the partial blocks in parallel
+// (a serial per-thread loop over ~64 partials costs ~16us in pure latency)
 __global__ void bn_finalize_kernel(const float* __restrict__ scratch, int nblocks,
                                    float* __restrict__ mean, float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
                                    int C, int64_t count, float eps, float momentum) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = blockIdx.x;
   if (c >= C) return;
   float s = 0, ss = 0;
-  for (int b = 0; b < nblocks; ++b) {
+  for (int b = threadIdx.x; b < nblocks; b += blockDim.x) {
     s += scratch[(int64_t)b * 2 * C + c];
     ss += scratch[(int64_t)b * 2 * C + C + c];
   }
+  s = faa_warp_reduce_sum(s);
+  ss = faa_warp_reduce_sum(ss);
+  if (threadIdx.x != 0) return;
   float m = s / count;
   float var = fmaxf(ss / count - m * m, 0.0f);
   mean[c] = m;
@@ -131,6 +136,11 @@ __global__ void bn_finalize_kernel(const float* __restrict__ scratch, int nblock
 }
 
 // ---------------------------------------------------------------- fwd apply
+__device__ __forceinline__ short f_to_bf16_bits(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
+}
+
 template <typename T>
 __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
                                 const float* __restrict__ mean,
@@ -138,11 +148,31 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
                                 int64_t total, int C) {
-  // precompute scale/shift per channel would need C regs; just read 4 floats
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (int64_t i = i0; i < total; i += stride) {
     int c0 = (int)(i % C);
+    if constexpr (sizeof(T) == 2) {
+      if (i + 8 <= total) {
+        const short* xs = reinterpret_cast<const short*>(x);
+        short4v v0 = *reinterpret_cast<const short4v*>(xs + i);
+        short4v v1 = *reinterpret_cast<const short4v*>(xs + i + 4);
+        short4v o0, o1;
+        #pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          int ca = c0 + k; while (ca >= C) ca -= C;
+          int cb = c0 + k + 4; while (cb >= C) cb -= C;
+          float a = (bf16_bits_to_f(v0[k]) - mean[ca]) * invstd[ca] * gamma[ca] + beta[ca];
+          float b = (bf16_bits_to_f(v1[k]) - mean[cb]) * invstd[cb] * gamma[cb] + beta[cb];
+          o0[k] = f_to_bf16_bits(fmaxf(a, 0.0f));
+          o1[k] = f_to_bf16_bits(fmaxf(b, 0.0f));
+        }
+        short* os = reinterpret_cast<short*>(out);
+        *reinterpret_cast<short4v*>(os + i) = o0;
+        *reinterpret_cast<short4v*>(os + i + 4) = o1;
+        continue;
+      }
+    }
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
       int64_t j = i + k;
@@ -247,12 +277,14 @@ __global__ void bn_bwd_reduce_scalar_kernel(const T* __restrict__ x, const T* __
 
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ scratch, int nblocks,
                                        float* __restrict__ sums, int C) {
-  // sums: [2C] = {sum_dy, sum_dy_xhat}
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  // sums: [2C] = {sum_dy, sum_dy_xhat}; one wave per output element
+  int c = blockIdx.x;
   if (c >= 2 * C) return;
   float s = 0;
-  for (int b = 0; b < nblocks; ++b) s += scratch[(int64_t)b * 2 * C + c];
-  sums[c] = s;
+  for (int b = threadIdx.x; b < nblocks; b += blockDim.x)
+    s += scratch[(int64_t)b * 2 * C + c];
+  s = faa_warp_reduce_sum(s);
+  if (threadIdx.x == 0) sums[c] = s;
 }
 
 template <typename T>
@@ -268,6 +300,44 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
   float inv_count = 1.0f / (float)count;
   for (int64_t i = i0; i < total; i += stride) {
     int c0 = (int)(i % C);
+    if constexpr (sizeof(T) == 2) {
+      if (i + 8 <= total) {
+        const short* xs = reinterpret_cast<const short*>(x);
+        const short* os = reinterpret_cast<const short*>(out);
+        const short* ds = reinterpret_cast<const short*>(dy);
+        short4v xv0 = *reinterpret_cast<const short4v*>(xs + i);
+        short4v xv1 = *reinterpret_cast<const short4v*>(xs + i + 4);
+        short4v ov0 = *reinterpret_cast<const short4v*>(os + i);
+        short4v ov1 = *reinterpret_cast<const short4v*>(os + i + 4);
+        short4v dv0 = *reinterpret_cast<const short4v*>(ds + i);
+        short4v dv1 = *reinterpret_cast<const short4v*>(ds + i + 4);
+        short4v r0, r1;
+        #pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          int ca = c0 + k; while (ca >= C) ca -= C;
+          int cb = c0 + k + 4; while (cb >= C) cb -= C;
+          float ga = bf16_bits_to_f(ov0[k]) > 0.0f ? bf16_bits_to_f(dv0[k]) : 0.0f;
+          float gb = bf16_bits_to_f(ov1[k]) > 0.0f ? bf16_bits_to_f(dv1[k]) : 0.0f;
+          float isa = invstd[ca], isb = invstd[cb];
+          float ra, rb;
+          if (training) {
+            float xa = (bf16_bits_to_f(xv0[k]) - mean[ca]) * isa;
+            float xb = (bf16_bits_to_f(xv1[k]) - mean[cb]) * isb;
+            ra = gamma[ca] * isa * (ga - sums[ca] * inv_count - xa * sums[C + ca] * inv_count);
+            rb = gamma[cb] * isb * (gb - sums[cb] * inv_count - xb * sums[C + cb] * inv_count);
+          } else {
+            ra = gamma[ca] * isa * ga;
+            rb = gamma[cb] * isb * gb;
+          }
+          r0[k] = f_to_bf16_bits(ra);
+          r1[k] = f_to_bf16_bits(rb);
+        }
+        short* dxs = reinterpret_cast<short*>(dx);
+        *reinterpret_cast<short4v*>(dxs + i) = r0;
+        *reinterpret_cast<short4v*>(dxs + i + 4) = r1;
+        continue;
+      }
+    }
     #pragma unroll
     for (int k = 0; k < 8; ++k) {
       int64_t j = i + k;
@@ -334,7 +404,7 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                            stream, (const scalar_t*)xc.data_ptr(),
                            scratch.data_ptr<float>(), total, C);
       });
-      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
                          scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
                          invstd.data_ptr<float>(),
                          running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
@@ -348,7 +418,7 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                            stream, (const scalar_t*)xc.data_ptr(),
                            scratch.data_ptr<float>(), total, C);
       });
-      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
                          scratch.data_ptr<float>(), 1, mean.data_ptr<float>(),
                          invstd.data_ptr<float>(),
                          running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
@@ -395,7 +465,7 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                          (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
     });
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((2 * C + 255) / 256), dim3(256), 0,
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(2 * C), dim3(64), 0,
                        stream, scratch.data_ptr<float>(), nb,
                        sums.data_ptr<float>(), C);
   } else {

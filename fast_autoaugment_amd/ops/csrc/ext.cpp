@@ -13,6 +13,10 @@ torch::Tensor label_smooth_ce_bwd(torch::Tensor softmax, torch::Tensor target,
 void sgd_fused_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
                     torch::Tensor normsq, torch::Tensor lr_t, int64_t n_decay,
                     double wd, double clip, double momentum, int64_t nesterov);
+void sgd_fused_step_mixed(torch::Tensor master, torch::Tensor work, torch::Tensor g,
+                          torch::Tensor buf, torch::Tensor normsq, torch::Tensor lr_t,
+                          int64_t n_decay, double wd, double clip, double momentum,
+                          int64_t nesterov);
 void ema_lerp_(torch::Tensor shadow, torch::Tensor x, double mu);
 torch::Tensor aug_pipeline(torch::Tensor images, torch::Tensor sel, torch::Tensor prog,
                            torch::Tensor post, torch::Tensor mean, torch::Tensor std,
@@ -37,6 +41,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("label_smooth_ce_bwd", &label_smooth_ce_bwd);
   m.def("sgd_fused_step", &sgd_fused_step,
         "fused manual-WD + global-clip + nesterov SGD on flat buffers");
+  m.def("sgd_fused_step_mixed", &sgd_fused_step_mixed,
+        "mixed bf16-work/fp32-master fused SGD step");
   m.def("ema_lerp_", &ema_lerp_);
   m.def("aug_pipeline", &aug_pipeline,
         "batched augmentation program executor (uint8 NHWC -> normalized bf16/f32)");

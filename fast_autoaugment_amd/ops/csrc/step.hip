@@ -114,7 +114,124 @@ __global__ void ema_lerp_kernel(float* __restrict__ shadow, const float* __restr
   }
 }
 
+// ---- mixed-precision variant: bf16 working weights/grads + fp32 master ----
+// The model computes natively in bf16 (no autocast cast kernels); grads land
+// in a flat bf16 buffer; the step reads bf16 grads, applies manual WD from
+// the fp32 master, clips by the global norm, updates the fp32 master and
+// re-quantizes the bf16 working copy. Norm pass is read-only (no wd write).
+
+typedef __attribute__((ext_vector_type(4))) short short4x;
+
+__device__ __forceinline__ float bits_to_f32(short u) {
+  union { float f; uint32_t i; } c;
+  c.i = ((uint32_t)(uint16_t)u) << 16;
+  return c.f;
+}
+
+__global__ void wd_norm_mixed_kernel(const short* __restrict__ g,
+                                     const float* __restrict__ p,
+                                     float* __restrict__ normsq,
+                                     int64_t n, int64_t n_decay, float wd) {
+  __shared__ float lds[8];
+  int64_t i0 = (int64_t)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  float acc = 0.0f;
+  for (int64_t i = i0; i < n; i += stride) {
+    if (i + 3 < n) {
+      short4x gv = *reinterpret_cast<const short4x*>(g + i);
+      #pragma unroll
+      for (int v = 0; v < 4; ++v) {
+        float gg = bits_to_f32(gv[v]);
+        if (wd != 0.0f && i + v < n_decay) gg += wd * p[i + v];
+        acc += gg * gg;
+      }
+    } else {
+      for (int64_t j = i; j < n; ++j) {
+        float gg = bits_to_f32(g[j]);
+        if (wd != 0.0f && j < n_decay) gg += wd * p[j];
+        acc += gg * gg;
+      }
+    }
+  }
+  acc = faa_block_reduce_sum(acc, lds);
+  if (threadIdx.x == 0) atomicAdd(normsq, acc);
+}
+
+__global__ void sgd_step_mixed_kernel(float* __restrict__ p, short* __restrict__ w,
+                                      const short* __restrict__ g,
+                                      float* __restrict__ buf,
+                                      const float* __restrict__ normsq,
+                                      const float* __restrict__ lr_p,
+                                      int64_t n, int64_t n_decay, float wd,
+                                      float clip, float mu, int nesterov) {
+  float lr = lr_p[0];
+  float coef = 1.0f;
+  if (clip > 0.0f) {
+    float norm = sqrtf(normsq[0]);
+    coef = fminf(1.0f, clip / (norm + 1e-6f));
+  }
+  int64_t i0 = (int64_t)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i = i0; i < n; i += stride) {
+    if (i + 3 < n) {
+      short4x gv = *reinterpret_cast<const short4x*>(g + i);
+      float4 bv = *reinterpret_cast<float4*>(buf + i);
+      float4 pv = *reinterpret_cast<float4*>(p + i);
+      short4x wv;
+      #pragma unroll
+      for (int v = 0; v < 4; ++v) {
+        float* bp = &bv.x + v; float* pp = &pv.x + v;
+        float gg = bits_to_f32(gv[v]);
+        if (wd != 0.0f && i + v < n_decay) gg += wd * *pp;
+        gg *= coef;
+        float bb = mu * *bp + gg;
+        float upd = nesterov ? (gg + mu * bb) : bb;
+        *bp = bb;
+        *pp = *pp - lr * upd;
+        __hip_bfloat16 h = __float2bfloat16(*pp);
+        wv[v] = *reinterpret_cast<short*>(&h);
+      }
+      *reinterpret_cast<float4*>(buf + i) = bv;
+      *reinterpret_cast<float4*>(p + i) = pv;
+      *reinterpret_cast<short4x*>(w + i) = wv;
+    } else {
+      for (int64_t j = i; j < n; ++j) {
+        float gg = bits_to_f32(g[j]);
+        if (wd != 0.0f && j < n_decay) gg += wd * p[j];
+        gg *= coef;
+        float bb = mu * buf[j] + gg;
+        float upd = nesterov ? (gg + mu * bb) : bb;
+        buf[j] = bb;
+        p[j] -= lr * upd;
+        __hip_bfloat16 h = __float2bfloat16(p[j]);
+        w[j] = *reinterpret_cast<short*>(&h);
+      }
+    }
+  }
+}
+
 }  // namespace
+
+void sgd_fused_step_mixed(torch::Tensor master, torch::Tensor work, torch::Tensor g,
+                          torch::Tensor buf, torch::Tensor normsq, torch::Tensor lr_t,
+                          int64_t n_decay, double wd, double clip, double momentum,
+                          int64_t nesterov) {
+  TORCH_CHECK(master.dtype() == torch::kFloat32 && work.dtype() == torch::kBFloat16
+              && g.dtype() == torch::kBFloat16);
+  int64_t n = master.numel();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const int block = 256;
+  int grid = faa_grid(n / 4 + 1, block);
+  hipLaunchKernelGGL(zero1_kernel, dim3(1), dim3(64), 0, stream, normsq.data_ptr<float>());
+  hipLaunchKernelGGL(wd_norm_mixed_kernel, dim3(grid), dim3(block), 0, stream,
+                     (const short*)g.data_ptr(), master.data_ptr<float>(),
+                     normsq.data_ptr<float>(), n, n_decay, (float)wd);
+  hipLaunchKernelGGL(sgd_step_mixed_kernel, dim3(grid), dim3(block), 0, stream,
+                     master.data_ptr<float>(), (short*)work.data_ptr(),
+                     (const short*)g.data_ptr(), buf.data_ptr<float>(),
+                     normsq.data_ptr<float>(), lr_t.data_ptr<float>(), n, n_decay,
+                     (float)wd, (float)clip, (float)momentum, (int)nesterov);
+}
 
 void sgd_fused_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
                     torch::Tensor normsq, torch::Tensor lr_t, int64_t n_decay,

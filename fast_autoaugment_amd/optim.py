@@ -77,7 +77,8 @@ class FusedSGD:
         self.nesterov = nesterov
         self.weight_decay = weight_decay
         self.grad_clip = grad_clip
-        self.momentum_buf = torch.zeros_like(flat.flat_param)
+        base = flat.flat_master if flat.flat_master is not None else flat.flat_param
+        self.momentum_buf = torch.zeros_like(base)
         self._normsq = torch.zeros(1, device=flat.flat_param.device, dtype=torch.float32)
         # lr lives on-device so the step kernels are hipGraph-replayable;
         # the host writes it through a pinned staging scalar before replay
@@ -106,26 +107,37 @@ class FusedSGD:
         p = self.flat.flat_param
         g = self.flat.flat_grad
         nd = self.flat.n_decay
+        master = self.flat.flat_master
         if p.is_cuda:
             from .ops import ext
             C = ext()
             self.sync_lr()
-            C.sgd_fused_step(p, g, self.momentum_buf, self._normsq, self._lr_t, nd,
-                             self.weight_decay, self.grad_clip,
-                             self.momentum, 1 if self.nesterov else 0)
+            if master is not None:
+                C.sgd_fused_step_mixed(master, p, g, self.momentum_buf, self._normsq,
+                                       self._lr_t, nd, self.weight_decay,
+                                       self.grad_clip, self.momentum,
+                                       1 if self.nesterov else 0)
+            else:
+                C.sgd_fused_step(p, g, self.momentum_buf, self._normsq, self._lr_t, nd,
+                                 self.weight_decay, self.grad_clip,
+                                 self.momentum, 1 if self.nesterov else 0)
             return
-        # CPU reference path
+        # CPU reference path (also defines the mixed-kernel semantics)
+        pm = master if master is not None else p
+        gf = g.float()
         if self.weight_decay > 0 and nd > 0:
-            g[:nd].add_(p[:nd], alpha=self.weight_decay)
+            gf[:nd].add_(pm[:nd].float(), alpha=self.weight_decay)
         if self.grad_clip > 0:
-            total = g.norm(2)
+            total = gf.norm(2)
             coef = min(1.0, self.grad_clip / (float(total) + 1e-6))
             if coef < 1.0:
-                g.mul_(coef)
+                gf.mul_(coef)
         buf = self.momentum_buf
-        buf.mul_(self.momentum).add_(g)
-        upd = g.add(buf, alpha=self.momentum) if self.nesterov else buf
-        p.add_(upd, alpha=-self.lr)
+        buf.mul_(self.momentum).add_(gf)
+        upd = gf.add(buf, alpha=self.momentum) if self.nesterov else buf
+        pm.add_(upd.to(pm.dtype), alpha=-self.lr)
+        if master is not None:
+            p.copy_(master.to(p.dtype))
 
     def state_dict(self):
         return {"momentum_buf": self.momentum_buf, "lr": self.lr,

@@ -27,12 +27,14 @@ from .flat import FlatParams, flatten_module
 
 class FlatDDP(nn.Module):
     def __init__(self, module: nn.Module, bucket_bytes: int = 16 << 20,
-                 process_group=None):
+                 process_group=None, work_dtype=None):
         super().__init__()
         self.module = module
         self.pg = process_group
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
-        self.flat: FlatParams = flatten_module(module)
+        import torch as _t
+        self.flat: FlatParams = flatten_module(
+            module, work_dtype=work_dtype or _t.float32)
         self._comm_stream: Optional[torch.cuda.Stream] = (
             torch.cuda.Stream() if self.flat.flat_param.is_cuda else None)
         self._sync_enabled = True

@@ -33,9 +33,11 @@ def bn_param_names(model: nn.Module) -> set:
 
 class FlatParams:
     def __init__(self, flat_param: torch.Tensor, flat_grad: torch.Tensor,
-                 n_decay: int, params: List[nn.Parameter]):
-        self.flat_param = flat_param
+                 n_decay: int, params: List[nn.Parameter],
+                 flat_master: torch.Tensor = None):
+        self.flat_param = flat_param      # the working buffer (views back it)
         self.flat_grad = flat_grad
+        self.flat_master = flat_master    # fp32 master when work dtype != fp32
         self.n_decay = n_decay            # elements in the decay segment
         self.params = params
 
@@ -44,10 +46,16 @@ class FlatParams:
         return self.flat_param.numel()
 
 
-def flatten_module(model: nn.Module, align: int = 64) -> FlatParams:
-    """Rebuild model params as views of one flat fp32 buffer.
+def flatten_module(model: nn.Module, align: int = 64,
+                   work_dtype: torch.dtype = torch.float32) -> FlatParams:
+    """Rebuild model params as views of one flat buffer.
 
-    align: element alignment per tensor (64 floats = 256 B) so each view is
+    work_dtype=float32: classic flat fp32 params+grads.
+    work_dtype=bfloat16: pure-bf16 compute — params AND grads are bf16 views
+    (no autocast cast kernels anywhere in fwd/bwd), with a flat fp32 master
+    updated by the mixed fused SGD kernel which re-quantizes the working copy.
+
+    align: element alignment per tensor (64 elements) keeps each view
     vector-load friendly in the fused kernels.
     """
     named = list(model.named_parameters())
@@ -70,15 +78,21 @@ def flatten_module(model: nn.Module, align: int = 64) -> FlatParams:
             n_decay = offsets[i][0] + padded(p.numel())
 
     device = ordered[0][1].device if ordered else torch.device("cpu")
-    flat_param = torch.zeros(total, dtype=torch.float32, device=device)
-    flat_grad = torch.zeros(total, dtype=torch.float32, device=device)
+    flat_param = torch.zeros(total, dtype=work_dtype, device=device)
+    flat_grad = torch.zeros(total, dtype=work_dtype, device=device)
+    flat_master = None
+    if work_dtype != torch.float32:
+        flat_master = torch.zeros(total, dtype=torch.float32, device=device)
 
     params = []
     for (name, p), (off, n) in zip(ordered, offsets):
+        if flat_master is not None:
+            with torch.no_grad():
+                flat_master[off:off + n].view_as(p).copy_(p.detach().float())
         view = flat_param[off:off + n].view_as(p)
         with torch.no_grad():
-            view.copy_(p.detach().float())
+            view.copy_(p.detach().to(work_dtype))
         p.data = view
         p.grad = flat_grad[off:off + n].view_as(p)
         params.append(p)
-    return FlatParams(flat_param, flat_grad, n_decay, params)
+    return FlatParams(flat_param, flat_grad, n_decay, params, flat_master)

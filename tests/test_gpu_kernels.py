@@ -314,3 +314,30 @@ def test_wrn_forward_backward_gpu():
     loss.backward()
     assert torch.isfinite(loss)
     assert all(torch.isfinite(p.grad).all() for p in m.parameters() if p.grad is not None)
+
+
+def test_sgd_fused_step_mixed_matches_cpu():
+    """bf16 working weights + fp32 master: GPU kernel vs CPU reference path."""
+    from fast_autoaugment_amd.parallel.flat import flatten_module
+    from fast_autoaugment_amd.optim import FusedSGD
+    torch.manual_seed(0)
+    mk = lambda: torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3, padding=1),
+                                     torch.nn.BatchNorm2d(8),
+                                     torch.nn.Conv2d(8, 8, 1))
+    m_gpu, m_cpu = mk().to(dev()), mk()
+    m_cpu.load_state_dict(m_gpu.state_dict())
+    fg = flatten_module(m_gpu, work_dtype=torch.bfloat16)
+    fc = flatten_module(m_cpu, work_dtype=torch.bfloat16)
+    og = FusedSGD(fg, lr=0.1, momentum=0.9, nesterov=True, weight_decay=0.01, grad_clip=5.0)
+    oc = FusedSGD(fc, lr=0.1, momentum=0.9, nesterov=True, weight_decay=0.01, grad_clip=5.0)
+    x = torch.randn(4, 3, 8, 8)
+    for i in range(3):
+        for m, o, xx in [(m_gpu, og, x.to(dev()).bfloat16()), (m_cpu, oc, x.bfloat16())]:
+            o.zero_grad()
+            m(xx).float().square().mean().backward()
+            o.step()
+    # bf16 grads differ slightly between CPU/GPU conv kernels; masters track
+    assert (fg.flat_master.cpu() - fc.flat_master).abs().max().item() < 5e-3
+    # working copy is the bf16 quantization of the master
+    assert (fg.flat_param.float().cpu() -
+            fg.flat_master.cpu().bfloat16().float()).abs().max().item() == 0
