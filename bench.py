@@ -90,6 +90,11 @@ def main():
         flat = model.flat
     else:
         flat = flatten_module(model, work_dtype=work_dtype)
+    if work_dtype == torch.bfloat16:
+        from fast_autoaugment_amd.ops.conv import patch_convs
+        n_patched = patch_convs(model)
+        if rank == 0:
+            print(f"# {n_patched} convs on MFMA kernels", flush=True)
     lr0 = conf["lr"] * world_size
     opt = FusedSGD(flat, lr=lr0, momentum=0.9, nesterov=True,
                    weight_decay=conf["optimizer"]["decay"], grad_clip=5.0)

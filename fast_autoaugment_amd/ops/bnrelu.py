@@ -21,7 +21,7 @@ class FusedBNReLUFn(torch.autograd.Function):
         x, out, mean, invstd, gamma = ctx.saved_tensors
         C = ext()
         dx, dgamma, dbeta = C.bn_relu_bwd(dy, x, out, mean, invstd, gamma, ctx.training)
-        return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, None, None, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 def fused_bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:

@@ -1,0 +1,82 @@
+"""Custom MFMA conv dispatch.
+
+`patch_convs(model)` rebinds every eligible nn.Conv2d's forward to the
+in-house CDNA4 implicit-GEMM kernels (csrc/conv_mfma.hip): bf16 NHWC,
+square kernel 1 or 3 with SAME-style padding, groups=1, dilation=1.
+Ineligible convs (grouped/depthwise, exotic kernels, non-bf16) keep the
+torch/MIOpen path. The stride-2 backward-data piece also falls back to
+torch.nn.grad.conv2d_input until the scatter variant lands.
+"""
+from __future__ import annotations
+
+import types
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import ext
+
+
+class FaaConv2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding):
+        C = ext()
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        ctx.padding = padding
+        ctx.has_bias = bias is not None
+        b = bias if bias is not None else torch.Tensor()
+        return C.conv2d_fwd(x, weight, b, stride, padding)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        C = ext()
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = dw = dbias = None
+        if ctx.needs_input_grad[0]:
+            if ctx.stride == 1:
+                dx = C.conv2d_bwd_data(dy, weight, 1, ctx.padding,
+                                       x.size(2), x.size(3))
+            else:
+                dx = torch.nn.grad.conv2d_input(list(x.shape), weight, dy,
+                                                stride=ctx.stride,
+                                                padding=ctx.padding)
+        if ctx.needs_input_grad[1] or (ctx.has_bias and ctx.needs_input_grad[2]):
+            dw, db = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
+                                         weight.size(2), weight.size(3),
+                                         ctx.has_bias)
+            if ctx.has_bias and ctx.needs_input_grad[2]:
+                dbias = db
+        return dx, dw, dbias, None, None
+
+
+def faa_conv2d(x, weight, bias, stride: int, padding: int):
+    return FaaConv2dFn.apply(x, weight, bias, stride, padding)
+
+
+def _eligible(m: torch.nn.Conv2d) -> bool:
+    k = m.kernel_size
+    return (k[0] == k[1] and k[0] in (1, 3)
+            and m.padding[0] == m.padding[1] and m.padding[0] == k[0] // 2
+            and m.stride[0] == m.stride[1] and m.stride[0] in (1, 2)
+            and m.dilation == (1, 1) and m.groups == 1)
+
+
+def _faa_forward(self, x):
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and self.weight.dtype == torch.bfloat16):
+        return faa_conv2d(x, self.weight, self.bias, self.stride[0], self.padding[0])
+    return F.conv2d(x, self.weight, self.bias, self.stride, self.padding,
+                    self.dilation, self.groups)
+
+
+def patch_convs(model: torch.nn.Module) -> int:
+    """Rebind eligible Conv2d forwards to the MFMA kernels. Returns count."""
+    n = 0
+    for m in model.modules():
+        if isinstance(m, torch.nn.Conv2d) and _eligible(m):
+            m.forward = types.MethodType(_faa_forward, m)
+            n += 1
+    return n

@@ -141,12 +141,12 @@ __device__ __forceinline__ short f_to_bf16_bits(float f) {
   return *reinterpret_cast<short*>(&h);
 }
 
-template <typename T>
+template <typename T, typename GT>
 __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
-                                const float* __restrict__ gamma,
-                                const float* __restrict__ beta,
+                                const GT* __restrict__ gamma,
+                                const GT* __restrict__ beta,
                                 int64_t total, int C) {
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
@@ -162,8 +162,8 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
         for (int k = 0; k < 4; ++k) {
           int ca = c0 + k; while (ca >= C) ca -= C;
           int cb = c0 + k + 4; while (cb >= C) cb -= C;
-          float a = (bf16_bits_to_f(v0[k]) - mean[ca]) * invstd[ca] * gamma[ca] + beta[ca];
-          float b = (bf16_bits_to_f(v1[k]) - mean[cb]) * invstd[cb] * gamma[cb] + beta[cb];
+          float a = (bf16_bits_to_f(v0[k]) - mean[ca]) * invstd[ca] * faa_to_float(gamma[ca]) + faa_to_float(beta[ca]);
+          float b = (bf16_bits_to_f(v1[k]) - mean[cb]) * invstd[cb] * faa_to_float(gamma[cb]) + faa_to_float(beta[cb]);
           o0[k] = f_to_bf16_bits(fmaxf(a, 0.0f));
           o1[k] = f_to_bf16_bits(fmaxf(b, 0.0f));
         }
@@ -178,7 +178,7 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
       int64_t j = i + k;
       if (j < total) {
         int c = c0 + k; while (c >= C) c -= C;
-        float v = (faa_to_float(x[j]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+        float v = (faa_to_float(x[j]) - mean[c]) * invstd[c] * faa_to_float(gamma[c]) + faa_to_float(beta[c]);
         faa_from_float(fmaxf(v, 0.0f), &out[j]);
       }
     }
@@ -275,24 +275,31 @@ __global__ void bn_bwd_reduce_scalar_kernel(const T* __restrict__ x, const T* __
   }
 }
 
+template <typename GT>
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ scratch, int nblocks,
-                                       float* __restrict__ sums, int C) {
-  // sums: [2C] = {sum_dy, sum_dy_xhat}; one wave per output element
+                                       float* __restrict__ sums,
+                                       GT* __restrict__ dbeta, GT* __restrict__ dgamma,
+                                       int C) {
+  // sums: [2C] = {sum_dy (=dbeta), sum_dy_xhat (=dgamma)}; one wave per element
   int c = blockIdx.x;
   if (c >= 2 * C) return;
   float s = 0;
   for (int b = threadIdx.x; b < nblocks; b += blockDim.x)
     s += scratch[(int64_t)b * 2 * C + c];
   s = faa_warp_reduce_sum(s);
-  if (threadIdx.x == 0) sums[c] = s;
+  if (threadIdx.x == 0) {
+    sums[c] = s;
+    if (c < C) faa_from_float(s, &dbeta[c]);
+    else faa_from_float(s, &dgamma[c - C]);
+  }
 }
 
-template <typename T>
+template <typename T, typename GT>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict__ out,
                                     const T* __restrict__ dy, T* __restrict__ dx,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
-                                    const float* __restrict__ gamma,
+                                    const GT* __restrict__ gamma,
                                     const float* __restrict__ sums,
                                     int64_t total, int C, int64_t count, int training) {
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
@@ -323,11 +330,11 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
           if (training) {
             float xa = (bf16_bits_to_f(xv0[k]) - mean[ca]) * isa;
             float xb = (bf16_bits_to_f(xv1[k]) - mean[cb]) * isb;
-            ra = gamma[ca] * isa * (ga - sums[ca] * inv_count - xa * sums[C + ca] * inv_count);
-            rb = gamma[cb] * isb * (gb - sums[cb] * inv_count - xb * sums[C + cb] * inv_count);
+            ra = faa_to_float(gamma[ca]) * isa * (ga - sums[ca] * inv_count - xa * sums[C + ca] * inv_count);
+            rb = faa_to_float(gamma[cb]) * isb * (gb - sums[cb] * inv_count - xb * sums[C + cb] * inv_count);
           } else {
-            ra = gamma[ca] * isa * ga;
-            rb = gamma[cb] * isb * gb;
+            ra = faa_to_float(gamma[ca]) * isa * ga;
+            rb = faa_to_float(gamma[cb]) * isb * gb;
           }
           r0[k] = f_to_bf16_bits(ra);
           r1[k] = f_to_bf16_bits(rb);
@@ -349,9 +356,9 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
         float res;
         if (training) {
           float xhat = (faa_to_float(x[j]) - mean[c]) * is;
-          res = gamma[c] * is * (g - sums[c] * inv_count - xhat * sums[C + c] * inv_count);
+          res = faa_to_float(gamma[c]) * is * (g - sums[c] * inv_count - xhat * sums[C + c] * inv_count);
         } else {
-          res = gamma[c] * is * g;
+          res = faa_to_float(gamma[c]) * is * g;
         }
         faa_from_float(res, &dx[j]);
       }
@@ -390,9 +397,9 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   auto mean = torch::empty({C}, f32);
   auto invstd = torch::empty({C}, f32);
   auto stream = at::hip::getCurrentHIPStream().stream();
-
-  auto g = gamma.to(torch::kFloat32).contiguous();
-  auto bta = beta.to(torch::kFloat32).contiguous();
+  auto g = gamma.contiguous();
+  auto bta = beta.contiguous();
+  TORCH_CHECK(g.scalar_type() == bta.scalar_type(), "gamma/beta dtype mismatch");
 
   if (training) {
     bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
@@ -431,10 +438,14 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   }
   int grid1 = faa_grid(total / 8 + 1, 256);
   DISPATCH_FB(xc.scalar_type(), "bn_apply", [&] {
-    hipLaunchKernelGGL((bn_apply_kernel<scalar_t>), dim3(grid1), dim3(256), 0, stream,
-                       (const scalar_t*)xc.data_ptr(), (scalar_t*)out.data_ptr(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       g.data_ptr<float>(), bta.data_ptr<float>(), total, C);
+    using data_t = scalar_t;
+    DISPATCH_FB(g.scalar_type(), "bn_apply_g", [&] {
+      hipLaunchKernelGGL((bn_apply_kernel<data_t, scalar_t>), dim3(grid1), dim3(256), 0,
+                         stream, (const data_t*)xc.data_ptr(), (data_t*)out.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         (const scalar_t*)g.data_ptr(), (const scalar_t*)bta.data_ptr(),
+                         total, C);
+    });
   });
   return {out, mean, invstd};
 }
@@ -451,8 +462,10 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
   int64_t rows = total / C;
   auto f32 = xc.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(xc);
-  auto g = gamma.to(torch::kFloat32).contiguous();
+  auto g = gamma.contiguous();
   auto sums = torch::empty({2 * C}, f32);
+  auto dgamma = torch::empty({C}, xc.options().dtype(g.scalar_type()));
+  auto dbeta = torch::empty({C}, xc.options().dtype(g.scalar_type()));
   auto stream = at::hip::getCurrentHIPStream().stream();
 
   bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
@@ -465,9 +478,11 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                          (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
     });
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(2 * C), dim3(64), 0,
-                       stream, scratch.data_ptr<float>(), nb,
-                       sums.data_ptr<float>(), C);
+    DISPATCH_FB(g.scalar_type(), "bn_bwd_fin", [&] {
+      hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
+                         stream, scratch.data_ptr<float>(), nb, sums.data_ptr<float>(),
+                         (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
+    });
   } else {
     sums.zero_();
     int grid = faa_grid(total, 256, 1024);
@@ -477,17 +492,20 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                          (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), sums.data_ptr<float>(), total, C);
     });
+    dbeta.copy_(sums.narrow(0, 0, C));
+    dgamma.copy_(sums.narrow(0, C, C));
   }
   int grid1 = faa_grid(total / 8 + 1, 256);
   DISPATCH_FB(xc.scalar_type(), "bn_bwd_apply", [&] {
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<scalar_t>), dim3(grid1), dim3(256), 0, stream,
-                       (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
-                       (const scalar_t*)dyc.data_ptr(), (scalar_t*)dx.data_ptr(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       g.data_ptr<float>(), sums.data_ptr<float>(),
-                       total, C, rows, training ? 1 : 0);
+    using data_t = scalar_t;
+    DISPATCH_FB(g.scalar_type(), "bn_bwd_apply_g", [&] {
+      hipLaunchKernelGGL((bn_bwd_apply_kernel<data_t, scalar_t>), dim3(grid1), dim3(256), 0,
+                         stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
+                         (const data_t*)dyc.data_ptr(), (data_t*)dx.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         (const scalar_t*)g.data_ptr(), sums.data_ptr<float>(),
+                         total, C, rows, training ? 1 : 0);
+    });
   });
-  auto dbeta = sums.narrow(0, 0, C).clone();
-  auto dgamma = sums.narrow(0, C, C).clone();
   return {dx, dgamma, dbeta};
 }

@@ -1,0 +1,461 @@
+// MFMA implicit-GEMM convolution for NHWC bf16 (CDNA4 / gfx950).
+//
+// The reference delegates all conv work to cuDNN; MIOpen's
+// igemm kernels on these CIFAR shapes (3x3/1x1, C<=640, 8..32px, batch 128)
+// run 20-40us each plus SubTensor/cast helper launches. These kernels map
+// the conv directly onto mfma_f32_16x16x32_bf16 with LDS-staged tiles:
+//
+//   fwd:       Y[m=(b,ho,wo)][n=cout] = sum_k A[m][k=(kh,kw,ci)] * W[n][k]
+//              (weights in torch channels_last layout [Cout][KH][KW][Cin]
+//               ARE [n][k] with k contiguous - no repack needed); bias fused.
+//   bwd-data:  same kernel on dY with flipped/transposed weights (repacked
+//              by a tiny kernel); stride-1 only (s=2 falls back to MIOpen).
+//   bwd-weight:dW[k][n] = sum_m A[m][k] * dY[m][n], split-K over m-slices
+//              with fp32 atomic accumulation, then cast back to the torch
+//              weight layout. dbias via a column-sum kernel.
+//
+// Tiles: BM=64 x BN=64 x BK=32, 4 waves/block (each wave a 32x32 C-tile of
+// 2x2 16x16 fragments), LDS rows padded to 40 bf16 to spread banks,
+// double-buffered. A-operand per-lane layout: row=lane%16, 8 contiguous k at
+// (lane/16)*8; B likewise with n=lane%16 (verified against torch in
+// tests/test_gpu_conv.py with asymmetric data per the CDNA4 guide G9).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "faa_common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int BM = 64, BN = 64, BK = 32;
+constexpr int LDSP = 40;             // padded LDS row stride (elems)
+
+__device__ __forceinline__ float b2f(short u) {
+  union { float f; uint32_t i; } c;
+  c.i = ((uint32_t)(uint16_t)u) << 16;
+  return c.f;
+}
+__device__ __forceinline__ short f2b(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
+}
+
+struct ConvGeom {
+  int B, H, Wd, Cin, Ho, Wo, Cout;
+  int KH, KW, stride, pad;
+  int cin_chunks;      // ceil(Cin/8)
+  int kpad;            // KH*KW*cin_chunks*8
+};
+
+// load 8 input channels (zero-filled out of bounds / tail) as bf16x8
+__device__ __forceinline__ bf16x8 load_x8(const short* __restrict__ X,
+                                          const ConvGeom g, int b, int hi, int wi,
+                                          int ci0) {
+  bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (b < g.B && hi >= 0 && hi < g.H && wi >= 0 && wi < g.Wd && ci0 < g.Cin) {
+    const short* p = X + (((int64_t)b * g.H + hi) * g.Wd + wi) * g.Cin + ci0;
+    if (ci0 + 8 <= g.Cin) {
+      v = *reinterpret_cast<const bf16x8*>(p);
+    } else {
+      for (int j = 0; j < g.Cin - ci0; ++j) v[j] = p[j];
+    }
+  }
+  return v;
+}
+
+// ------------------------------------------------------------------- fwd
+
+template <bool HAS_BIAS>
+__global__ __launch_bounds__(256)
+void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
+                     const short* __restrict__ bias, short* __restrict__ Y,
+                     ConvGeom g, int M, int grid_m) {
+  __shared__ short ldsA[2][BM * LDSP];
+  __shared__ short ldsB[2][BN * LDSP];
+
+  // XCD-aware swizzle: consecutive logical tiles share B-panels; keep them
+  // on one XCD's L2 (guide T1). Bijective for any grid size.
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, idx = wg / 8;
+    if (q > 0) wg = (xcd < r) ? (xcd * (q + 1) + idx) : (r * (q + 1) + (xcd - r) * q + idx);
+  }
+  int bm = wg % grid_m;
+  int bn = wg / grid_m;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 32;     // wave row offset in tile
+  const int wc = (wave & 1) * 32;      // wave col offset
+
+  // per-thread A staging coords: row tid/4, k-chunk tid%4
+  const int a_row = tid >> 2;
+  const int a_kc = (tid & 3);
+  const int m_g = bm * BM + a_row;
+  int xb = 0, xho = 0, xwo = 0;
+  if (m_g < M) {
+    xb = m_g / (g.Ho * g.Wo);
+    int rem = m_g - xb * (g.Ho * g.Wo);
+    xho = rem / g.Wo;
+    xwo = rem - xho * g.Wo;
+  } else {
+    xb = g.B;  // out of range flag
+  }
+  // B staging: row tid/4 (cout), k-chunk tid%4
+  const int b_row = tid >> 2;
+  const int n_g = bn * BN + b_row;
+
+  const int nk = g.kpad / BK;
+  f32x4 acc[2][2] = {};
+
+  auto stage = [&](int kt, int buf) {
+    // A: 8 channels for (cell, ci) chunk
+    {
+      int kc = kt * 4 + a_kc;                    // global 8-chunk index
+      int cell = kc / g.cin_chunks;
+      int ci0 = (kc - cell * g.cin_chunks) * 8;
+      int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
+      int hi = xho * g.stride - g.pad + kh;
+      int wi = xwo * g.stride - g.pad + kw;
+      bf16x8 v = load_x8(X, g, xb, hi, wi, ci0);
+      short* dst = &ldsA[buf][a_row * LDSP + a_kc * 8];
+      *reinterpret_cast<bf16x8*>(dst) = v;
+    }
+    // B: weights [n][k] with k = cell*Cin + ci contiguous
+    {
+      int kc = kt * 4 + a_kc;
+      int cell = kc / g.cin_chunks;
+      int ci0 = (kc - cell * g.cin_chunks) * 8;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (n_g < g.Cout && ci0 < g.Cin) {
+        const short* p = Wt + ((int64_t)n_g * g.KH * g.KW + cell) * g.Cin + ci0;
+        if (ci0 + 8 <= g.Cin) {
+          v = *reinterpret_cast<const bf16x8*>(p);
+        } else {
+          for (int j = 0; j < g.Cin - ci0; ++j) v[j] = p[j];
+        }
+      }
+      short* dst = &ldsB[buf][b_row * LDSP + a_kc * 8];
+      *reinterpret_cast<bf16x8*>(dst) = v;
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  const int fr = lane & 15;            // fragment row/col within 16
+  const int kq = (lane >> 4) * 8;      // 8-elem k offset
+
+  for (int kt = 0; kt < nk; ++kt) {
+    int buf = kt & 1;
+    if (kt + 1 < nk) {
+      // prefetch next tile into the other LDS buffer AFTER compute barrier
+    }
+    bf16x8 afrag[2], bfrag[2];
+    #pragma unroll
+    for (int f = 0; f < 2; ++f) {
+      afrag[f] = *reinterpret_cast<const bf16x8*>(&ldsA[buf][(wr + f * 16 + fr) * LDSP + kq]);
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(&ldsB[buf][(wc + f * 16 + fr) * LDSP + kq]);
+    }
+    #pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+      #pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[fm], bfrag[fn],
+                                                              acc[fm][fn], 0, 0, 0);
+    if (kt + 1 < nk) {
+      __syncthreads();
+      stage(kt + 1, buf ^ 1);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: D[m][n], m = wr + fm*16 + (lane>>4)*4 + r, n = wc + fn*16 + fr
+  #pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m_loc = wr + fm * 16 + (lane >> 4) * 4 + r;
+      int m = bm * BM + m_loc;
+      if (m >= M) continue;
+      #pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        int n = bn * BN + wc + fn * 16 + fr;
+        if (n >= g.Cout) continue;
+        float v = acc[fm][fn][r];
+        if (HAS_BIAS) v += b2f(bias[n]);
+        Y[(int64_t)m * g.Cout + n] = f2b(v);
+      }
+    }
+  }
+}
+
+// ------------------------------------------------- weight repack (bwd-data)
+// W[co][kh][kw][ci] -> W2[ci][KH-1-kh][KW-1-kw][co]
+__global__ void weight_flip_kernel(const short* __restrict__ W, short* __restrict__ W2,
+                                   int Cout, int KH, int KW, int Cin) {
+  int64_t total = (int64_t)Cout * KH * KW * Cin;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int ci = i % Cin;
+  int64_t t = i / Cin;
+  int kw = t % KW; t /= KW;
+  int kh = t % KH; t /= KH;
+  int co = t;
+  W2[(((int64_t)ci * KH + (KH - 1 - kh)) * KW + (KW - 1 - kw)) * Cout + co] = W[i];
+}
+
+// ------------------------------------------------------------- bwd-weight
+// dW_acc[k=(kh,kw,ci)][n=co] += sum_m A[m][k] * dY[m][n]  (fp32 atomics)
+// block tile: 32 k-rows x 64 co, m-chunks of 32, split over blockIdx.z
+__global__ __launch_bounds__(256)
+void conv_wrw_kernel(const short* __restrict__ X, const short* __restrict__ dY,
+                     float* __restrict__ dWacc, ConvGeom g, int M, int m_slices) {
+  __shared__ short ldsA[32 * 40];      // [k'=32][m=32] padded
+  __shared__ short ldsB[64 * 40];      // [co=64][m=32] padded
+
+  const int kt = blockIdx.x;           // k-tile (32 rows of kpad)
+  const int nt = blockIdx.y;           // cout tile (64)
+  const int slice = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  int64_t m_per = (M + m_slices - 1) / m_slices;
+  int64_t m0 = (int64_t)slice * m_per;
+  int64_t m1 = min((int64_t)M, m0 + m_per);
+
+  f32x4 acc[2] = {};                   // wave covers [2x16 k'] x [16 co]
+  const int fr = lane & 15;
+  const int kq = (lane >> 4) * 8;
+
+  for (int64_t mc = m0; mc < m1; mc += 32) {
+    // ---- stage A^T: [k'][m] ; threads: 128 threads x 8 = 1024 elems
+    if (tid < 128) {
+      int m_loc = tid >> 2;            // 0..31
+      int kc8 = tid & 3;               // which 8-chunk of k'
+      int64_t m = mc + m_loc;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (m < m1) {
+        int b = (int)(m / (g.Ho * g.Wo));
+        int rem = (int)(m - (int64_t)b * g.Ho * g.Wo);
+        int ho = rem / g.Wo, wo = rem - (rem / g.Wo) * g.Wo;
+        int kc = kt * 4 + kc8;
+        int cell = kc / g.cin_chunks;
+        int ci0 = (kc - cell * g.cin_chunks) * 8;
+        int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
+        int hi = ho * g.stride - g.pad + kh;
+        int wi = wo * g.stride - g.pad + kw;
+        v = load_x8(X, g, b, hi, wi, ci0);
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        ldsA[(kc8 * 8 + j) * 40 + m_loc] = v[j];
+    } else {
+      // ---- stage dY^T: [co][m]; 128 threads x 16 = 2048 elems
+      int t = tid - 128;
+      int m_loc = t >> 2;              // 0..31
+      int cg = t & 3;                  // 16-chunk of cout
+      int64_t m = mc + m_loc;
+      short v[16] = {};
+      if (m < m1) {
+        const short* p = dY + m * g.Cout + nt * 64 + cg * 16;
+        #pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          int n = nt * 64 + cg * 16 + j;
+          v[j] = (n < g.Cout) ? p[j] : (short)0;
+        }
+      }
+      #pragma unroll
+      for (int j = 0; j < 16; ++j)
+        ldsB[(cg * 16 + j) * 40 + m_loc] = v[j];
+    }
+    __syncthreads();
+
+    // wave w handles co block [w*16, w*16+16), k' rows [0..32)
+    bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(&ldsB[(wave * 16 + fr) * 40 + kq]);
+    #pragma unroll
+    for (int fk = 0; fk < 2; ++fk) {
+      bf16x8 afrag = *reinterpret_cast<const bf16x8*>(&ldsA[(fk * 16 + fr) * 40 + kq]);
+      acc[fk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[fk], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: D[k'][co]: k' = fk*16 + (lane>>4)*4 + r, co = wave*16 + fr
+  #pragma unroll
+  for (int fk = 0; fk < 2; ++fk) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int kp = kt * 32 + fk * 16 + (lane >> 4) * 4 + r;
+      int n = nt * 64 + wave * 16 + fr;
+      if (kp < g.kpad && n < g.Cout)
+        atomicAdd(&dWacc[(int64_t)kp * g.Cout + n], acc[fk][r]);
+    }
+  }
+}
+
+// cast dW accumulator [kpad][Cout] back to torch layout [Cout][KH][KW][Cin]
+__global__ void wrw_cast_kernel(const float* __restrict__ dWacc, short* __restrict__ dW,
+                                ConvGeom g) {
+  int64_t total = (int64_t)g.Cout * g.KH * g.KW * g.Cin;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int ci = i % g.Cin;
+  int64_t t = i / g.Cin;
+  int cell = t % (g.KH * g.KW);
+  int co = t / (g.KH * g.KW);
+  int kp = cell * (g.cin_chunks * 8) + ci;
+  dW[i] = f2b(dWacc[(int64_t)kp * g.Cout + co]);
+}
+
+// column sum: dbias[n] = sum_m dY[m][n] (bf16 in, bf16 out via fp32)
+__global__ void colsum_kernel(const short* __restrict__ dY, float* __restrict__ partial,
+                              int64_t M, int C) {
+  // grid x: 64 partial blocks; each thread owns channels via linear stride
+  float s[8] = {};
+  const int c0 = (int)((((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8) % C);
+  int64_t total = M * C;
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t i = i0; i < total; i += stride) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(dY + i);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) s[j] += b2f(v[j]);
+  }
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = c0 + j; while (c >= C) c -= C;
+    atomicAdd(&partial[c], s[j]);
+  }
+}
+
+__global__ void cast_f32_bf16_kernel(const float* __restrict__ in, short* __restrict__ out,
+                                     int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = f2b(in[i]);
+}
+
+}  // namespace
+
+static ConvGeom make_geom(const torch::Tensor& x, const torch::Tensor& w,
+                          int64_t stride, int64_t pad) {
+  ConvGeom g;
+  g.B = x.size(0); g.Cin = x.size(1); g.H = x.size(2); g.Wd = x.size(3);
+  g.Cout = w.size(0); g.KH = w.size(2); g.KW = w.size(3);
+  g.stride = stride; g.pad = pad;
+  g.Ho = (g.H + 2 * g.pad - g.KH) / g.stride + 1;
+  g.Wo = (g.Wd + 2 * g.pad - g.KW) / g.stride + 1;
+  g.cin_chunks = (g.Cin + 7) / 8;
+  g.kpad = g.KH * g.KW * g.cin_chunks * 8;
+  // kpad must be a multiple of BK for the k-loop
+  g.kpad = (g.kpad + BK - 1) / BK * BK;
+  return g;
+}
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && w.scalar_type() == torch::kBFloat16,
+              "conv2d_fwd: bf16 only");
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto wc = w.contiguous(torch::MemoryFormat::ChannelsLast);
+  ConvGeom g = make_geom(xc, wc, stride, pad);
+  int M = g.B * g.Ho * g.Wo;
+  auto y = torch::empty({g.B, g.Cout, g.Ho, g.Wo},
+                        xc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  int grid_m = (M + BM - 1) / BM;
+  int grid_n = (g.Cout + BN - 1) / BN;
+  dim3 grid(grid_m * grid_n);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  bool has_bias = bias.defined() && bias.numel() > 0;
+  if (has_bias) {
+    auto bc = bias.contiguous();
+    TORCH_CHECK(bc.scalar_type() == torch::kBFloat16);
+    hipLaunchKernelGGL((conv_fwd_kernel<true>), grid, dim3(256), 0, stream,
+                       (const short*)xc.data_ptr(), (const short*)wc.data_ptr(),
+                       (const short*)bc.data_ptr(), (short*)y.data_ptr(), g, M, grid_m);
+  } else {
+    hipLaunchKernelGGL((conv_fwd_kernel<false>), grid, dim3(256), 0, stream,
+                       (const short*)xc.data_ptr(), (const short*)wc.data_ptr(),
+                       nullptr, (short*)y.data_ptr(), g, M, grid_m);
+  }
+  return y;
+}
+
+torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w,
+                              int64_t stride, int64_t pad,
+                              int64_t H, int64_t W_in) {
+  TORCH_CHECK(stride == 1, "conv2d_bwd_data: stride-1 only (caller falls back)");
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto wc = w.contiguous(torch::MemoryFormat::ChannelsLast);
+  int Cout = wc.size(0), Cin = wc.size(1), KH = wc.size(2), KW = wc.size(3);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  // repack: W2 [Cin][KH][KW][Cout] flipped
+  auto w2 = torch::empty({Cin, Cout, KH, KW},
+                         wc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)Cout * KH * KW * Cin;
+  hipLaunchKernelGGL(weight_flip_kernel, dim3((total + 255) / 256), dim3(256), 0, stream,
+                     (const short*)wc.data_ptr(), (short*)w2.data_ptr(),
+                     Cout, KH, KW, Cin);
+  // full-correlation: dX = conv(dY, W2, pad = KH-1-pad)
+  return conv2d_fwd(dyc, w2, torch::Tensor(), 1, KH - 1 - pad);
+}
+
+std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
+                                             int64_t stride, int64_t pad,
+                                             int64_t KH, int64_t KW, bool want_bias) {
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  int Cout = dyc.size(1);
+  ConvGeom g;
+  g.B = xc.size(0); g.Cin = xc.size(1); g.H = xc.size(2); g.Wd = xc.size(3);
+  g.Cout = Cout; g.KH = KH; g.KW = KW; g.stride = stride; g.pad = pad;
+  g.Ho = dyc.size(2); g.Wo = dyc.size(3);
+  g.cin_chunks = (g.Cin + 7) / 8;
+  g.kpad = (g.KH * g.KW * g.cin_chunks * 8 + BK - 1) / BK * BK;
+  int M = g.B * g.Ho * g.Wo;
+  auto stream = at::hip::getCurrentHIPStream().stream();
+
+  auto f32 = xc.options().dtype(torch::kFloat32);
+  auto dwacc = torch::zeros({g.kpad, Cout}, f32);
+  int grid_k = g.kpad / 32;
+  int grid_n = (Cout + 63) / 64;
+  // pick m-slices for ~>=512 blocks without over-splitting tiny workloads
+  int m_slices = 1;
+  while (grid_k * grid_n * m_slices < 512 && (int64_t)m_slices * 2 * 32 <= M)
+    m_slices *= 2;
+  dim3 grid(grid_k, grid_n, m_slices);
+  hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), 0, stream,
+                     (const short*)xc.data_ptr(), (const short*)dyc.data_ptr(),
+                     dwacc.data_ptr<float>(), g, M, m_slices);
+  auto dw = torch::empty({Cout, g.Cin, (int64_t)KH, (int64_t)KW},
+                         xc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)Cout * KH * KW * g.Cin;
+  hipLaunchKernelGGL(wrw_cast_kernel, dim3((total + 255) / 256), dim3(256), 0, stream,
+                     dwacc.data_ptr<float>(), (short*)dw.data_ptr(), g);
+
+  torch::Tensor dbias;
+  if (want_bias) {
+    auto part = torch::zeros({Cout}, f32);
+    if (Cout % 8 == 0) {
+      // stride-aligned colsum (channel set fixed per thread modulo C)
+      auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
+      int q = Cout / gcd(Cout, 2048);
+      int nb = ((64 + q - 1) / q) * q;
+      hipLaunchKernelGGL(colsum_kernel, dim3(nb), dim3(256), 0, stream,
+                         (const short*)dyc.data_ptr(), part.data_ptr<float>(),
+                         (int64_t)M, Cout);
+      dbias = torch::empty({Cout}, xc.options());
+      hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3((Cout + 255) / 256), dim3(256), 0,
+                         stream, part.data_ptr<float>(), (short*)dbias.data_ptr(), Cout);
+    } else {
+      dbias = dyc.sum(/*dim=*/{0, 2, 3}).to(xc.scalar_type());
+    }
+  }
+  return {dw, dbias};
+}

@@ -29,6 +29,13 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor out, torch::Tensor mean,
                                        torch::Tensor invstd, torch::Tensor gamma,
                                        bool training);
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pad);
+torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w, int64_t stride,
+                              int64_t pad, int64_t H, int64_t W_in);
+std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
+                                             int64_t stride, int64_t pad,
+                                             int64_t KH, int64_t KW, bool want_bias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_bcast", &scale_bcast, "out = x * s[b] (per-sample broadcast)");
@@ -47,5 +54,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("aug_pipeline", &aug_pipeline,
         "batched augmentation program executor (uint8 NHWC -> normalized bf16/f32)");
   m.def("bn_relu_fwd", &bn_relu_fwd);
+  m.def("conv2d_fwd", &conv2d_fwd, "MFMA implicit-GEMM NHWC bf16 conv forward");
+  m.def("conv2d_bwd_data", &conv2d_bwd_data);
+  m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("bn_relu_bwd", &bn_relu_bwd);
 }

@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "step.hip"),
     os.path.join(CSRC, "aug_kernels.hip"),
     os.path.join(CSRC, "bnrelu.hip"),
+    os.path.join(CSRC, "conv_mfma.hip"),
 ]
 
 setup(

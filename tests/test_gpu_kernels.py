@@ -321,8 +321,9 @@ def test_sgd_fused_step_mixed_matches_cpu():
     from fast_autoaugment_amd.parallel.flat import flatten_module
     from fast_autoaugment_amd.optim import FusedSGD
     torch.manual_seed(0)
+    # conv-only model: torch CPU BatchNorm cannot mix bf16 activations with
+    # fp32 buffers (the GPU path uses the fused HIP BN instead)
     mk = lambda: torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3, padding=1),
-                                     torch.nn.BatchNorm2d(8),
                                      torch.nn.Conv2d(8, 8, 1))
     m_gpu, m_cpu = mk().to(dev()), mk()
     m_cpu.load_state_dict(m_gpu.state_dict())
