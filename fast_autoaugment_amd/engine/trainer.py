@@ -83,11 +83,14 @@ def run_epoch(model, loader, loss_fn, optimizer, desc_default="", epoch=0,
             loss.backward()
             if hasattr(model, "finish_gradient_sync"):
                 model.finish_gradient_sync()
-            if wd > 0.0:
-                _apply_manual_wd(decay_params, wd)
-            grad_clip = conf["optimizer"].get("clip", 5.0)
-            if grad_clip > 0:
-                nn.utils.clip_grad_norm_(model.parameters(), grad_clip)
+            from ..optim import FusedSGD
+            if not isinstance(optimizer, FusedSGD):
+                # FusedSGD folds manual WD + global clip into its kernels
+                if wd > 0.0:
+                    _apply_manual_wd(decay_params, wd)
+                grad_clip = conf["optimizer"].get("clip", 5.0)
+                if grad_clip > 0:
+                    nn.utils.clip_grad_norm_(model.parameters(), grad_clip)
             optimizer.step()
             optimizer.zero_grad(set_to_none=False)
             if ema is not None:
@@ -154,7 +157,14 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
         multinode=(local_rank >= 0), rank=rank, world_size=world_size,
         device=device, out_dtype=out_dtype)
 
-    model = get_model(conf["model"], nc, local_rank=local_rank, device=device)
+    # Fast path: pure-bf16 flat weights + fused SGD + MFMA convs on GPU.
+    use_fast = (use_cuda and conf["optimizer"]["type"] == "sgd"
+                and autocast_dtype == torch.bfloat16
+                and conf.get_value("pure_bf16", True))
+    work_dtype = torch.bfloat16 if use_fast else None
+
+    model = get_model(conf["model"], nc, local_rank=local_rank, device=device,
+                      work_dtype=work_dtype)
     model_ema = get_model(conf["model"], nc, local_rank=-1, device=device)
     model_ema.eval()
 
@@ -162,7 +172,21 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
     if conf.get_value("mixup", 0.0) > 0.0:
         criterion = CrossEntropyMixUpLabelSmooth(nc, conf.get_value("lb_smooth", 0))
 
-    optimizer = build_optimizer(conf["optimizer"], model.parameters(), conf["lr"])
+    if use_fast:
+        from ..ops.conv import patch_convs
+        from ..optim import FusedSGD
+        from ..parallel.flat import flatten_module
+        flat = model.flat if hasattr(model, "flat") else \
+            flatten_module(model, work_dtype=torch.bfloat16)
+        patch_convs(model)
+        optimizer = FusedSGD(flat, lr=conf["lr"],
+                             momentum=conf["optimizer"].get("momentum", 0.9),
+                             nesterov=conf["optimizer"].get("nesterov", True),
+                             weight_decay=conf["optimizer"].get("decay", 0.0),
+                             grad_clip=conf["optimizer"].get("clip", 5.0))
+        autocast_dtype = None      # the model computes natively in bf16
+    else:
+        optimizer = build_optimizer(conf["optimizer"], model.parameters(), conf["lr"])
     scheduler = build_scheduler(conf.conf, optimizer, conf["lr"])
 
     writers = [get_summary_writer(f"./logs/{tag}/{x}", bool(tag) and is_master)
@@ -184,7 +208,12 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
             sd = data[key]
             raw = model.module if hasattr(model, "module") else model
             raw.load_state_dict({k.replace("module.", ""): v for k, v in sd.items()})
-            optimizer.load_state_dict(data["optimizer"])
+            if use_fast:
+                flat.flat_master.copy_(flat.flat_param.float())
+            try:
+                optimizer.load_state_dict(data["optimizer"])
+            except Exception as e:
+                logger.warning("optimizer state not loaded (%s); fresh momentum", e)
             if data["epoch"] < max_epoch:
                 epoch_start = data["epoch"]
             else:
@@ -307,6 +336,10 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
                 if is_master and save_path:
                     raw = model.module if hasattr(model, "module") else model
                     logger.info("save model@%d to %s", epoch, save_path)
+                    # .pth layout compatible with the reference (train.py:307-317);
+                    # bf16 working weights are stored as fp32 for interchange
+                    sd_out = {k: (v.float() if v.dtype == torch.bfloat16 else v)
+                              for k, v in raw.state_dict().items()}
                     torch.save({
                         "epoch": epoch,
                         "log": {
@@ -315,7 +348,7 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
                             "test": rs["test"].get_dict(),
                         },
                         "optimizer": optimizer.state_dict(),
-                        "model": raw.state_dict(),
+                        "model": sd_out,
                         "ema": ema.state_dict() if ema is not None else None,
                     }, save_path)
 

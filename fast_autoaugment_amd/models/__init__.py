@@ -70,7 +70,8 @@ def _tf_style_init(model: nn.Module) -> None:
 
 
 def get_model(conf, num_class: int = 10, local_rank: int = -1,
-              device: str = "cuda", channels_last: bool = True) -> nn.Module:
+              device: str = "cuda", channels_last: bool = True,
+              work_dtype=None) -> nn.Module:
     model = build_model(conf, num_class)
     name = conf["type"]
     if "efficientnet" in name and local_rank >= 0:
@@ -82,7 +83,7 @@ def get_model(conf, num_class: int = 10, local_rank: int = -1,
             model = model.to(memory_format=torch.channels_last)
     if local_rank >= 0:
         from ..parallel.ddp import FlatDDP
-        model = FlatDDP(model)
+        model = FlatDDP(model, work_dtype=work_dtype)
     return model
 
 
