@@ -167,6 +167,10 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
                       work_dtype=work_dtype)
     model_ema = get_model(conf["model"], nc, local_rank=-1, device=device)
     model_ema.eval()
+    if use_fast:
+        # EMA evaluation model follows the bf16 working dtype (buffers stay fp32)
+        for p in model_ema.parameters():
+            p.data = p.data.to(torch.bfloat16)
 
     criterion_ce = criterion = CrossEntropyLabelSmooth(nc, conf.get_value("lb_smooth", 0))
     if conf.get_value("mixup", 0.0) > 0.0:
