@@ -119,10 +119,13 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
       int kc = kt * 4 + a_kc;                    // global 8-chunk index
       int cell = kc / g.cin_chunks;
       int ci0 = (kc - cell * g.cin_chunks) * 8;
-      int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
-      int hi = xho * g.stride - g.pad + kh;
-      int wi = xwo * g.stride - g.pad + kw;
-      bf16x8 v = load_x8(X, g, xb, hi, wi, ci0);
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (cell < g.KH * g.KW) {                  // guard kpad round-up cells
+        int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
+        int hi = xho * g.stride - g.pad + kh;
+        int wi = xwo * g.stride - g.pad + kw;
+        v = load_x8(X, g, xb, hi, wi, ci0);
+      }
       short* dst = &ldsA[buf][a_row * LDSP + a_kc * 8];
       *reinterpret_cast<bf16x8*>(dst) = v;
     }
@@ -132,7 +135,7 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
       int cell = kc / g.cin_chunks;
       int ci0 = (kc - cell * g.cin_chunks) * 8;
       bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (n_g < g.Cout && ci0 < g.Cin) {
+      if (n_g < g.Cout && ci0 < g.Cin && cell < g.KH * g.KW) {
         const short* p = Wt + ((int64_t)n_g * g.KH * g.KW + cell) * g.Cin + ci0;
         if (ci0 + 8 <= g.Cin) {
           v = *reinterpret_cast<const bf16x8*>(p);
@@ -248,10 +251,12 @@ void conv_wrw_kernel(const short* __restrict__ X, const short* __restrict__ dY,
         int kc = kt * 4 + kc8;
         int cell = kc / g.cin_chunks;
         int ci0 = (kc - cell * g.cin_chunks) * 8;
-        int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
-        int hi = ho * g.stride - g.pad + kh;
-        int wi = wo * g.stride - g.pad + kw;
-        v = load_x8(X, g, b, hi, wi, ci0);
+        if (cell < g.KH * g.KW) {
+          int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
+          int hi = ho * g.stride - g.pad + kh;
+          int wi = wo * g.stride - g.pad + kw;
+          v = load_x8(X, g, b, hi, wi, ci0);
+        }
       }
       #pragma unroll
       for (int j = 0; j < 8; ++j)
