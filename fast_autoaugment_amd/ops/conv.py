@@ -18,6 +18,14 @@ import torch.nn.functional as F
 from . import ext
 
 
+# Measured dispatch rule (profiles/conv_bench_r01.txt, MI355X b128):
+# the MFMA fwd/bwd-data kernels beat MIOpen for C<=128 (all WRN-40-2
+# shapes, 1.2-4x); MIOpen keeps larger channels and the wrw direction
+# until the wrw v2 staging lands.
+_FAA_MAX_CH = 128
+_FAA_WRW = False   # flip when conv_wrw v2 beats igemm_wrw
+
+
 class FaaConv2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
@@ -44,11 +52,20 @@ class FaaConv2dFn(torch.autograd.Function):
                                                 stride=ctx.stride,
                                                 padding=ctx.padding)
         if ctx.needs_input_grad[1] or (ctx.has_bias and ctx.needs_input_grad[2]):
-            dw, db = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
-                                         weight.size(2), weight.size(3),
-                                         ctx.has_bias)
-            if ctx.has_bias and ctx.needs_input_grad[2]:
-                dbias = db
+            if _FAA_WRW:
+                dw, db = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
+                                             weight.size(2), weight.size(3),
+                                             ctx.has_bias)
+                if ctx.has_bias and ctx.needs_input_grad[2]:
+                    dbias = db
+            else:
+                if ctx.needs_input_grad[1]:
+                    dw = torch.nn.grad.conv2d_weight(
+                        x, list(weight.shape), dy, stride=ctx.stride,
+                        padding=ctx.padding).contiguous(
+                            memory_format=torch.channels_last)
+                if ctx.has_bias and ctx.needs_input_grad[2]:
+                    dbias = dy.sum(dim=(0, 2, 3))
         return dx, dw, dbias, None, None
 
 
@@ -61,7 +78,8 @@ def _eligible(m: torch.nn.Conv2d) -> bool:
     return (k[0] == k[1] and k[0] in (1, 3)
             and m.padding[0] == m.padding[1] and m.padding[0] == k[0] // 2
             and m.stride[0] == m.stride[1] and m.stride[0] in (1, 2)
-            and m.dilation == (1, 1) and m.groups == 1)
+            and m.dilation == (1, 1) and m.groups == 1
+            and m.in_channels <= _FAA_MAX_CH and m.out_channels <= _FAA_MAX_CH)
 
 
 def _faa_forward(self, x):
