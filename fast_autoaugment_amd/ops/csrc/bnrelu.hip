@@ -377,10 +377,13 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
 
 // partial-sum block count: the grid stride (nblocks*256*8) must be a
 // multiple of C so each thread's channel octet is loop-invariant.
-static int bn_nblocks(int C) {
+static int bn_nblocks(int C, int64_t total) {
   auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
   int q = C / gcd(C, 2048);
-  return ((64 + q - 1) / q) * q;
+  // target ~8 vector iterations per thread, clamp [64, 512] then align to q
+  int64_t want = total / 8 / 256 / 8;
+  int base = (int)std::min<int64_t>(std::max<int64_t>(want, 64), 512);
+  return ((base + q - 1) / q) * q;
 }
 
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
@@ -404,7 +407,7 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   if (training) {
     bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
     if (vec) {
-      int nb = bn_nblocks(C);
+      int nb = bn_nblocks(C, total);
       auto scratch = torch::empty({nb, 2 * C}, f32);
       DISPATCH_FB(xc.scalar_type(), "bn_reduce", [&] {
         hipLaunchKernelGGL((bn_reduce_vec_kernel<scalar_t>), dim3(nb), dim3(256), 0,
@@ -470,7 +473,7 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
 
   bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
   if (vec) {
-    int nb = bn_nblocks(C);
+    int nb = bn_nblocks(C, total);
     auto scratch = torch::empty({nb, 2 * C}, f32);
     DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce", [&] {
       hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<scalar_t>), dim3(nb), dim3(256), 0,

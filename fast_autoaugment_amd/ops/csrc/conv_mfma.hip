@@ -67,16 +67,15 @@ __device__ __forceinline__ bf16x8 load_x8(const short* __restrict__ X,
 
 // ------------------------------------------------------------------- fwd
 
-template <bool HAS_BIAS>
+template <bool HAS_BIAS, int BNT>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
                      const short* __restrict__ bias, short* __restrict__ Y,
                      ConvGeom g, int M, int grid_m) {
+  // BNT=64: 4 waves as 2x2 (32x32 each); BNT=32: 4 waves as 4x1 (16x32 each)
   __shared__ short ldsA[2][BM * LDSP];
-  __shared__ short ldsB[2][BN * LDSP];
+  __shared__ short ldsB[2][BNT * LDSP];
 
-  // XCD-aware swizzle: consecutive logical tiles share B-panels; keep them
-  // on one XCD's L2 (guide T1). Bijective for any grid size.
   int nwg = gridDim.x;
   int wg = blockIdx.x;
   {
@@ -90,10 +89,10 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wr = (wave >> 1) * 32;     // wave row offset in tile
-  const int wc = (wave & 1) * 32;      // wave col offset
+  const int wr = (BNT == 64) ? (wave >> 1) * 32 : wave * 16;
+  const int wc = (BNT == 64) ? (wave & 1) * 32 : 0;
+  constexpr int MFRAG = (BNT == 64) ? 2 : 1;
 
-  // per-thread A staging coords: row tid/4, k-chunk tid%4
   const int a_row = tid >> 2;
   const int a_kc = (tid & 3);
   const int m_g = bm * BM + a_row;
@@ -104,83 +103,88 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
     xho = rem / g.Wo;
     xwo = rem - xho * g.Wo;
   } else {
-    xb = g.B;  // out of range flag
+    xb = g.B;
   }
-  // B staging: row tid/4 (cout), k-chunk tid%4
-  const int b_row = tid >> 2;
-  const int n_g = bn * BN + b_row;
+  // B staging: BNT rows x 32 k = BNT*4 chunk-loads; with 256 threads each
+  // thread loads ceil(BNT*4/256) chunks (1 for BNT=64, every other thread
+  // idle for BNT=32)
+  const int b_row = (tid >> 2) % BNT;
+  const bool b_act = (tid >> 2) < BNT;
+  const int n_g = bn * BNT + b_row;
 
   const int nk = g.kpad / BK;
-  f32x4 acc[2][2] = {};
+  f32x4 acc[MFRAG][2] = {};
+  const int CELLS = g.KH * g.KW;
 
-  auto stage = [&](int kt, int buf) {
-    // A: 8 channels for (cell, ci) chunk
-    {
-      int kc = kt * 4 + a_kc;                    // global 8-chunk index
-      int cell = kc / g.cin_chunks;
-      int ci0 = (kc - cell * g.cin_chunks) * 8;
-      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (cell < g.KH * g.KW) {                  // guard kpad round-up cells
-        int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
-        int hi = xho * g.stride - g.pad + kh;
-        int wi = xwo * g.stride - g.pad + kw;
-        v = load_x8(X, g, xb, hi, wi, ci0);
-      }
-      short* dst = &ldsA[buf][a_row * LDSP + a_kc * 8];
-      *reinterpret_cast<bf16x8*>(dst) = v;
+  auto load_a = [&](int kt) -> bf16x8 {
+    int kc = kt * 4 + a_kc;
+    int cell = kc / g.cin_chunks;
+    int ci0 = (kc - cell * g.cin_chunks) * 8;
+    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (cell < CELLS) {
+      int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
+      int hi = xho * g.stride - g.pad + kh;
+      int wi = xwo * g.stride - g.pad + kw;
+      v = load_x8(X, g, xb, hi, wi, ci0);
     }
-    // B: weights [n][k] with k = cell*Cin + ci contiguous
-    {
-      int kc = kt * 4 + a_kc;
-      int cell = kc / g.cin_chunks;
-      int ci0 = (kc - cell * g.cin_chunks) * 8;
-      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (n_g < g.Cout && ci0 < g.Cin && cell < g.KH * g.KW) {
-        const short* p = Wt + ((int64_t)n_g * g.KH * g.KW + cell) * g.Cin + ci0;
-        if (ci0 + 8 <= g.Cin) {
-          v = *reinterpret_cast<const bf16x8*>(p);
-        } else {
-          for (int j = 0; j < g.Cin - ci0; ++j) v[j] = p[j];
-        }
+    return v;
+  };
+  auto load_b = [&](int kt) -> bf16x8 {
+    int kc = kt * 4 + a_kc;
+    int cell = kc / g.cin_chunks;
+    int ci0 = (kc - cell * g.cin_chunks) * 8;
+    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (b_act && n_g < g.Cout && ci0 < g.Cin && cell < CELLS) {
+      const short* p = Wt + ((int64_t)n_g * CELLS + cell) * g.Cin + ci0;
+      if (ci0 + 8 <= g.Cin) {
+        v = *reinterpret_cast<const bf16x8*>(p);
+      } else {
+        for (int j = 0; j < g.Cin - ci0; ++j) v[j] = p[j];
       }
-      short* dst = &ldsB[buf][b_row * LDSP + a_kc * 8];
-      *reinterpret_cast<bf16x8*>(dst) = v;
     }
+    return v;
+  };
+  auto write_lds = [&](int buf, bf16x8 va, bf16x8 vb) {
+    *reinterpret_cast<bf16x8*>(&ldsA[buf][a_row * LDSP + a_kc * 8]) = va;
+    if (b_act)
+      *reinterpret_cast<bf16x8*>(&ldsB[buf][b_row * LDSP + a_kc * 8]) = vb;
   };
 
-  stage(0, 0);
+  write_lds(0, load_a(0), load_b(0));
   __syncthreads();
 
-  const int fr = lane & 15;            // fragment row/col within 16
-  const int kq = (lane >> 4) * 8;      // 8-elem k offset
+  const int fr = lane & 15;
+  const int kq = (lane >> 4) * 8;
 
   for (int kt = 0; kt < nk; ++kt) {
     int buf = kt & 1;
-    if (kt + 1 < nk) {
-      // prefetch next tile into the other LDS buffer AFTER compute barrier
+    bf16x8 na = {}, nb = {};
+    if (kt + 1 < nk) {      // issue next-tile global loads BEFORE the MFMAs
+      na = load_a(kt + 1);
+      nb = load_b(kt + 1);
     }
-    bf16x8 afrag[2], bfrag[2];
+    bf16x8 afrag[MFRAG], bfrag[2];
     #pragma unroll
-    for (int f = 0; f < 2; ++f) {
+    for (int f = 0; f < MFRAG; ++f)
       afrag[f] = *reinterpret_cast<const bf16x8*>(&ldsA[buf][(wr + f * 16 + fr) * LDSP + kq]);
-      bfrag[f] = *reinterpret_cast<const bf16x8*>(&ldsB[buf][(wc + f * 16 + fr) * LDSP + kq]);
-    }
     #pragma unroll
-    for (int fm = 0; fm < 2; ++fm)
+    for (int f = 0; f < 2; ++f)
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(&ldsB[buf][(wc + f * 16 + fr) * LDSP + kq]);
+    #pragma unroll
+    for (int fm = 0; fm < MFRAG; ++fm)
       #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[fm], bfrag[fn],
                                                               acc[fm][fn], 0, 0, 0);
     if (kt + 1 < nk) {
       __syncthreads();
-      stage(kt + 1, buf ^ 1);
+      write_lds(buf ^ 1, na, nb);
     }
     __syncthreads();
   }
 
-  // epilogue: D[m][n], m = wr + fm*16 + (lane>>4)*4 + r, n = wc + fn*16 + fr
   #pragma unroll
-  for (int fm = 0; fm < 2; ++fm) {
+  for (int fm = 0; fm < MFRAG; ++fm) {
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int m_loc = wr + fm * 16 + (lane >> 4) * 4 + r;
@@ -188,7 +192,7 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
       if (m >= M) continue;
       #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
-        int n = bn * BN + wc + fn * 16 + fr;
+        int n = bn * BNT + wc + fn * 16 + fr;
         if (n >= g.Cout) continue;
         float v = acc[fm][fn][r];
         if (HAS_BIAS) v += b2f(bias[n]);
@@ -374,21 +378,25 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   auto y = torch::empty({g.B, g.Cout, g.Ho, g.Wo},
                         xc.options().memory_format(torch::MemoryFormat::ChannelsLast));
   int grid_m = (M + BM - 1) / BM;
-  int grid_n = (g.Cout + BN - 1) / BN;
+  int bnt = (g.Cout <= 32) ? 32 : 64;
+  int grid_n = (g.Cout + bnt - 1) / bnt;
   dim3 grid(grid_m * grid_n);
   auto stream = at::hip::getCurrentHIPStream().stream();
   bool has_bias = bias.defined() && bias.numel() > 0;
+  const short* bptr = nullptr;
+  torch::Tensor bc;
   if (has_bias) {
-    auto bc = bias.contiguous();
+    bc = bias.contiguous();
     TORCH_CHECK(bc.scalar_type() == torch::kBFloat16);
-    hipLaunchKernelGGL((conv_fwd_kernel<true>), grid, dim3(256), 0, stream,
-                       (const short*)xc.data_ptr(), (const short*)wc.data_ptr(),
-                       (const short*)bc.data_ptr(), (short*)y.data_ptr(), g, M, grid_m);
-  } else {
-    hipLaunchKernelGGL((conv_fwd_kernel<false>), grid, dim3(256), 0, stream,
-                       (const short*)xc.data_ptr(), (const short*)wc.data_ptr(),
-                       nullptr, (short*)y.data_ptr(), g, M, grid_m);
+    bptr = (const short*)bc.data_ptr();
   }
+  #define CF_LAUNCH(HB, BNT)                                                    \
+    hipLaunchKernelGGL((conv_fwd_kernel<HB, BNT>), grid, dim3(256), 0, stream,  \
+                       (const short*)xc.data_ptr(), (const short*)wc.data_ptr(),\
+                       bptr, (short*)y.data_ptr(), g, M, grid_m)
+  if (has_bias) { if (bnt == 32) CF_LAUNCH(true, 32); else CF_LAUNCH(true, 64); }
+  else { if (bnt == 32) CF_LAUNCH(false, 32); else CF_LAUNCH(false, 64); }
+  #undef CF_LAUNCH
   return y;
 }
 
