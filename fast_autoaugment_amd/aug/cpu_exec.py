@@ -232,3 +232,76 @@ def run_pipeline_cpu(batch: np.ndarray, prog: np.ndarray, post: np.ndarray,
                      mean: np.ndarray, std: np.ndarray) -> np.ndarray:
     """Full train-time pipeline on CPU: programs -> post stage. float32 NHWC out."""
     return apply_post_batch(apply_program_batch(batch, prog), post, mean, std)
+
+
+# --------------------------------------------------- imagenet pipeline (CPU)
+
+def _cubic_pil(x: np.ndarray) -> np.ndarray:
+    """PIL bicubic kernel, a=-0.5."""
+    x = np.abs(x)
+    out = np.zeros_like(x)
+    m1 = x < 1.0
+    m2 = (x >= 1.0) & (x < 2.0)
+    out[m1] = ((1.5 * x[m1] - 2.5) * x[m1]) * x[m1] + 1.0
+    out[m2] = (((-0.5 * x[m2]) + 2.5) * x[m2] - 4.0) * x[m2] + 2.0
+    return out
+
+
+def resize_box_bicubic(img: np.ndarray, box, oh: int, ow: int, flip: bool) -> np.ndarray:
+    """Antialiased bicubic crop-resize (single-pass fp32; mirrors the HIP
+    kernel op_resize_box — PIL-style support widening on downscale)."""
+    H, W = img.shape[:2]
+    bx0, by0, bw, bh = box
+    ix0, iy0, ibw, ibh = int(bx0), int(by0), int(bw), int(bh)
+    sx, sy = bw / ow, bh / oh
+    ssx, ssy = (1.0 / sx if sx > 1 else 1.0), (1.0 / sy if sy > 1 else 1.0)
+    supx, supy = 2.0 * max(sx, 1.0), 2.0 * max(sy, 1.0)
+
+    out = np.empty((oh, ow, 3), np.float32)
+    f = img.astype(np.float32)
+    ys = np.arange(oh)
+    cys = by0 + (ys + 0.5) * sy
+    for oy in range(oh):
+        cy = np.float32(cys[oy])
+        ymin = max(int(cy - supy + 0.5), iy0)
+        ymax = min(int(cy + supy + 0.5), iy0 + ibh)
+        wy = _cubic_pil(((np.arange(ymin, ymax) + 0.5 - cy) * ssy).astype(np.float32))
+        for ox in range(ow):
+            sxp = (ow - 1 - ox) if flip else ox
+            cx = np.float32(bx0 + (sxp + 0.5) * sx)
+            xmin = max(int(cx - supx + 0.5), ix0)
+            xmax = min(int(cx + supx + 0.5), ix0 + ibw)
+            wx = _cubic_pil(((np.arange(xmin, xmax) + 0.5 - cx) * ssx).astype(np.float32))
+            wmat = wy[:, None] * wx[None, :]
+            wsum = wmat.sum()
+            patch = f[ymin:ymax, xmin:xmax]
+            out[oy, ox] = (wmat[:, :, None] * patch).sum(axis=(0, 1)) / (wsum if wsum != 0 else 1.0)
+    return np.clip(np.round(out), 0, 255).astype(np.uint8)
+
+
+def apply_post_imagenet_batch(batch: np.ndarray, post: np.ndarray,
+                              mean: np.ndarray, std: np.ndarray,
+                              oh: int, ow: int) -> np.ndarray:
+    """CPU reference for the imagenet post stage (see aug/imagenet.py layout)."""
+    from .ops import OpCode
+    B = batch.shape[0]
+    out = np.empty((B, oh, ow, 3), np.float32)
+    for b in range(B):
+        pp = post[b]
+        img = resize_box_bicubic(batch[b], pp[1:5], oh, ow, pp[5] > 0.5)
+        for s in range(3):
+            code, f = int(pp[9 + s * 2]), float(pp[10 + s * 2])
+            if code == OpCode.BRIGHTNESS:
+                img = brightness(img, f)
+            elif code == OpCode.CONTRAST:
+                img = contrast(img, f)
+            elif code == OpCode.COLOR:
+                img = color(img, f)
+        m = mean.reshape(1, 1, 3) - pp[6:9].reshape(1, 1, 3)
+        out[b] = (img.astype(np.float32) / 255.0 - m) / std.reshape(1, 1, 3)
+    return out
+
+
+def run_pipeline_imagenet_cpu(batch, prog, post, mean, std, oh, ow):
+    return apply_post_imagenet_batch(apply_program_batch(batch, prog), post,
+                                     mean, std, oh, ow)

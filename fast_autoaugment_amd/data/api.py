@@ -67,6 +67,16 @@ def get_dataloaders(dataset: str, batch: int, dataroot: str, split: float = 0.15
     policy = resolve_aug(conf["aug"]) if "aug" in conf.conf else []
     cutout = int(conf.get_value("cutout", 0))
     mean, std = dataset_stats(dataset)
+    # ImageNet: EffNet-style crop-resize pipeline sized per model
+    # (reference data.py:49-80; EfficientNet input size from the model table)
+    imagenet_size = 0
+    if "imagenet" in dataset:
+        mt = conf["model"]["type"] if "model" in conf.conf else ""
+        if "efficientnet" in mt:
+            from ..models.efficientnet import EfficientNet
+            imagenet_size = EfficientNet.get_image_size(mt)
+        else:
+            imagenet_size = 224
 
     # CV fold split (reference data.py:192-203)
     if split > 0.0:
@@ -81,7 +91,8 @@ def get_dataloaders(dataset: str, batch: int, dataroot: str, split: float = 0.15
     train_store = _get_store(store_key + ("train",), tr_imgs, tr_labels, device)
     test_store = _get_store(store_key + ("test",), te_imgs, te_labels, device)
 
-    common = dict(mean=mean, std=std, out_dtype=out_dtype, seed=seed)
+    common = dict(mean=mean, std=std, out_dtype=out_dtype, seed=seed,
+                  imagenet_size=imagenet_size)
     trainloader = AugLoader(train_store, batch, policy, train=True, cutout=cutout,
                             indices=train_idx, rank=rank if multinode else 0,
                             world_size=world_size if multinode else 1, **common)

@@ -57,7 +57,8 @@ class AugLoader:
                  indices: Optional[np.ndarray] = None,
                  shuffle: Optional[bool] = None, drop_last: Optional[bool] = None,
                  rank: int = 0, world_size: int = 1, seed: int = 0,
-                 out_dtype: torch.dtype = torch.float32, prefetch: int = 2):
+                 out_dtype: torch.dtype = torch.float32, prefetch: int = 2,
+                 imagenet_size: int = 0):
         self.store = store
         self.batch = batch
         self.policy = policy or []
@@ -75,6 +76,7 @@ class AugLoader:
         self.epoch = 0
         self.out_dtype = out_dtype
         self.prefetch = prefetch
+        self.imagenet_size = imagenet_size   # >0: EffNet crop-resize pipeline
         self._mean_t = None
         self._std_t = None
 
@@ -102,6 +104,15 @@ class AugLoader:
     def _gen_host(self, idx: np.ndarray, rng: np.random.Generator, b: int):
         sel = idx[b * self.batch:(b + 1) * self.batch]
         H, W = self.store.hw
+        if self.imagenet_size > 0:
+            from ..aug.imagenet import compile_post_imagenet
+            if self.train:
+                prog = aug_ops.compile_program(self.policy, len(sel), W, H, rng)
+            else:
+                prog = np.zeros((len(sel), aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH), np.float32)
+            post = compile_post_imagenet(len(sel), W, H, rng, self.imagenet_size,
+                                         train=self.train)
+            return sel, prog, post
         if self.train:
             prog = aug_ops.compile_program(self.policy, len(sel), W, H, rng)
             post = aug_ops.compile_post(len(sel), W, H, rng, pad=self.pad,
@@ -143,7 +154,12 @@ class AugLoader:
         if self.store.device.type == "cuda":
             return self._make_batch_gpu(sel, prog, post)
         imgs = self.store.images_np[sel]
-        out = cpu_exec.run_pipeline_cpu(imgs, prog, post, self.mean, self.std)
+        if self.imagenet_size > 0:
+            out = cpu_exec.run_pipeline_imagenet_cpu(imgs, prog, post, self.mean,
+                                                     self.std, self.imagenet_size,
+                                                     self.imagenet_size)
+        else:
+            out = cpu_exec.run_pipeline_cpu(imgs, prog, post, self.mean, self.std)
         data = torch.from_numpy(out).permute(0, 3, 1, 2).contiguous()
         label = self.store.labels[torch.from_numpy(np.ascontiguousarray(sel))]
         return data.to(self.out_dtype), label
@@ -158,8 +174,14 @@ class AugLoader:
         post_t = torch.from_numpy(post).to(dev, non_blocking=True)
         from ..ops import ext
         C = ext()
-        out = C.aug_pipeline(self.store.images, sel_t, prog_t, post_t,
-                             self._mean_t, self._std_t,
-                             self.out_dtype == torch.bfloat16)
+        if self.imagenet_size > 0:
+            out = C.aug_pipeline_imagenet(self.store.images, sel_t, prog_t, post_t,
+                                          self._mean_t, self._std_t,
+                                          self.imagenet_size, self.imagenet_size,
+                                          self.out_dtype == torch.bfloat16)
+        else:
+            out = C.aug_pipeline(self.store.images, sel_t, prog_t, post_t,
+                                 self._mean_t, self._std_t,
+                                 self.out_dtype == torch.bfloat16)
         label = self.store.labels[sel_t]
         return out, label

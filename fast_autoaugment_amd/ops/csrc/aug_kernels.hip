@@ -195,6 +195,97 @@ __device__ void op_cutout(const uint32_t* src, uint32_t* dst, int W, int H,
   }
 }
 
+// run one image's op program over ping-pong buffers; returns final src
+__device__ uint32_t* run_program(uint32_t* src, uint32_t* dst, int W, int H,
+                                 const float* bp, uint32_t* hist, uint8_t* lut,
+                                 uint32_t* scalar_acc) {
+  const int n = W * H;
+  for (int s = 0; s < PROG_SLOTS; ++s) {
+    const float* slot = bp + s * PROG_WIDTH;
+    int code = (int)slot[0];
+    const float* p = slot + 1;
+    if (code == OP_NOP) continue;
+    switch (code) {
+      case OP_AFFINE:
+        op_affine(src, dst, W, H, p);
+        break;
+      case OP_AUTOCONTRAST: {
+        build_histogram(src, n, hist);
+        if (threadIdx.x < 3) autocontrast_lut(hist + threadIdx.x * 256, lut + threadIdx.x * 256);
+        __syncthreads();
+        op_pointwise_lut(src, dst, n, lut, lut + 256, lut + 512);
+        break;
+      }
+      case OP_EQUALIZE: {
+        build_histogram(src, n, hist);
+        if (threadIdx.x < 3) equalize_lut(hist + threadIdx.x * 256, n, lut + threadIdx.x * 256);
+        __syncthreads();
+        op_pointwise_lut(src, dst, n, lut, lut + 256, lut + 512);
+        break;
+      }
+      case OP_INVERT:
+        for (int i = threadIdx.x; i < n; i += blockDim.x) {
+          uint32_t px = src[i];
+          dst[i] = pack_rgb(255 - ch_r(px), 255 - ch_g(px), 255 - ch_b(px));
+        }
+        break;
+      case OP_FLIP:
+        for (int i = threadIdx.x; i < n; i += blockDim.x) {
+          int x = i % W, y = i / W;
+          dst[i] = src[y * W + (W - 1 - x)];
+        }
+        break;
+      case OP_SOLARIZE: {
+        float th = p[0];
+        for (int i = threadIdx.x; i < n; i += blockDim.x) {
+          uint32_t px = src[i];
+          int r = ch_r(px), g = ch_g(px), bb = ch_b(px);
+          dst[i] = pack_rgb((float)r >= th ? 255 - r : r,
+                            (float)g >= th ? 255 - g : g,
+                            (float)bb >= th ? 255 - bb : bb);
+        }
+        break;
+      }
+      case OP_POSTERIZE: {
+        int bits = (int)p[0];
+        uint32_t m8 = bits >= 8 ? 0xFFu : (uint32_t)(0xFF & ~((1 << (8 - bits)) - 1));
+        uint32_t mask = m8 | (m8 << 8) | (m8 << 16);
+        for (int i = threadIdx.x; i < n; i += blockDim.x)
+          dst[i] = src[i] & mask;
+        break;
+      }
+      case OP_CONTRAST: {
+        if (threadIdx.x == 0) scalar_acc[0] = 0;
+        __syncthreads();
+        uint32_t part = 0;
+        for (int i = threadIdx.x; i < n; i += blockDim.x) part += (uint32_t)lum(src[i]);
+        atomicAdd(&scalar_acc[0], part);
+        __syncthreads();
+        int meanv = (int)((double)scalar_acc[0] / n + 0.5);
+        op_blend_const(src, dst, n, meanv, meanv, meanv, p[0]);
+        break;
+      }
+      case OP_COLOR:
+        op_color(src, dst, n, p[0]);
+        break;
+      case OP_BRIGHTNESS:
+        op_blend_const(src, dst, n, 0, 0, 0, p[0]);
+        break;
+      case OP_SHARPNESS:
+        op_sharpness(src, dst, W, H, p[0]);
+        break;
+      case OP_CUTOUT:
+        op_cutout(src, dst, W, H, p);
+        break;
+      default:
+        for (int i = threadIdx.x; i < n; i += blockDim.x) dst[i] = src[i];
+    }
+    __syncthreads();
+    uint32_t* t = src; src = dst; dst = t;
+  }
+  return src;
+}
+
 // ------------------------------------------------------------- main kernel
 
 template <typename OutT, bool IN_LDS>
@@ -234,92 +325,8 @@ __global__ void aug_pipeline_kernel(
     uint32_t* src = bufA;
     uint32_t* dst = bufB;
     const float* bp = prog + (int64_t)b * PROG_SLOTS * PROG_WIDTH;
-
-    // ---- program ops -------------------------------------------------------
-    for (int s = 0; s < PROG_SLOTS; ++s) {
-      const float* slot = bp + s * PROG_WIDTH;
-      int code = (int)slot[0];
-      const float* p = slot + 1;
-      if (code == OP_NOP) continue;
-      switch (code) {
-        case OP_AFFINE:
-          op_affine(src, dst, W, H, p);
-          break;
-        case OP_AUTOCONTRAST: {
-          build_histogram(src, n, hist);
-          if (threadIdx.x < 3) autocontrast_lut(hist + threadIdx.x * 256, lut + threadIdx.x * 256);
-          __syncthreads();
-          op_pointwise_lut(src, dst, n, lut, lut + 256, lut + 512);
-          break;
-        }
-        case OP_EQUALIZE: {
-          build_histogram(src, n, hist);
-          if (threadIdx.x < 3) equalize_lut(hist + threadIdx.x * 256, n, lut + threadIdx.x * 256);
-          __syncthreads();
-          op_pointwise_lut(src, dst, n, lut, lut + 256, lut + 512);
-          break;
-        }
-        case OP_INVERT:
-          for (int i = threadIdx.x; i < n; i += blockDim.x) {
-            uint32_t px = src[i];
-            dst[i] = pack_rgb(255 - ch_r(px), 255 - ch_g(px), 255 - ch_b(px));
-          }
-          break;
-        case OP_FLIP:
-          for (int i = threadIdx.x; i < n; i += blockDim.x) {
-            int x = i % W, y = i / W;
-            dst[i] = src[y * W + (W - 1 - x)];
-          }
-          break;
-        case OP_SOLARIZE: {
-          float th = p[0];
-          for (int i = threadIdx.x; i < n; i += blockDim.x) {
-            uint32_t px = src[i];
-            int r = ch_r(px), g = ch_g(px), bb = ch_b(px);
-            dst[i] = pack_rgb((float)r >= th ? 255 - r : r,
-                              (float)g >= th ? 255 - g : g,
-                              (float)bb >= th ? 255 - bb : bb);
-          }
-          break;
-        }
-        case OP_POSTERIZE: {
-          int bits = (int)p[0];
-          uint32_t m8 = bits >= 8 ? 0xFFu : (uint32_t)(0xFF & ~((1 << (8 - bits)) - 1));
-          uint32_t mask = m8 | (m8 << 8) | (m8 << 16);
-          for (int i = threadIdx.x; i < n; i += blockDim.x)
-            dst[i] = src[i] & mask;
-          break;
-        }
-        case OP_CONTRAST: {
-          // integer luminance sum -> PIL mean = int(sum/n + 0.5)
-          if (threadIdx.x == 0) scalar_acc[0] = 0;
-          __syncthreads();
-          uint32_t part = 0;
-          for (int i = threadIdx.x; i < n; i += blockDim.x) part += (uint32_t)lum(src[i]);
-          atomicAdd(&scalar_acc[0], part);
-          __syncthreads();
-          int meanv = (int)((double)scalar_acc[0] / n + 0.5);
-          op_blend_const(src, dst, n, meanv, meanv, meanv, p[0]);
-          break;
-        }
-        case OP_COLOR:
-          op_color(src, dst, n, p[0]);
-          break;
-        case OP_BRIGHTNESS:
-          op_blend_const(src, dst, n, 0, 0, 0, p[0]);
-          break;
-        case OP_SHARPNESS:
-          op_sharpness(src, dst, W, H, p[0]);
-          break;
-        case OP_CUTOUT:
-          op_cutout(src, dst, W, H, p);
-          break;
-        default:
-          for (int i = threadIdx.x; i < n; i += blockDim.x) dst[i] = src[i];
-      }
-      __syncthreads();
-      uint32_t* t = src; src = dst; dst = t;
-    }
+    src = run_program(src, dst, W, H, bp, hist, lut, scalar_acc);
+    dst = (src == bufA) ? bufB : bufA;
 
     // ---- post stage: crop-shift, hflip, normalize, cutout-to-zero ---------
     const float* pp = post + (int64_t)b * 6;
@@ -358,7 +365,170 @@ __global__ void aug_pipeline_kernel(
   }
 }
 
+
+// ------------------------------------------------------- imagenet pipeline
+// phase A program ops at source res -> box-crop bicubic resize (+flip) to
+// (OH,OW) -> ColorJitter ops -> normalize with lighting-shifted mean.
+// post layout: [0]=1, [1:5]=box x0,y0,w,h, [5]=flip, [6:9]=lighting rgb,
+// [9:15]=3 x (jitter code, factor).
+
+__device__ __forceinline__ float cubic_pil(float x) {
+  x = fabsf(x);
+  if (x < 1.0f) return ((1.5f * x - 2.5f) * x) * x + 1.0f;
+  if (x < 2.0f) return (((-0.5f * x) + 2.5f) * x - 4.0f) * x + 2.0f;
+  return 0.0f;
+}
+
+__device__ void op_resize_box(const uint32_t* src, uint32_t* dst, int W, int H,
+                              int OW, int OH, float bx0, float by0, float bw,
+                              float bh, bool flip) {
+  #pragma clang fp contract(off)
+  float sx_scale = bw / OW, sy_scale = bh / OH;
+  float ssx = sx_scale > 1.0f ? 1.0f / sx_scale : 1.0f;
+  float ssy = sy_scale > 1.0f ? 1.0f / sy_scale : 1.0f;
+  float supx = 2.0f * fmaxf(sx_scale, 1.0f);
+  float supy = 2.0f * fmaxf(sy_scale, 1.0f);
+  int ix0 = (int)bx0, iy0 = (int)by0, ibw = (int)bw, ibh = (int)bh;
+  for (int i = threadIdx.x; i < OW * OH; i += blockDim.x) {
+    int ox = i % OW, oy = i / OW;
+    int sxp = flip ? (OW - 1 - ox) : ox;
+    float cx = bx0 + (sxp + 0.5f) * sx_scale;
+    float cy = by0 + (oy + 0.5f) * sy_scale;
+    int xmin = (int)(cx - supx + 0.5f); if (xmin < ix0) xmin = ix0;
+    int xmax = (int)(cx + supx + 0.5f); if (xmax > ix0 + ibw) xmax = ix0 + ibw;
+    int ymin = (int)(cy - supy + 0.5f); if (ymin < iy0) ymin = iy0;
+    int ymax = (int)(cy + supy + 0.5f); if (ymax > iy0 + ibh) ymax = iy0 + ibh;
+    float accr = 0, accg = 0, accb = 0, wsum = 0;
+    for (int y = ymin; y < ymax; ++y) {
+      float wy = cubic_pil((y + 0.5f - cy) * ssy);
+      for (int x = xmin; x < xmax; ++x) {
+        float w = wy * cubic_pil((x + 0.5f - cx) * ssx);
+        uint32_t px = src[y * W + x];
+        accr += w * ch_r(px); accg += w * ch_g(px); accb += w * ch_b(px);
+        wsum += w;
+      }
+    }
+    float inv = wsum != 0.0f ? 1.0f / wsum : 0.0f;
+    int r = (int)fminf(fmaxf(rintf(accr * inv), 0.0f), 255.0f);
+    int gch = (int)fminf(fmaxf(rintf(accg * inv), 0.0f), 255.0f);
+    int bb = (int)fminf(fmaxf(rintf(accb * inv), 0.0f), 255.0f);
+    dst[i] = pack_rgb(r, gch, bb);
+  }
+}
+
+template <typename OutT>
+__global__ void aug_pipeline_in_kernel(
+    const uint8_t* __restrict__ images, int64_t img_stride,
+    const int64_t* __restrict__ sel,
+    const float* __restrict__ prog,        // [B,6,7]
+    const float* __restrict__ post,        // [B,18]
+    const float* __restrict__ mean3, const float* __restrict__ std3,
+    OutT* __restrict__ out,                // [B,OH,OW,3] channels_last
+    uint32_t* __restrict__ gws,            // [gridDim, 2, max(HW,OHW)]
+    int B, int H, int W, int OH, int OW) {
+  extern __shared__ uint32_t smem[];
+  uint32_t* hist = smem;
+  __shared__ uint8_t lut[3 * 256];
+  __shared__ uint32_t scalar_acc[1];
+  const int n = H * W;
+  const int no = OH * OW;
+  const int nmax = n > no ? n : no;
+  uint32_t* bufA = gws + (int64_t)blockIdx.x * 2 * nmax;
+  uint32_t* bufB = bufA + nmax;
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const uint8_t* img = images + sel[b] * img_stride;
+    for (int i = threadIdx.x; i < n; i += blockDim.x)
+      bufA[i] = pack_rgb(img[i * 3], img[i * 3 + 1], img[i * 3 + 2]);
+    __syncthreads();
+
+    const float* bp = prog + (int64_t)b * PROG_SLOTS * PROG_WIDTH;
+    uint32_t* src = run_program(bufA, bufB, W, H, bp, hist, lut, scalar_acc);
+    uint32_t* dst = (src == bufA) ? bufB : bufA;
+
+    const float* pp = post + (int64_t)b * 18;
+    op_resize_box(src, dst, W, H, OW, OH, pp[1], pp[2], pp[3], pp[4], pp[5] > 0.5f);
+    __syncthreads();
+    { uint32_t* t = src; src = dst; dst = t; }
+
+    // jitter ops on the resized image
+    for (int sjit = 0; sjit < 3; ++sjit) {
+      int code = (int)pp[9 + sjit * 2];
+      float f = pp[10 + sjit * 2];
+      if (code == OP_BRIGHTNESS) {
+        op_blend_const(src, dst, no, 0, 0, 0, f);
+      } else if (code == OP_CONTRAST) {
+        if (threadIdx.x == 0) scalar_acc[0] = 0;
+        __syncthreads();
+        uint32_t part = 0;
+        for (int i = threadIdx.x; i < no; i += blockDim.x) part += (uint32_t)lum(src[i]);
+        atomicAdd(&scalar_acc[0], part);
+        __syncthreads();
+        int meanv = (int)((double)scalar_acc[0] / no + 0.5);
+        op_blend_const(src, dst, no, meanv, meanv, meanv, f);
+      } else if (code == OP_COLOR) {
+        op_color(src, dst, no, f);
+      } else {
+        continue;
+      }
+      __syncthreads();
+      uint32_t* t = src; src = dst; dst = t;
+    }
+
+    // normalize with lighting folded into the mean
+    float m0 = mean3[0] - pp[6], m1 = mean3[1] - pp[7], m2 = mean3[2] - pp[8];
+    float s0 = std3[0], s1 = std3[1], s2 = std3[2];
+    OutT* orow = out + (int64_t)b * no * 3;
+    for (int i = threadIdx.x; i < no; i += blockDim.x) {
+      uint32_t px = src[i];
+      faa_from_float((ch_r(px) / 255.0f - m0) / s0, &orow[i * 3 + 0]);
+      faa_from_float((ch_g(px) / 255.0f - m1) / s1, &orow[i * 3 + 1]);
+      faa_from_float((ch_b(px) / 255.0f - m2) / s2, &orow[i * 3 + 2]);
+    }
+    __syncthreads();
+  }
+}
+
 }  // namespace
+
+torch::Tensor aug_pipeline_imagenet(torch::Tensor images, torch::Tensor sel,
+                                    torch::Tensor prog, torch::Tensor post,
+                                    torch::Tensor mean, torch::Tensor std,
+                                    int64_t out_h, int64_t out_w, bool bf16_out) {
+  TORCH_CHECK(images.is_cuda() && images.dtype() == torch::kUInt8 && images.dim() == 4);
+  TORCH_CHECK(post.size(1) == 18, "imagenet post width 18 expected");
+  int H = images.size(1), W = images.size(2);
+  int B = sel.size(0);
+  auto sel_c = sel.to(torch::kInt64).contiguous();
+  auto prog_c = prog.contiguous();
+  auto post_c = post.contiguous();
+  auto mean_c = mean.to(torch::kFloat32).contiguous();
+  auto std_c = std.to(torch::kFloat32).contiguous();
+  auto opts = images.options().dtype(bf16_out ? torch::kBFloat16 : torch::kFloat32);
+  auto out = torch::empty({B, 3, out_h, out_w},
+                          opts.memory_format(torch::MemoryFormat::ChannelsLast));
+  int64_t nmax = std::max<int64_t>((int64_t)H * W, out_h * out_w);
+  int grid = B;
+  auto gws = torch::empty({(int64_t)grid, 2, nmax}, images.options().dtype(torch::kInt32));
+  size_t lds = 3 * 256 * 4;
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  if (bf16_out) {
+    hipLaunchKernelGGL((aug_pipeline_in_kernel<__hip_bfloat16>), dim3(grid), dim3(256),
+                       lds, stream, images.data_ptr<uint8_t>(), images.stride(0),
+                       sel_c.data_ptr<int64_t>(), prog_c.data_ptr<float>(),
+                       post_c.data_ptr<float>(), mean_c.data_ptr<float>(),
+                       std_c.data_ptr<float>(), (__hip_bfloat16*)out.data_ptr(),
+                       (uint32_t*)gws.data_ptr(), B, H, W, (int)out_h, (int)out_w);
+  } else {
+    hipLaunchKernelGGL((aug_pipeline_in_kernel<float>), dim3(grid), dim3(256),
+                       lds, stream, images.data_ptr<uint8_t>(), images.stride(0),
+                       sel_c.data_ptr<int64_t>(), prog_c.data_ptr<float>(),
+                       post_c.data_ptr<float>(), mean_c.data_ptr<float>(),
+                       std_c.data_ptr<float>(), (float*)out.data_ptr(),
+                       (uint32_t*)gws.data_ptr(), B, H, W, (int)out_h, (int)out_w);
+  }
+  return out;
+}
 
 torch::Tensor aug_pipeline(torch::Tensor images, torch::Tensor sel, torch::Tensor prog,
                            torch::Tensor post, torch::Tensor mean, torch::Tensor std,

@@ -21,6 +21,10 @@ void ema_lerp_(torch::Tensor shadow, torch::Tensor x, double mu);
 torch::Tensor aug_pipeline(torch::Tensor images, torch::Tensor sel, torch::Tensor prog,
                            torch::Tensor post, torch::Tensor mean, torch::Tensor std,
                            bool bf16_out);
+torch::Tensor aug_pipeline_imagenet(torch::Tensor images, torch::Tensor sel,
+                                    torch::Tensor prog, torch::Tensor post,
+                                    torch::Tensor mean, torch::Tensor std,
+                                    int64_t out_h, int64_t out_w, bool bf16_out);
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
@@ -51,6 +55,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_fused_step_mixed", &sgd_fused_step_mixed,
         "mixed bf16-work/fp32-master fused SGD step");
   m.def("ema_lerp_", &ema_lerp_);
+  m.def("aug_pipeline_imagenet", &aug_pipeline_imagenet,
+        "imagenet pipeline: program ops + EffNet box crop-resize + jitter + lighting");
   m.def("aug_pipeline", &aug_pipeline,
         "batched augmentation program executor (uint8 NHWC -> normalized bf16/f32)");
   m.def("bn_relu_fwd", &bn_relu_fwd);

@@ -342,3 +342,52 @@ def test_sgd_fused_step_mixed_matches_cpu():
     # working copy is the bf16 quantization of the master
     assert (fg.flat_param.float().cpu() -
             fg.flat_master.cpu().bfloat16().float()).abs().max().item() == 0
+
+
+def test_aug_pipeline_imagenet():
+    """EffNet crop-resize + jitter + lighting pipeline vs the CPU reference.
+    Small sizes keep the pure-python bicubic reference fast."""
+    from fast_autoaugment_amd.aug import cpu_exec, ops as aug_ops
+    from fast_autoaugment_amd.aug.imagenet import compile_post_imagenet
+    rng = np.random.default_rng(11)
+    H = W = 48
+    OS = 32
+    B = 6
+    imgs = rng.integers(0, 256, size=(16, H, W, 3), dtype=np.uint8)
+    sel = rng.integers(0, 16, size=B)
+    prog = np.zeros((B, aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH), np.float32)
+    prog[:, 0, 0] = 3  # invert as a phase-A op
+    post = compile_post_imagenet(B, W, H, rng, OS, train=True)
+    mean = np.array([0.485, 0.456, 0.406], np.float32)
+    std = np.array([0.229, 0.224, 0.225], np.float32)
+
+    ref = cpu_exec.run_pipeline_imagenet_cpu(imgs[sel], prog, post, mean, std, OS, OS)
+
+    t = lambda a: torch.from_numpy(np.ascontiguousarray(a)).to(dev())
+    out = C.aug_pipeline_imagenet(t(imgs), t(sel), t(prog), t(post), t(mean), t(std),
+                                  OS, OS, False)
+    got = out.permute(0, 2, 3, 1).contiguous().cpu().numpy()
+    # resize accumulation order differs CPU vs GPU: allow 1/255 per pixel
+    err = np.abs(got - ref) * std.reshape(1, 1, 1, 3) * 255.0
+    assert err.max() < 1.5, f"imagenet pipeline max err {err.max()}"
+
+
+def test_aug_pipeline_imagenet_eval_centercrop():
+    from fast_autoaugment_amd.aug import cpu_exec, ops as aug_ops
+    from fast_autoaugment_amd.aug.imagenet import compile_post_imagenet
+    rng = np.random.default_rng(12)
+    H = W = 64
+    OS = 40
+    B = 4
+    imgs = rng.integers(0, 256, size=(8, H, W, 3), dtype=np.uint8)
+    sel = np.arange(B)
+    prog = np.zeros((B, aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH), np.float32)
+    post = compile_post_imagenet(B, W, H, rng, OS, train=False)
+    mean = np.zeros(3, np.float32)
+    std = np.ones(3, np.float32)
+    ref = cpu_exec.run_pipeline_imagenet_cpu(imgs[sel], prog, post, mean, std, OS, OS)
+    t = lambda a: torch.from_numpy(np.ascontiguousarray(a)).to(dev())
+    out = C.aug_pipeline_imagenet(t(imgs), t(sel), t(prog), t(post), t(mean), t(std),
+                                  OS, OS, False)
+    got = out.permute(0, 2, 3, 1).contiguous().cpu().numpy()
+    assert np.abs(got - ref).max() * 255 < 1.5
