@@ -35,6 +35,8 @@ def cv_split(labels: np.ndarray, split: float, split_idx: int,
 def reduce_dataset(labels: np.ndarray, keep: int, random_state: int = 0) -> np.ndarray:
     """reduced_cifar10 (4k) / reduced_svhn (1k) index selection
     (reference data.py:117-144): stratified keep-subset."""
+    if keep >= len(labels):      # capped synthetic sets can already be <= keep
+        return np.arange(len(labels))
     train_idx, _ = stratified_split(labels, test_size=len(labels) - keep,
                                     random_state=random_state)
     return train_idx
