@@ -65,7 +65,9 @@ class FaaConv2dFn(torch.autograd.Function):
                         padding=ctx.padding).contiguous(
                             memory_format=torch.channels_last)
                 if ctx.has_bias and ctx.needs_input_grad[2]:
-                    dbias = dy.sum(dim=(0, 2, 3))
+                    nch = dy.size(1)
+                    dbias = (C.colsum_bf16(dy) if dy.dtype == torch.bfloat16
+                             and nch % 8 == 0 else dy.sum(dim=(0, 2, 3)))
         return dx, dw, dbias, None, None
 
 

@@ -419,6 +419,26 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w,
   return conv2d_fwd(dyc, w2, torch::Tensor(), 1, KH - 1 - pad);
 }
 
+torch::Tensor colsum_bf16(torch::Tensor dy) {
+  // dbias[n] = sum over rows of a [*, C] bf16 channels-last tensor
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  int C = dyc.size(1);
+  int64_t M = dyc.numel() / C;
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto part = torch::zeros({C}, dyc.options().dtype(torch::kFloat32));
+  TORCH_CHECK(C % 8 == 0, "colsum_bf16: C % 8 == 0 expected");
+  auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
+  int q = C / gcd(C, 2048);
+  int nb = ((64 + q - 1) / q) * q;
+  hipLaunchKernelGGL(colsum_kernel, dim3(nb), dim3(256), 0, stream,
+                     (const short*)dyc.data_ptr(), part.data_ptr<float>(),
+                     M, C);
+  auto out = torch::empty({C}, dyc.options());
+  hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     stream, part.data_ptr<float>(), (short*)out.data_ptr(), C);
+  return out;
+}
+
 std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                              int64_t stride, int64_t pad,
                                              int64_t KH, int64_t KW, bool want_bias) {

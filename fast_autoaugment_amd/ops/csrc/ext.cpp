@@ -40,6 +40,7 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w, int64_t stride,
 std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                              int64_t stride, int64_t pad,
                                              int64_t KH, int64_t KW, bool want_bias);
+torch::Tensor colsum_bf16(torch::Tensor dy);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_bcast", &scale_bcast, "out = x * s[b] (per-sample broadcast)");
@@ -63,5 +64,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd, "MFMA implicit-GEMM NHWC bf16 conv forward");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
+  m.def("colsum_bf16", &colsum_bf16, "channel column-sum (bias grad)");
   m.def("bn_relu_bwd", &bn_relu_bwd);
 }
