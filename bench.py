@@ -121,9 +121,9 @@ def main():
 
     def gen_host():
         sel = rng.integers(0, n_imgs, size=args.batch)
-        prog = aug_ops.compile_program(policy, args.batch, 32, 32, rng)
-        post = aug_ops.compile_post(args.batch, 32, 32, rng, pad=4,
-                                    cutout_len=16, train=True)
+        prog = aug_ops.compile_program_fast(policy, args.batch, 32, 32, rng)
+        post = aug_ops.compile_post_fast(args.batch, 32, 32, rng, pad=4,
+                                         cutout_len=16, train=True)
         return sel, prog, post
 
     host_q: "_q.Queue" = _q.Queue(maxsize=6)

@@ -197,3 +197,150 @@ def compile_post(batch: int, w: int, h: int, rng: np.random.Generator,
             post[b, 4] = float(rng.integers(0, h))
             post[b, 5] = float(cutout_len)
     return post
+
+
+# ------------------------------------------------- vectorized batch compiler
+
+_AFFINE_OPS = {"ShearX", "ShearY", "TranslateX", "TranslateY", "Rotate",
+               "TranslateXAbs", "TranslateYAbs"}
+
+
+class CompiledPolicy:
+    """Policy table pre-resolved to numpy arrays for the fast compiler."""
+
+    def __init__(self, policy: Sequence):
+        self.n_sub = len(policy)
+        self.max_ops = max((len(sub) for sub in policy), default=0)
+        shape = (self.n_sub, self.max_ops)
+        self.name = np.full(shape, "", dtype=object)
+        self.prob = np.zeros(shape, np.float64)
+        self.level = np.zeros(shape, np.float64)
+        self.active = np.zeros(shape, bool)
+        for i, sub in enumerate(policy):
+            for j, (name, pr, lv) in enumerate(sub):
+                self.name[i, j] = name
+                self.prob[i, j] = pr
+                self.level[i, j] = lv
+                self.active[i, j] = True
+
+
+_COMPILED_CACHE = {}
+
+
+def _compiled(policy) -> CompiledPolicy:
+    key = id(policy)
+    cp = _COMPILED_CACHE.get(key)
+    if cp is None or cp.n_sub != len(policy):
+        cp = CompiledPolicy(policy)
+        _COMPILED_CACHE[key] = cp
+    return cp
+
+
+def compile_program_fast(policy: Sequence, batch: int, w: int, h: int,
+                         rng: np.random.Generator) -> np.ndarray:
+    """Vectorized compile_program: identical op semantics, batched RNG.
+
+    Draw order differs from the scalar compiler (whole-batch draws instead of
+    per-image sequences), so outputs are distribution-equal, not bit-equal.
+    """
+    prog = np.zeros((batch, PROG_SLOTS, PROG_WIDTH), dtype=np.float32)
+    if not policy:
+        return prog
+    cp = _compiled(policy)
+    sub = rng.integers(0, cp.n_sub, size=batch)
+    gates = np.zeros((batch, cp.max_ops), bool)
+    mirror = rng.random((batch, cp.max_ops)) > 0.5
+    ur = rng.random((batch, cp.max_ops))            # prob gate draws
+    cut_u = rng.random((batch, cp.max_ops, 2))      # cutout centers
+    probs = cp.prob[sub]
+    gates = (ur <= probs) & cp.active[sub]
+    slot_pos = np.cumsum(gates, axis=1) - gates.astype(int)
+
+    names = cp.name[sub]
+    levels = cp.level[sub]
+
+    for j in range(cp.max_ops):
+        fired = gates[:, j]
+        if not fired.any():
+            continue
+        for name in np.unique(names[fired, j]):
+            m = fired & (names[:, j] == name)
+            idx = np.nonzero(m)[0]
+            lv = levels[idx, j]
+            lo, hi = OP_RANGES[name]
+            val = lv * (hi - lo) + lo
+            if name in _MIRRORED:
+                val = np.where(mirror[idx, j], -val, val)
+            pos = slot_pos[idx, j]
+            pos_ok = pos < PROG_SLOTS
+            idx, val, pos = idx[pos_ok], val[pos_ok], pos[pos_ok]
+            if name in _AFFINE_OPS:
+                a = np.ones_like(val); b = np.zeros_like(val); c = np.zeros_like(val)
+                d = np.zeros_like(val); e = np.ones_like(val); f = np.zeros_like(val)
+                if name == "ShearX":
+                    b = val
+                elif name == "ShearY":
+                    d = val
+                elif name == "TranslateX":
+                    c = val * w
+                elif name == "TranslateY":
+                    f = val * h
+                elif name == "TranslateXAbs":
+                    c = val
+                elif name == "TranslateYAbs":
+                    f = val
+                else:  # Rotate (PIL matrix about center)
+                    ang = -np.radians(val)
+                    ca, sa = np.cos(ang), np.sin(ang)
+                    cx, cy = w / 2.0, h / 2.0
+                    a, b = ca, sa
+                    d, e = -sa, ca
+                    c = ca * (-cx) + sa * (-cy) + cx
+                    f = -sa * (-cx) + ca * (-cy) + cy
+                prog[idx, pos, 0] = OpCode.AFFINE
+                prog[idx, pos, 1] = a; prog[idx, pos, 2] = b; prog[idx, pos, 3] = c
+                prog[idx, pos, 4] = d; prog[idx, pos, 5] = e; prog[idx, pos, 6] = f
+            elif name in ("Cutout", "CutoutAbs"):
+                v = val * w if name == "Cutout" else val
+                ok = v > 0
+                idx, v, pos = idx[ok], v[ok], pos[ok]
+                u = cut_u[idx, j]
+                x0 = np.maximum(0, u[:, 0] * w - v / 2.0).astype(np.int64)
+                y0 = np.maximum(0, u[:, 1] * h - v / 2.0).astype(np.int64)
+                prog[idx, pos, 0] = OpCode.CUTOUT
+                prog[idx, pos, 1] = x0
+                prog[idx, pos, 2] = y0
+                prog[idx, pos, 3] = np.minimum(w, x0 + v)
+                prog[idx, pos, 4] = np.minimum(h, y0 + v)
+            else:
+                code_map = {"AutoContrast": OpCode.AUTOCONTRAST, "Invert": OpCode.INVERT,
+                            "Equalize": OpCode.EQUALIZE, "Flip": OpCode.FLIP,
+                            "Solarize": OpCode.SOLARIZE, "Posterize": OpCode.POSTERIZE,
+                            "Posterize2": OpCode.POSTERIZE, "Contrast": OpCode.CONTRAST,
+                            "Color": OpCode.COLOR, "Brightness": OpCode.BRIGHTNESS,
+                            "Sharpness": OpCode.SHARPNESS}
+                code = code_map.get(name)
+                if code is None:
+                    continue
+                prog[idx, pos, 0] = code
+                p0 = val
+                if name in ("Posterize", "Posterize2"):
+                    p0 = np.floor(val)
+                prog[idx, pos, 1] = p0
+    return prog
+
+
+def compile_post_fast(batch: int, w: int, h: int, rng: np.random.Generator,
+                      pad: int = 4, cutout_len: int = 0, train: bool = True) -> np.ndarray:
+    post = np.zeros((batch, 6), dtype=np.float32)
+    if not train:
+        return post
+    if pad > 0:
+        post[:, 0] = rng.integers(0, 2 * pad + 1, size=batch) - pad
+        post[:, 1] = rng.integers(0, 2 * pad + 1, size=batch) - pad
+    post[:, 2] = (rng.random(batch) < 0.5).astype(np.float32)
+    if cutout_len > 0:
+        post[:, 3] = rng.integers(0, w, size=batch)
+        post[:, 4] = rng.integers(0, h, size=batch)
+        post[:, 5] = cutout_len
+    return post

@@ -107,16 +107,16 @@ class AugLoader:
         if self.imagenet_size > 0:
             from ..aug.imagenet import compile_post_imagenet
             if self.train:
-                prog = aug_ops.compile_program(self.policy, len(sel), W, H, rng)
+                prog = aug_ops.compile_program_fast(self.policy, len(sel), W, H, rng)
             else:
                 prog = np.zeros((len(sel), aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH), np.float32)
             post = compile_post_imagenet(len(sel), W, H, rng, self.imagenet_size,
                                          train=self.train)
             return sel, prog, post
         if self.train:
-            prog = aug_ops.compile_program(self.policy, len(sel), W, H, rng)
-            post = aug_ops.compile_post(len(sel), W, H, rng, pad=self.pad,
-                                        cutout_len=self.cutout, train=True)
+            prog = aug_ops.compile_program_fast(self.policy, len(sel), W, H, rng)
+            post = aug_ops.compile_post_fast(len(sel), W, H, rng, pad=self.pad,
+                                             cutout_len=self.cutout, train=True)
         else:
             prog = np.zeros((len(sel), aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH), np.float32)
             post = np.zeros((len(sel), 6), np.float32)
