@@ -323,12 +323,12 @@ __global__ void wrw_cast_kernel(const float* __restrict__ dWacc, short* __restri
   dW[i] = f2b(dWacc[(int64_t)kp * g.Cout + co]);
 }
 
-// column sum: dbias[n] = sum_m dY[m][n] (bf16 in, bf16 out via fp32)
+// column sum: dbias[n] = sum_m dY[m][n] (bf16 in, fp32 partials).
+// per-thread register accumulation -> per-block LDS fold by channel octet
+// -> ONE atomic per channel per block (per-thread atomics serialize badly).
 __global__ void colsum_kernel(const short* __restrict__ dY, float* __restrict__ partial,
                               int64_t M, int C) {
-  // grid x: 64 partial blocks; each thread owns channels via linear stride
   float s[8] = {};
-  const int c0 = (int)((((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8) % C);
   int64_t total = M * C;
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
@@ -337,10 +337,19 @@ __global__ void colsum_kernel(const short* __restrict__ dY, float* __restrict__ 
     #pragma unroll
     for (int j = 0; j < 8; ++j) s[j] += b2f(v[j]);
   }
+  __shared__ float lds[256 * 8];
   #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    int c = c0 + j; while (c >= C) c -= C;
-    atomicAdd(&partial[c], s[j]);
+  for (int j = 0; j < 8; ++j) lds[threadIdx.x * 8 + j] = s[j];
+  __syncthreads();
+  const int groups = C / 8;
+  const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % groups);
+  if ((int)threadIdx.x < C) {
+    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups)
+      acc += lds[t * 8 + lane];
+    atomicAdd(&partial[threadIdx.x], acc);
   }
 }
 
