@@ -109,3 +109,61 @@ def test_loader_distributed_sharding():
     # shards are disjoint and cover everything
     assert seen[0] | seen[1] == set(range(100))
     assert not (seen[0] & seen[1])
+
+
+def _bench_dp_worker(rank, world, port, results):
+    """Mirrors bench.py's distributed step on CPU/gloo: AugLoader batches ->
+    FlatDDP fwd/bwd -> finish_gradient_sync -> FusedSGD step."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(50 + rank)
+
+    from fast_autoaugment_amd.data.loader import AugLoader, TensorStore
+    from fast_autoaugment_amd.metrics import CrossEntropyLabelSmooth
+    from fast_autoaugment_amd.models import build_model
+    from fast_autoaugment_amd.optim import FusedSGD
+    from fast_autoaugment_amd.parallel.ddp import FlatDDP
+    from fast_autoaugment_amd.policies import resolve_aug
+
+    rng = np.random.default_rng(9)
+    imgs = rng.integers(0, 256, (64, 32, 32, 3), dtype=np.uint8)
+    labels = (np.arange(64) % 10).astype(np.int64)
+    store = TensorStore(imgs, labels, device="cpu")
+    mean = np.zeros(3, np.float32)
+    std = np.ones(3, np.float32)
+    loader = AugLoader(store, 8, resolve_aug("fa_reduced_cifar10"), train=True,
+                       mean=mean, std=std, cutout=4, rank=rank, world_size=world,
+                       seed=3, prefetch=0)
+    model = build_model({"type": "wresnet40_2"}, 10)
+    ddp = FlatDDP(model)
+    opt = FusedSGD(ddp.flat, lr=0.05, weight_decay=2e-4, grad_clip=5.0)
+    crit = CrossEntropyLabelSmooth(10, 0.0)
+    it = iter(loader)
+    for _ in range(2):
+        data, label = next(it)
+        opt.zero_grad()
+        loss = crit(ddp(data), label)
+        loss.backward()
+        ddp.finish_gradient_sync()
+        opt.step()
+    results[rank] = ddp.flat.flat_param.detach().clone()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_bench_style_dp_step():
+    world = 2
+    port = _find_port()
+    ctx = mp.get_context("spawn")
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        procs = [ctx.Process(target=_bench_dp_worker, args=(r, world, port, results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(280)
+            assert p.exitcode == 0
+        assert torch.allclose(results[0], results[1], atol=1e-6)
+        assert torch.isfinite(results[0]).all()
