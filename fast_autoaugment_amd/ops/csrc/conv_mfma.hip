@@ -67,13 +67,18 @@ __device__ __forceinline__ bf16x8 load_x8(const short* __restrict__ X,
 
 // ------------------------------------------------------------------- fwd
 
-template <bool HAS_BIAS, int BNT>
+template <bool HAS_BIAS, int BMT, int BNT>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
                      const short* __restrict__ bias, short* __restrict__ Y,
                      ConvGeom g, int M, int grid_m) {
-  // BNT=64: 4 waves as 2x2 (32x32 each); BNT=32: 4 waves as 4x1 (16x32 each)
-  __shared__ short ldsA[2][BM * LDSP];
+  // 4 waves in a fixed 2x2 grid; per-wave sub-tile (BMT/2) x (BNT/2),
+  // i.e. MFRAG=BMT/32 x NFRAG=BNT/32 fragments of 16x16. Smaller tiles
+  // multiply the workgroup count for shapes that underfill 256 CUs
+  // (profiles/pmc_conv_r01.txt: occupancy-bound at BM=64 on 8x8 stages).
+  constexpr int MFRAG = BMT / 32;
+  constexpr int NFRAG = BNT / 32;
+  __shared__ short ldsA[2][BMT * LDSP];
   __shared__ short ldsB[2][BNT * LDSP];
 
   int nwg = gridDim.x;
@@ -89,15 +94,16 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wr = (BNT == 64) ? (wave >> 1) * 32 : wave * 16;
-  const int wc = (BNT == 64) ? (wave & 1) * 32 : 0;
-  constexpr int MFRAG = (BNT == 64) ? 2 : 1;
+  const int wr = (wave >> 1) * (BMT / 2);
+  const int wc = (wave & 1) * (BNT / 2);
 
-  const int a_row = tid >> 2;
+  // A staging: BMT rows x 4 k-chunks; threads beyond BMT*4 idle on A
+  const int a_row = (tid >> 2) % BMT;
+  const bool a_act = (tid >> 2) < BMT;
   const int a_kc = (tid & 3);
-  const int m_g = bm * BM + a_row;
+  const int m_g = bm * BMT + a_row;
   int xb = 0, xho = 0, xwo = 0;
-  if (m_g < M) {
+  if (a_act && m_g < M) {
     xb = m_g / (g.Ho * g.Wo);
     int rem = m_g - xb * (g.Ho * g.Wo);
     xho = rem / g.Wo;
@@ -105,22 +111,20 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
   } else {
     xb = g.B;
   }
-  // B staging: BNT rows x 32 k = BNT*4 chunk-loads; with 256 threads each
-  // thread loads ceil(BNT*4/256) chunks (1 for BNT=64, every other thread
-  // idle for BNT=32)
   const int b_row = (tid >> 2) % BNT;
   const bool b_act = (tid >> 2) < BNT;
   const int n_g = bn * BNT + b_row;
 
   const int nk = g.kpad / BK;
-  f32x4 acc[MFRAG][2] = {};
+  f32x4 acc[MFRAG][NFRAG] = {};
   const int CELLS = g.KH * g.KW;
 
   auto load_a = [&](int kt) -> bf16x8 {
+    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (!a_act) return v;
     int kc = kt * 4 + a_kc;
     int cell = kc / g.cin_chunks;
     int ci0 = (kc - cell * g.cin_chunks) * 8;
-    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
     if (cell < CELLS) {
       int kh = cell / g.KW, kw = cell - (cell / g.KW) * g.KW;
       int hi = xho * g.stride - g.pad + kh;
@@ -130,11 +134,12 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
     return v;
   };
   auto load_b = [&](int kt) -> bf16x8 {
+    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (!b_act) return v;
     int kc = kt * 4 + a_kc;
     int cell = kc / g.cin_chunks;
     int ci0 = (kc - cell * g.cin_chunks) * 8;
-    bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (b_act && n_g < g.Cout && ci0 < g.Cin && cell < CELLS) {
+    if (n_g < g.Cout && ci0 < g.Cin && cell < CELLS) {
       const short* p = Wt + ((int64_t)n_g * CELLS + cell) * g.Cin + ci0;
       if (ci0 + 8 <= g.Cin) {
         v = *reinterpret_cast<const bf16x8*>(p);
@@ -145,7 +150,8 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
     return v;
   };
   auto write_lds = [&](int buf, bf16x8 va, bf16x8 vb) {
-    *reinterpret_cast<bf16x8*>(&ldsA[buf][a_row * LDSP + a_kc * 8]) = va;
+    if (a_act)
+      *reinterpret_cast<bf16x8*>(&ldsA[buf][a_row * LDSP + a_kc * 8]) = va;
     if (b_act)
       *reinterpret_cast<bf16x8*>(&ldsB[buf][b_row * LDSP + a_kc * 8]) = vb;
   };
@@ -163,17 +169,17 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
       na = load_a(kt + 1);
       nb = load_b(kt + 1);
     }
-    bf16x8 afrag[MFRAG], bfrag[2];
+    bf16x8 afrag[MFRAG], bfrag[NFRAG];
     #pragma unroll
     for (int f = 0; f < MFRAG; ++f)
       afrag[f] = *reinterpret_cast<const bf16x8*>(&ldsA[buf][(wr + f * 16 + fr) * LDSP + kq]);
     #pragma unroll
-    for (int f = 0; f < 2; ++f)
+    for (int f = 0; f < NFRAG; ++f)
       bfrag[f] = *reinterpret_cast<const bf16x8*>(&ldsB[buf][(wc + f * 16 + fr) * LDSP + kq]);
     #pragma unroll
     for (int fm = 0; fm < MFRAG; ++fm)
       #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
+      for (int fn = 0; fn < NFRAG; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[fm], bfrag[fn],
                                                               acc[fm][fn], 0, 0, 0);
     if (kt + 1 < nk) {
@@ -188,10 +194,10 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int m_loc = wr + fm * 16 + (lane >> 4) * 4 + r;
-      int m = bm * BM + m_loc;
+      int m = bm * BMT + m_loc;
       if (m >= M) continue;
       #pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
+      for (int fn = 0; fn < NFRAG; ++fn) {
         int n = bn * BNT + wc + fn * 16 + fr;
         if (n >= g.Cout) continue;
         float v = acc[fm][fn][r];
@@ -386,8 +392,15 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   int M = g.B * g.Ho * g.Wo;
   auto y = torch::empty({g.B, g.Cout, g.Ho, g.Wo},
                         xc.options().memory_format(torch::MemoryFormat::ChannelsLast));
-  int grid_m = (M + BM - 1) / BM;
   int bnt = (g.Cout <= 32) ? 32 : 64;
+  int bmt = 64;
+  {
+    auto blocks = [&](int bm_, int bn_) {
+      return (int64_t)((M + bm_ - 1) / bm_) * ((g.Cout + bn_ - 1) / bn_);
+    };
+    if (blocks(64, bnt) < 1024 && blocks(32, bnt) <= 4096) bmt = 32;
+  }
+  int grid_m = (M + bmt - 1) / bmt;
   int grid_n = (g.Cout + bnt - 1) / bnt;
   dim3 grid(grid_m * grid_n);
   auto stream = at::hip::getCurrentHIPStream().stream();
@@ -399,12 +412,22 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     TORCH_CHECK(bc.scalar_type() == torch::kBFloat16);
     bptr = (const short*)bc.data_ptr();
   }
-  #define CF_LAUNCH(HB, BNT)                                                    \
-    hipLaunchKernelGGL((conv_fwd_kernel<HB, BNT>), grid, dim3(256), 0, stream,  \
-                       (const short*)xc.data_ptr(), (const short*)wc.data_ptr(),\
-                       bptr, (short*)y.data_ptr(), g, M, grid_m)
-  if (has_bias) { if (bnt == 32) CF_LAUNCH(true, 32); else CF_LAUNCH(true, 64); }
-  else { if (bnt == 32) CF_LAUNCH(false, 32); else CF_LAUNCH(false, 64); }
+  #define CF_LAUNCH(HB, BMT_, BNT_)                                            \
+    hipLaunchKernelGGL((conv_fwd_kernel<HB, BMT_, BNT_>), grid, dim3(256), 0,  \
+                       stream, (const short*)xc.data_ptr(),                    \
+                       (const short*)wc.data_ptr(), bptr,                      \
+                       (short*)y.data_ptr(), g, M, grid_m)
+  if (has_bias) {
+    if (bmt == 32 && bnt == 32) CF_LAUNCH(true, 32, 32);
+    else if (bmt == 32) CF_LAUNCH(true, 32, 64);
+    else if (bnt == 32) CF_LAUNCH(true, 64, 32);
+    else CF_LAUNCH(true, 64, 64);
+  } else {
+    if (bmt == 32 && bnt == 32) CF_LAUNCH(false, 32, 32);
+    else if (bmt == 32) CF_LAUNCH(false, 32, 64);
+    else if (bnt == 32) CF_LAUNCH(false, 64, 32);
+    else CF_LAUNCH(false, 64, 64);
+  }
   #undef CF_LAUNCH
   return y;
 }
