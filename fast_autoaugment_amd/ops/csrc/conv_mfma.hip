@@ -280,10 +280,14 @@ void conv_wrw_kernel(const short* __restrict__ X, const short* __restrict__ dY,
       short v[16] = {};
       if (m < m1) {
         const short* p = dY + m * g.Cout + nt * 64 + cg * 16;
-        #pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          int n = nt * 64 + cg * 16 + j;
-          v[j] = (n < g.Cout) ? p[j] : (short)0;
+        if (nt * 64 + cg * 16 + 16 <= g.Cout) {
+          *reinterpret_cast<bf16x8*>(&v[0]) = *reinterpret_cast<const bf16x8*>(p);
+          *reinterpret_cast<bf16x8*>(&v[8]) = *reinterpret_cast<const bf16x8*>(p + 8);
+        } else {
+          for (int j = 0; j < 16; ++j) {
+            int n = nt * 64 + cg * 16 + j;
+            v[j] = (n < g.Cout) ? p[j] : (short)0;
+          }
         }
       }
       #pragma unroll
