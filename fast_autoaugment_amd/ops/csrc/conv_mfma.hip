@@ -280,7 +280,7 @@ void conv_wrw_kernel(const short* __restrict__ X, const short* __restrict__ dY,
       short v[16] = {};
       if (m < m1) {
         const short* p = dY + m * g.Cout + nt * 64 + cg * 16;
-        if (nt * 64 + cg * 16 + 16 <= g.Cout) {
+        if (nt * 64 + cg * 16 + 16 <= g.Cout && (g.Cout % 8) == 0) {
           *reinterpret_cast<bf16x8*>(&v[0]) = *reinterpret_cast<const bf16x8*>(p);
           *reinterpret_cast<bf16x8*>(&v[8]) = *reinterpret_cast<const bf16x8*>(p + 8);
         } else {
