@@ -27,9 +27,12 @@ _SPECS: Dict[str, dict] = {
 }
 
 # synthetic sizes are capped so smoke/bench runs don't spend minutes
-# generating data; override with FAA_SYNTH_TRAIN/TEST env vars.
-_SYNTH_CAP_TRAIN = int(os.environ.get("FAA_SYNTH_TRAIN", 50000))
-_SYNTH_CAP_TEST = int(os.environ.get("FAA_SYNTH_TEST", 10000))
+# generating data; override with FAA_SYNTH_TRAIN/TEST env vars (read at
+# call time so tests/smokes can adjust them).
+def _synth_cap(train: bool) -> int:
+    if train:
+        return int(os.environ.get("FAA_SYNTH_TRAIN", 50000))
+    return int(os.environ.get("FAA_SYNTH_TEST", 10000))
 
 
 def synthetic_arrays(n: int, size: int, classes: int, seed: int) -> Arrays:
@@ -89,7 +92,7 @@ def load_dataset_arrays(dataset: str, dataroot: str, train: bool = True,
         except (FileNotFoundError, OSError):
             if synthetic == "never":
                 raise
-    n = min(spec["n_train"], _SYNTH_CAP_TRAIN) if train else min(spec["n_test"], _SYNTH_CAP_TEST)
+    n = min(spec["n_train"], _synth_cap(True)) if train else min(spec["n_test"], _synth_cap(False))
     return synthetic_arrays(n, spec["size"], spec["classes"], seed=hash((base, train)) % (2**31))
 
 

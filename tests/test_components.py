@@ -215,3 +215,58 @@ def test_tpu_bn_single_process_eval():
     ref = (x - bn.running_mean.view(1, -1, 1, 1)) / torch.sqrt(
         bn.running_var.view(1, -1, 1, 1) + bn.eps)
     assert torch.allclose(bn(x), ref, atol=1e-5)
+
+
+def test_trainer_mixup_and_svhn_cpu(tmp_path):
+    """One CPU epoch with mixup conf semantics and the SVHN conf path."""
+    os.environ["FAA_SYNTH_TRAIN"] = "96"
+    os.environ["FAA_SYNTH_TEST"] = "64"
+    try:
+        from fast_autoaugment_amd.data import api as data_api
+        from fast_autoaugment_amd.engine import train_and_eval
+        data_api._STORE_CACHE.clear()
+        conf = {
+            "model": {"type": "wresnet40_2"}, "dataset": "svhn",
+            "aug": "fa_reduced_svhn", "cutout": 20, "batch": 32, "epoch": 1,
+            "lr": 0.01, "mixup": 0.2,
+            "lr_schedule": {"type": "cosine", "warmup": {"multiplier": 1, "epoch": 0}},
+            "optimizer": {"type": "sgd", "decay": 1e-4, "nesterov": True, "ema": 0},
+        }
+        C.replace(conf)
+        r = train_and_eval("", "./data", save_path=str(tmp_path / "m.pth"),
+                           evaluation_interval=1)
+        assert "top1_test" in r or "top1_train" in r
+    finally:
+        os.environ.pop("FAA_SYNTH_TRAIN", None)
+        os.environ.pop("FAA_SYNTH_TEST", None)
+        from fast_autoaugment_amd.data import api as data_api
+        data_api._STORE_CACHE.clear()
+
+
+def test_trainer_ema_rmsprop_cpu(tmp_path):
+    """EffNet-style optimizer path: RMSpropTF + per-step EMA (1 tiny epoch)."""
+    os.environ["FAA_SYNTH_TRAIN"] = "64"
+    os.environ["FAA_SYNTH_TEST"] = "32"
+    try:
+        from fast_autoaugment_amd.data import api as data_api
+        from fast_autoaugment_amd.engine import train_and_eval
+        data_api._STORE_CACHE.clear()
+        conf = {
+            "model": {"type": "wresnet40_2"}, "dataset": "cifar10",
+            "aug": "default", "cutout": 0, "batch": 32, "epoch": 1,
+            "lr": 0.001, "lb_smooth": 0.1,
+            "lr_schedule": {"type": "efficientnet",
+                            "warmup": {"multiplier": 1, "epoch": 0}},
+            "optimizer": {"type": "rmsprop", "decay": 1e-5, "clip": 0,
+                          "ema": 0.999, "ema_interval": -1},
+        }
+        C.replace(conf)
+        r = train_and_eval("", "./data", save_path=str(tmp_path / "e.pth"),
+                           evaluation_interval=1)
+        data = torch.load(str(tmp_path / "e.pth"), weights_only=False)
+        assert data["ema"] is not None and len(data["ema"]) > 0
+    finally:
+        os.environ.pop("FAA_SYNTH_TRAIN", None)
+        os.environ.pop("FAA_SYNTH_TEST", None)
+        from fast_autoaugment_amd.data import api as data_api
+        data_api._STORE_CACHE.clear()
