@@ -18,12 +18,16 @@ import torch.nn.functional as F
 from . import ext
 
 
-# Measured dispatch rule (profiles/conv_bench_r01.txt, MI355X b128):
-# the MFMA fwd/bwd-data kernels beat MIOpen for C<=128 (all WRN-40-2
-# shapes, 1.2-4x); MIOpen keeps larger channels and the wrw direction
-# until the wrw v2 staging lands.
+# Measured dispatch rules (profiles/conv_bench_r01.txt + wrw v2 re-measure,
+# MI355X b128): the MFMA fwd/bwd-data kernels beat MIOpen for C<=128 (all
+# WRN-40-2 shapes, 1.2-4x); MIOpen keeps larger channels. The wrw v2 kernel
+# wins on the stem (tiny Cin) and the deep stages (Cin>=128), loses 1.2-2x
+# in between -> shape-conditional.
 _FAA_MAX_CH = 128
-_FAA_WRW = False   # flip when conv_wrw v2 beats igemm_wrw
+
+
+def _faa_wrw_wins(cin: int) -> bool:
+    return cin < 8 or cin >= 128
 
 
 class FaaConv2dFn(torch.autograd.Function):
@@ -52,7 +56,7 @@ class FaaConv2dFn(torch.autograd.Function):
                                                 stride=ctx.stride,
                                                 padding=ctx.padding)
         if ctx.needs_input_grad[1] or (ctx.has_bias and ctx.needs_input_grad[2]):
-            if _FAA_WRW:
+            if _faa_wrw_wins(x.size(1)):
                 dw, db = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
                                              weight.size(2), weight.size(3),
                                              ctx.has_bias)
