@@ -38,7 +38,10 @@ def parse_args():
 
 def main():
     args = parse_args()
-    assert torch.cuda.is_available(), "bench.py requires a GPU"
+    # FAA_BENCH_CPU=1: integration dry-run of the exact distributed bench
+    # path on CPU/gloo (used by tests; numbers are meaningless there)
+    cpu_mode = os.environ.get("FAA_BENCH_CPU") == "1"
+    assert cpu_mode or torch.cuda.is_available(), "bench.py requires a GPU"
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -46,10 +49,12 @@ def main():
     distributed = world_size > 1
     if distributed:
         import torch.distributed as dist
-        dist.init_process_group("nccl", init_method="env://")
-        torch.cuda.set_device(local_rank)
-    dev = torch.device("cuda", local_rank)
-    torch.cuda.set_device(dev)
+        dist.init_process_group("gloo" if cpu_mode else "nccl", init_method="env://")
+        if not cpu_mode:
+            torch.cuda.set_device(local_rank)
+    dev = torch.device("cpu") if cpu_mode else torch.device("cuda", local_rank)
+    if not cpu_mode:
+        torch.cuda.set_device(dev)
 
     from fast_autoaugment_amd.config import Config as C
     from fast_autoaugment_amd.data.loader import AugLoader, TensorStore
@@ -83,7 +88,7 @@ def main():
     model = build_model(conf["model"], nc).to(dev).to(memory_format=torch.channels_last)
     # pure-bf16 compute: params+grads are bf16 flat views (no autocast cast
     # kernels), fp32 master lives in the fused optimizer
-    work_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    work_dtype = torch.bfloat16 if (args.dtype == "bf16" and not cpu_mode) else torch.float32
     if distributed:
         from fast_autoaugment_amd.parallel.ddp import FlatDDP
         model = FlatDDP(model, work_dtype=work_dtype)
@@ -106,10 +111,11 @@ def main():
     steps_per_epoch = max(len(loader), 1)
     step_idx = 0
 
-    use_graphs = bool(args.graphs) and not distributed
+    use_graphs = bool(args.graphs) and not distributed and not cpu_mode
     from fast_autoaugment_amd.aug import ops as aug_ops
-    from fast_autoaugment_amd.ops import ext
-    CX = ext()
+    if not cpu_mode:
+        from fast_autoaugment_amd.ops import ext
+        CX = ext()
 
     # host-side program generation, prefetched on a thread (the analog of the
     # reference's 8 DataLoader workers): the GPU never waits for host RNG
@@ -143,9 +149,10 @@ def main():
     prog_s = torch.zeros((args.batch, aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH),
                          dtype=torch.float32, device=dev)
     post_s = torch.zeros((args.batch, 6), dtype=torch.float32, device=dev)
-    sel_h = torch.zeros_like(sel_s, device="cpu", pin_memory=True)
-    prog_h = torch.zeros_like(prog_s, device="cpu", pin_memory=True)
-    post_h = torch.zeros_like(post_s, device="cpu", pin_memory=True)
+    pin = not cpu_mode
+    sel_h = torch.zeros_like(sel_s, device="cpu", pin_memory=pin)
+    prog_h = torch.zeros_like(prog_s, device="cpu", pin_memory=pin)
+    post_h = torch.zeros_like(post_s, device="cpu", pin_memory=pin)
 
     def upload_next():
         sel, prog, post = host_q.get()
@@ -156,10 +163,20 @@ def main():
         prog_s.copy_(prog_h, non_blocking=True)
         post_s.copy_(post_h, non_blocking=True)
 
+    def make_batch_cpu():
+        from fast_autoaugment_amd.aug import cpu_exec
+        sel, prog, post2 = host_q.get()
+        out = cpu_exec.run_pipeline_cpu(store.images_np[sel], prog, post2, mean, std)
+        data = torch.from_numpy(out).permute(0, 3, 1, 2).contiguous()
+        return data, store.labels[torch.from_numpy(np.ascontiguousarray(sel))]
+
     def step_body():
         opt.zero_grad()
-        data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s, mean_t, std_t, bf16)
-        label = store.labels.index_select(0, sel_s)
+        if cpu_mode:
+            data, label = make_batch_cpu()
+        else:
+            data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s, mean_t, std_t, bf16)
+            label = store.labels.index_select(0, sel_s)
         preds = model(data)
         loss = crit(preds, label)
         loss.backward()
@@ -192,7 +209,8 @@ def main():
             upload_next()
             graph.replay()
         else:
-            upload_next()
+            if not cpu_mode:
+                upload_next()
             step_body()
         step_idx += 1
 
@@ -203,11 +221,13 @@ def main():
     if distributed:
         import torch.distributed as dist
         dist.barrier()
-    torch.cuda.synchronize()
+    if not cpu_mode:
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         one_step()
-    torch.cuda.synchronize()
+    if not cpu_mode:
+        torch.cuda.synchronize()
     if distributed:
         import torch.distributed as dist
         dist.barrier()
@@ -215,7 +235,7 @@ def main():
 
     if distributed:
         import torch.distributed as dist
-        t = torch.tensor([elapsed], device=dev)
+        t = torch.tensor([elapsed], device=dev if not cpu_mode else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 

@@ -167,3 +167,28 @@ def test_bench_style_dp_step():
             assert p.exitcode == 0
         assert torch.allclose(results[0], results[1], atol=1e-6)
         assert torch.isfinite(results[0]).all()
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_cpu_dry_run():
+    """The driver launches bench.py under torch.distributed.run; dry-run that
+    exact invocation on CPU/gloo (FAA_BENCH_CPU=1) and check the JSON line."""
+    import json
+    import subprocess
+    import sys
+    env = dict(os.environ, FAA_BENCH_CPU="1", FAA_SYNTH_TRAIN="256",
+               FAA_SYNTH_TEST="64")
+    root = os.path.join(os.path.dirname(__file__), "..")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(_find_port()), "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--batch", "16"],
+        env=env, cwd=root, capture_output=True, text=True, timeout=550)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 2
+    assert out["metric"] == "images/sec"
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["config"]["global_batch"] == 32
