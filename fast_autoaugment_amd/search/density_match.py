@@ -21,6 +21,29 @@ from ..metrics import Accumulator
 from ..models import get_model, num_class
 from ..policies import policy_decoder
 
+# Worker-process cache: scheduler workers evaluate ~hundreds of trials
+# against the SAME frozen fold checkpoint (reference reloads it per trial,
+# search.py:80-84); caching the loaded model cuts per-trial overhead to the
+# inference itself.
+_MODEL_CACHE: Dict = {}
+
+
+def _cached_model(conf, nc, save_path: str, device: str):
+    import os as _os
+    key = (save_path, conf["model"].get("type"), nc)
+    mtime = _os.path.getmtime(save_path)
+    hit = _MODEL_CACHE.get(key)
+    if hit is not None and hit[0] == mtime:
+        return hit[1]
+    model = get_model(conf["model"], nc, local_rank=-1, device=device)
+    ckpt = torch.load(save_path, map_location=device, weights_only=False)
+    sd = ckpt["model"] if "model" in ckpt else ckpt
+    model.load_state_dict({k.replace("module.", ""): v for k, v in sd.items()})
+    model.eval()
+    _MODEL_CACHE.clear()
+    _MODEL_CACHE[key] = (mtime, model)
+    return model
+
 
 def eval_tta(conf_dict: Dict, augment: Dict, reporter=None) -> float:
     C.replace(conf_dict)
@@ -33,11 +56,7 @@ def eval_tta(conf_dict: Dict, augment: Dict, reporter=None) -> float:
     use_cuda = torch.cuda.is_available()
     device = "cuda" if use_cuda else "cpu"
     nc = num_class(conf["dataset"])
-    model = get_model(conf["model"], nc, local_rank=-1, device=device)
-    ckpt = torch.load(save_path, map_location=device, weights_only=False)
-    sd = ckpt["model"] if "model" in ckpt else ckpt
-    model.load_state_dict({k.replace("module.", ""): v for k, v in sd.items()})
-    model.eval()
+    model = _cached_model(conf, nc, save_path, device)
 
     autocast_dtype = torch.bfloat16 if use_cuda and conf.get_value("precision", "bf16") == "bf16" else None
     out_dtype = torch.bfloat16 if autocast_dtype else torch.float32

@@ -19,11 +19,15 @@ def main():
     ap.add_argument("--epoch", type=int, default=1)
     ap.add_argument("--num-search", type=int, default=3)
     ap.add_argument("--cv-num", type=int, default=1)
+    ap.add_argument("--dataset", type=str, default=None)
+    ap.add_argument("--batch", type=int, default=64)
     args = ap.parse_args()
 
     C.load("confs/wresnet40x2_cifar.yaml")
     C.get()["epoch"] = args.epoch
-    C.get()["batch"] = 64
+    C.get()["batch"] = args.batch
+    if args.dataset:
+        C.get()["dataset"] = args.dataset
     r = run_search("./data", until=args.until, num_op=2, num_policy=2,
                    num_search=args.num_search, cv_ratio=0.4, cv_num=args.cv_num,
                    num_result_per_cv=2, n_workers=args.workers, resume=True)
