@@ -84,15 +84,24 @@ def flatten_module(model: nn.Module, align: int = 64,
     if work_dtype != torch.float32:
         flat_master = torch.zeros(total, dtype=torch.float32, device=device)
 
+    def make_view(buf, off, n, p):
+        # 4D channels_last params get channels_last-strided views so conv
+        # weight gradients accumulate along matching strides (vectorized
+        # add instead of the permuting slow path)
+        if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last):
+            o, i, kh, kw = p.shape
+            return buf[off:off + n].view(o, kh, kw, i).permute(0, 3, 1, 2)
+        return buf[off:off + n].view_as(p)
+
     params = []
     for (name, p), (off, n) in zip(ordered, offsets):
         if flat_master is not None:
             with torch.no_grad():
-                flat_master[off:off + n].view_as(p).copy_(p.detach().float())
-        view = flat_param[off:off + n].view_as(p)
+                make_view(flat_master, off, n, p).copy_(p.detach().float())
+        view = make_view(flat_param, off, n, p)
         with torch.no_grad():
             view.copy_(p.detach().to(work_dtype))
         p.data = view
-        p.grad = flat_grad[off:off + n].view_as(p)
+        p.grad = make_view(flat_grad, off, n, p)
         params.append(p)
     return FlatParams(flat_param, flat_grad, n_decay, params, flat_master)
