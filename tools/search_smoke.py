@@ -20,6 +20,7 @@ def main():
     ap.add_argument("--num-search", type=int, default=3)
     ap.add_argument("--cv-num", type=int, default=1)
     ap.add_argument("--dataset", type=str, default=None)
+    ap.add_argument("--num-policy", type=int, default=2)
     ap.add_argument("--batch", type=int, default=64)
     args = ap.parse_args()
 
@@ -28,7 +29,7 @@ def main():
     C.get()["batch"] = args.batch
     if args.dataset:
         C.get()["dataset"] = args.dataset
-    r = run_search("./data", until=args.until, num_op=2, num_policy=2,
+    r = run_search("./data", until=args.until, num_op=2, num_policy=args.num_policy,
                    num_search=args.num_search, cv_ratio=0.4, cv_num=args.cv_num,
                    num_result_per_cv=2, n_workers=args.workers, resume=True)
     print("search keys:", sorted(r.keys()),
