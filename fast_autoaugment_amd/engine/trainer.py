@@ -287,6 +287,8 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
                                 is_master=is_master, device=device,
                                 autocast_dtype=autocast_dtype)
         model.eval()
+        from ..ops.bnrelu import sync_bn_trackers
+        sync_bn_trackers(model.module if hasattr(model, "module") else model)
 
         if math.isnan(rs["train"]["loss"]):
             raise Exception("train loss is NaN.")

@@ -27,7 +27,19 @@ class FusedBNReLUFn(torch.autograd.Function):
 def fused_bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
     training = bn.training
     if training and bn.num_batches_tracked is not None:
-        bn.num_batches_tracked.add_(1)
+        # a per-call .add_(1) is a 5us GPU kernel x 37 BN sites x step; count
+        # on host and fold into the buffer lazily (sync_bn_trackers)
+        bn._faa_nbt_pending = getattr(bn, "_faa_nbt_pending", 0) + 1
     momentum = bn.momentum if bn.momentum is not None else 0.1
     return FusedBNReLUFn.apply(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
                                training, momentum, bn.eps)
+
+
+def sync_bn_trackers(model: torch.nn.Module) -> None:
+    """Fold host-side BN forward counts into num_batches_tracked buffers.
+    Call at epoch end / before state_dict save."""
+    for m in model.modules():
+        pend = getattr(m, "_faa_nbt_pending", 0)
+        if pend and getattr(m, "num_batches_tracked", None) is not None:
+            m.num_batches_tracked.add_(pend)
+            m._faa_nbt_pending = 0
