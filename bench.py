@@ -33,6 +33,11 @@ def parse_args():
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--graphs", type=int, default=1,
                    help="1: capture the train step in a hipGraph (default), 0: eager")
+    p.add_argument("--grad-mode", type=str, default="gather",
+                   choices=["gather", "flat"],
+                   help="gather: backward assigns fresh grads (no per-param "
+                        "accumulation kernels) packed by one multi-tensor "
+                        "kernel; flat: grads accumulate into flat views")
     return p.parse_args()
 
 
@@ -185,7 +190,22 @@ def main():
         opt.step()
         return loss
 
+    # --grad-mode gather (graphs only): backward ASSIGNS fresh grad tensors
+    # (p.grad=None -> no accumulation add kernels); under graph capture the
+    # allocations are address-stable across replays, so one multi-tensor
+    # gather kernel packs them into the flat buffer for the fused step.
+    gather_mode = (args.grad_mode == "gather" and use_graphs)
+
+    def fwd_bwd_body():
+        data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s, mean_t, std_t, bf16)
+        label = store.labels.index_select(0, sel_s)
+        preds = model(data)
+        loss = crit(preds, label)
+        loss.backward()
+        return loss
+
     graph = None
+    gather_table = None
     if use_graphs:
         opt.sync_lr()
         side = torch.cuda.Stream()
@@ -193,13 +213,32 @@ def main():
         with torch.cuda.stream(side):
             for _ in range(3):
                 upload_next()
-                step_body()
+                if gather_mode:
+                    fwd_bwd_body()
+                else:
+                    step_body()
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()
         upload_next()
-        with torch.cuda.graph(graph):
-            step_body()
+        if gather_mode:
+            for p_ in flat.params:
+                p_.grad = None          # capture assignment-mode backward
+            with torch.cuda.graph(graph):
+                fwd_bwd_body()
+            # grads now live in the graph pool at replay-stable addresses
+            base = flat.flat_param.data_ptr()
+            rows = []
+            for p_ in flat.params:
+                g_ = p_.grad
+                assert g_ is not None and g_.dtype == torch.bfloat16
+                off = (p_.data.data_ptr() - base) // 2
+                rows.append([g_.data_ptr(), off, g_.numel()])
+            gather_table = torch.tensor(rows, dtype=torch.int64, device=dev)
+            flat.flat_grad.zero_()      # pad gaps stay zero forever
+        else:
+            with torch.cuda.graph(graph):
+                step_body()
 
     def one_step():
         nonlocal step_idx
@@ -208,6 +247,9 @@ def main():
             opt.sync_lr()
             upload_next()
             graph.replay()
+            if gather_mode:
+                CX.gather_grads(gather_table, flat.flat_grad)
+                opt.step()
         else:
             if not cpu_mode:
                 upload_next()

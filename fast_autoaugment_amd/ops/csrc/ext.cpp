@@ -18,6 +18,7 @@ void sgd_fused_step_mixed(torch::Tensor master, torch::Tensor work, torch::Tenso
                           int64_t n_decay, double wd, double clip, double momentum,
                           int64_t nesterov);
 void ema_lerp_(torch::Tensor shadow, torch::Tensor x, double mu);
+void gather_grads(torch::Tensor table, torch::Tensor flat);
 torch::Tensor aug_pipeline(torch::Tensor images, torch::Tensor sel, torch::Tensor prog,
                            torch::Tensor post, torch::Tensor mean, torch::Tensor std,
                            bool bf16_out);
@@ -56,6 +57,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_fused_step_mixed", &sgd_fused_step_mixed,
         "mixed bf16-work/fp32-master fused SGD step");
   m.def("ema_lerp_", &ema_lerp_);
+  m.def("gather_grads", &gather_grads,
+        "pack scattered autograd grads into the flat bf16 buffer");
   m.def("aug_pipeline_imagenet", &aug_pipeline_imagenet,
         "imagenet pipeline: program ops + EffNet box crop-resize + jitter + lighting");
   m.def("aug_pipeline", &aug_pipeline,

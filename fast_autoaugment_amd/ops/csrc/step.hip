@@ -210,7 +210,40 @@ __global__ void sgd_step_mixed_kernel(float* __restrict__ p, short* __restrict__
   }
 }
 
+// ---- multi-tensor grad gather: scattered autograd grads -> flat buffer ----
+// table rows: [src_ptr, dst_offset, numel] (int64). One block per tensor,
+// vectorized bf16x8 copy. Lets backward run with .grad=None (no per-param
+// accumulation add kernels); addresses are stable under hipGraph replay.
+typedef __attribute__((ext_vector_type(4))) short short4g;
+
+__global__ void gather_grads_kernel(const int64_t* __restrict__ table, int n,
+                                    short* __restrict__ flat) {
+  for (int e = blockIdx.x; e < n; e += gridDim.x) {
+    const short* src = (const short*)table[e * 3];
+    int64_t dst = table[e * 3 + 1];
+    int64_t len = table[e * 3 + 2];
+    int64_t i = (int64_t)threadIdx.x * 8;
+    for (; i + 8 <= len; i += (int64_t)blockDim.x * 8) {
+      *reinterpret_cast<short4g*>(flat + dst + i) =
+          *reinterpret_cast<const short4g*>(src + i);
+      *reinterpret_cast<short4g*>(flat + dst + i + 4) =
+          *reinterpret_cast<const short4g*>(src + i + 4);
+    }
+    if (threadIdx.x == 0)
+      for (int64_t j = (len / 8) * 8; j < len; ++j) flat[dst + j] = src[j];
+  }
+}
+
 }  // namespace
+
+void gather_grads(torch::Tensor table, torch::Tensor flat) {
+  TORCH_CHECK(table.dtype() == torch::kInt64 && table.is_cuda());
+  TORCH_CHECK(flat.dtype() == torch::kBFloat16);
+  int n = table.size(0);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(gather_grads_kernel, dim3(std::min(n, 512)), dim3(256), 0, stream,
+                     table.data_ptr<int64_t>(), n, (short*)flat.data_ptr());
+}
 
 void sgd_fused_step_mixed(torch::Tensor master, torch::Tensor work, torch::Tensor g,
                           torch::Tensor buf, torch::Tensor normsq, torch::Tensor lr_t,
