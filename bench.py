@@ -82,7 +82,8 @@ def main():
 
     nc = num_class(args.dataset)
     # synthetic data, random-init weights (no network on the box)
-    imgs, labels = synthetic_arrays(50000, 32, nc, seed=1234 + rank)
+    n_synth = int(os.environ.get("FAA_BENCH_IMGS", "50000"))
+    imgs, labels = synthetic_arrays(n_synth, 32, nc, seed=1234 + rank)
     store = TensorStore(imgs, labels, device=str(dev))
     mean, std = dataset_stats(args.dataset)
     out_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
@@ -280,6 +281,22 @@ def main():
         t = torch.tensor([elapsed], device=dev if not cpu_mode else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
+
+    if os.environ.get("FAA_BENCH_DEBUG") == "1" and rank == 0:
+        # sanity: read back the last loss (graph modes keep it in a static
+        # tensor inside the capture pool via crit; recompute one eager loss)
+        with torch.no_grad():
+            sel = torch.arange(min(args.batch, len(store)), device=dev)
+            zero_prog = torch.zeros_like(prog_s)
+            zero_post = torch.zeros_like(post_s)
+            data = (CX.aug_pipeline(store.images, sel, zero_prog, zero_post,
+                                    mean_t, std_t, bf16) if not cpu_mode else None)
+            if data is not None:
+                model.eval()
+                preds = model(data)
+                l = crit(preds, store.labels.index_select(0, sel))
+                print(f"# debug eval loss: {l.item():.4f}", file=__import__('sys').stderr)
+                model.train()
 
     total_images = args.batch * world_size * args.steps
     ips = total_images / elapsed
