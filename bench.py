@@ -298,6 +298,11 @@ def main():
                 print(f"# debug eval loss: {l.item():.4f}", file=__import__('sys').stderr)
                 model.train()
 
+    save_p = os.environ.get("FAA_BENCH_SAVE")
+    if save_p and rank == 0:
+        torch.save((flat.flat_master if flat.flat_master is not None
+                    else flat.flat_param).detach().cpu(), save_p)
+
     total_images = args.batch * world_size * args.steps
     ips = total_images / elapsed
     if rank == 0:
