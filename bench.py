@@ -214,10 +214,10 @@ def main():
         with torch.cuda.stream(side):
             for _ in range(3):
                 upload_next()
-                if gather_mode:
-                    fwd_bwd_body()
-                else:
-                    step_body()
+                # warmup always runs the FULL step (flat grad views are still
+                # attached in gather mode) so both grad modes reach capture
+                # with identical param/momentum state
+                step_body()
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()
