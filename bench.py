@@ -233,6 +233,12 @@ def main():
             for p_ in flat.params:
                 g_ = p_.grad
                 assert g_ is not None and g_.dtype == torch.bfloat16
+                # layout must match the flat region's element order
+                if g_.dim() == 4:
+                    assert g_.is_contiguous(memory_format=torch.channels_last), \
+                        f"grad layout mismatch for {tuple(g_.shape)}"
+                else:
+                    assert g_.is_contiguous()
                 off = (p_.data.data_ptr() - base) // 2
                 rows.append([g_.data_ptr(), off, g_.numel()])
             gather_table = torch.tensor(rows, dtype=torch.int64, device=dev)
@@ -300,8 +306,11 @@ def main():
 
     save_p = os.environ.get("FAA_BENCH_SAVE")
     if save_p and rank == 0:
-        torch.save((flat.flat_master if flat.flat_master is not None
-                    else flat.flat_param).detach().cpu(), save_p)
+        if os.environ.get("FAA_BENCH_SAVE_GRAD") == "1":
+            torch.save(flat.flat_grad.detach().float().cpu(), save_p)
+        else:
+            torch.save((flat.flat_master if flat.flat_master is not None
+                        else flat.flat_param).detach().cpu(), save_p)
 
     total_images = args.batch * world_size * args.steps
     ips = total_images / elapsed
