@@ -160,14 +160,17 @@ def main():
     prog_h = torch.zeros_like(prog_s, device="cpu", pin_memory=pin)
     post_h = torch.zeros_like(post_s, device="cpu", pin_memory=pin)
 
+    sync_upload = os.environ.get("FAA_BENCH_SYNC_UPLOAD", "0") == "1"
+
     def upload_next():
         sel, prog, post = host_q.get()
         sel_h.copy_(torch.from_numpy(sel))
         prog_h.copy_(torch.from_numpy(prog))
         post_h.copy_(torch.from_numpy(post))
-        sel_s.copy_(sel_h, non_blocking=True)
-        prog_s.copy_(prog_h, non_blocking=True)
-        post_s.copy_(post_h, non_blocking=True)
+        nb = not sync_upload
+        sel_s.copy_(sel_h, non_blocking=nb)
+        prog_s.copy_(prog_h, non_blocking=nb)
+        post_s.copy_(post_h, non_blocking=nb)
 
     def make_batch_cpu():
         from fast_autoaugment_amd.aug import cpu_exec
