@@ -98,6 +98,9 @@ def _faa_forward(self, x):
 
 def patch_convs(model: torch.nn.Module) -> int:
     """Rebind eligible Conv2d forwards to the MFMA kernels. Returns count."""
+    import os
+    if os.environ.get("FAA_NO_PATCH") == "1":
+        return 0
     n = 0
     for m in model.modules():
         if isinstance(m, torch.nn.Conv2d) and _eligible(m):

@@ -14,9 +14,13 @@ import torch.nn.functional as F
 from . import has_ext
 
 
+import os
+
+
 def bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
     """BatchNorm2d followed by ReLU, fused on GPU."""
-    if x.is_cuda and has_ext():
+    if (x.is_cuda and has_ext()
+            and os.environ.get("FAA_NO_FUSED_BN") != "1"):
         from .bnrelu import fused_bn_relu
         return fused_bn_relu(x, bn)
     return F.relu(bn(x))

@@ -2,11 +2,11 @@ import os, subprocess, sys
 import torch
 sys.path.insert(0, ".")
 
-def run(tag, sync="0", n=4):
+def run(tag, sync="0", n=4, **extra_env):
     bad = 0
     for i in range(n):
         env = dict(os.environ, FAA_BENCH_IMGS="2048", FAA_BENCH_SAVE="/tmp/f.pt",
-                   FAA_BENCH_SYNC_UPLOAD=sync)
+                   FAA_BENCH_SYNC_UPLOAD=sync, **extra_env)
         r = subprocess.run([sys.executable, "bench.py", "--steps", "1", "--warmup", "2"],
                            env=env, capture_output=True, text=True, timeout=280)
         assert r.returncode == 0, r.stderr[-400:]
@@ -15,5 +15,7 @@ def run(tag, sync="0", n=4):
             bad += 1
     print(f"{tag}: {bad}/{n} runs NaN")
 
-run("async-upload", "0")
-run("sync-upload", "1")
+run("baseline", n=4)
+run("no-mfma-convs", n=4, FAA_NO_PATCH="1")
+run("no-fused-bn", n=4, FAA_NO_FUSED_BN="1")
+run("neither", n=4, FAA_NO_PATCH="1", FAA_NO_FUSED_BN="1")
