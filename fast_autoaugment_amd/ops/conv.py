@@ -26,7 +26,15 @@ from . import ext
 _FAA_MAX_CH = 128
 
 
+import os as _os
+
+
 def _faa_wrw_wins(cin: int) -> bool:
+    mode = _os.environ.get("FAA_WRW", "auto")
+    if mode == "faa":
+        return True
+    if mode == "torch":
+        return False
     return cin < 8 or cin >= 128
 
 
@@ -48,7 +56,7 @@ class FaaConv2dFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=torch.channels_last)
         dx = dw = dbias = None
         if ctx.needs_input_grad[0]:
-            if ctx.stride == 1:
+            if ctx.stride == 1 and _os.environ.get("FAA_BWD_DATA", "faa") != "torch":
                 dx = C.conv2d_bwd_data(dy, weight, 1, ctx.padding,
                                        x.size(2), x.size(3))
             else:

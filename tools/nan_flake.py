@@ -16,6 +16,7 @@ def run(tag, sync="0", n=4, **extra_env):
     print(f"{tag}: {bad}/{n} runs NaN")
 
 run("baseline", n=4)
-run("no-mfma-convs", n=4, FAA_NO_PATCH="1")
-run("no-fused-bn", n=4, FAA_NO_FUSED_BN="1")
-run("neither", n=4, FAA_NO_PATCH="1", FAA_NO_FUSED_BN="1")
+run("wrw-all-torch", n=4, FAA_WRW="torch")
+run("wrw-all-faa", n=4, FAA_WRW="faa")
+run("bw0d-torch", n=4, FAA_BWD_DATA="torch")
+run("wrwT+bwdT(fwd-only)", n=4, FAA_WRW="torch", FAA_BWD_DATA="torch")
