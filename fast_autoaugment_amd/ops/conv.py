@@ -78,8 +78,9 @@ class FaaConv2dFn(torch.autograd.Function):
                             memory_format=torch.channels_last)
                 if ctx.has_bias and ctx.needs_input_grad[2]:
                     nch = dy.size(1)
-                    dbias = (C.colsum_bf16(dy) if dy.dtype == torch.bfloat16
-                             and nch % 8 == 0 else dy.sum(dim=(0, 2, 3)))
+                    use_cs = (dy.dtype == torch.bfloat16 and nch % 8 == 0
+                              and _os.environ.get("FAA_NO_COLSUM") != "1")
+                    dbias = C.colsum_bf16(dy) if use_cs else dy.sum(dim=(0, 2, 3))
         return dx, dw, dbias, None, None
 
 

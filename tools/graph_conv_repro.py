@@ -38,3 +38,36 @@ for s in range(6):
     tot += bad
     print(f"trial {s}: {bad}/10 replays wrong")
 print("TOTAL bad replays:", tot)
+
+
+def trial_colsum(seed):
+    torch.manual_seed(seed)
+    dy = (torch.randn(128, 32, 32, 32, device="cuda") * 0.3).bfloat16().contiguous(
+        memory_format=torch.channels_last)
+    ref = C.colsum_bf16(dy).float().clone()
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for _ in range(3):
+            C.colsum_bf16(dy)
+    torch.cuda.current_stream().wait_stream(side)
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        out = C.colsum_bf16(dy)
+    bad = 0
+    for i in range(10):
+        g.replay()
+        torch.cuda.synchronize()
+        if (out.float() - ref).abs().max().item() > 0.5 or torch.isnan(out.float()).any():
+            bad += 1
+    return bad
+
+
+print("=== colsum graph replay ===")
+tot = 0
+for s in range(6):
+    b = trial_colsum(s)
+    tot += b
+    print(f"colsum trial {s}: {b}/10 wrong")
+print("COLSUM total bad:", tot)
