@@ -15,8 +15,9 @@ def run(tag, sync="0", n=4, **extra_env):
             bad += 1
     print(f"{tag}: {bad}/{n} runs NaN")
 
-run("baseline", n=4)
-run("wrw-all-torch", n=4, FAA_WRW="torch")
-run("wrw-all-faa", n=4, FAA_WRW="faa")
-run("bw0d-torch", n=4, FAA_BWD_DATA="torch")
-run("wrwT+bwdT(fwd-only)", n=4, FAA_WRW="torch", FAA_BWD_DATA="torch")
+import sys
+specs = sys.argv[1:] or ["baseline:"]
+for spec in specs:
+    name, _, envs = spec.partition(":")
+    extra = dict(kv.split("=") for kv in envs.split(",") if kv)
+    run(name, n=6, **extra)
