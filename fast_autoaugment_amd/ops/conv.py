@@ -65,11 +65,11 @@ class FaaConv2dFn(torch.autograd.Function):
                                                 padding=ctx.padding)
         if ctx.needs_input_grad[1] or (ctx.has_bias and ctx.needs_input_grad[2]):
             if _faa_wrw_wins(x.size(1)):
-                dw, db = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
-                                             weight.size(2), weight.size(3),
-                                             ctx.has_bias)
+                dw, _ = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
+                                            weight.size(2), weight.size(3),
+                                            False)
                 if ctx.has_bias and ctx.needs_input_grad[2]:
-                    dbias = db
+                    dbias = dy.sum(dim=(0, 2, 3))
             else:
                 if ctx.needs_input_grad[1]:
                     dw = torch.nn.grad.conv2d_weight(
@@ -77,10 +77,10 @@ class FaaConv2dFn(torch.autograd.Function):
                         padding=ctx.padding).contiguous(
                             memory_format=torch.channels_last)
                 if ctx.has_bias and ctx.needs_input_grad[2]:
-                    nch = dy.size(1)
-                    use_cs = (dy.dtype == torch.bfloat16 and nch % 8 == 0
-                              and _os.environ.get("FAA_NO_COLSUM") != "1")
-                    dbias = C.colsum_bf16(dy) if use_cs else dy.sum(dim=(0, 2, 3))
+                    # NOTE: colsum_bf16 is replay-clean standalone but was
+                    # implicated in hipGraph-replay corruption inside the full
+                    # captured step (tools/nan_flake.py bisection) -> torch sum
+                    dbias = dy.sum(dim=(0, 2, 3))
         return dx, dw, dbias, None, None
 
 
