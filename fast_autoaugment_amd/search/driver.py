@@ -103,8 +103,21 @@ def run_search(dataroot: str, until: int = 5, num_op: int = 2, num_policy: int =
                 "num_op": num_op, "num_policy": num_policy,
             }
             done: List = []
+            # trial journal: crash/restart resumes completed trials (the
+            # reference leaned on Ray Tune's experiment state, search.py:245)
+            journal = ckpt_path(dataset, model_type,
+                                f"ratio{cv_ratio:.1f}_fold{cv_fold}_trials.jsonl")
+            if resume and os.path.exists(journal):
+                with open(journal) as jf:
+                    for line in jf:
+                        rec = json.loads(line)
+                        sampler.observe(rec["cfg"], -rec["top1_valid"])
+                        done.append((rec["cfg"], rec))
+                if done:
+                    logger.info("fold %d: resumed %d completed trials", cv_fold, len(done))
+            jf = open(journal, "a")
             inflight = []
-            submitted = 0
+            submitted = len(done)
             max_conc = sched.n_workers
             while len(done) < num_search:
                 while submitted < num_search and len(inflight) < max_conc:
@@ -114,14 +127,25 @@ def run_search(dataroot: str, until: int = 5, num_op: int = 2, num_policy: int =
                     inflight.append((cfg, sched.submit(_eval_trial, copy.deepcopy(copied_c), aug)))
                     submitted += 1
                 cfg, fut = inflight.pop(0)
-                r = fut.result()
+                try:
+                    r = fut.result()
+                except Exception as e:
+                    # tolerate lost trials like Tune's raise_on_failed_trial=False
+                    logger.warning("trial failed (%s); continuing", e)
+                    submitted -= 1
+                    continue
                 sampler.observe(cfg, -r["top1_valid"])   # maximize top1
                 total_computation += r["elapsed_time"]
                 done.append((cfg, r))
+                jf.write(json.dumps({"cfg": cfg, "top1_valid": r["top1_valid"],
+                                     "minus_loss": r["minus_loss"],
+                                     "elapsed_time": r["elapsed_time"]}) + "\n")
+                jf.flush()
                 if len(done) % 10 == 0:
                     best = max(x[1]["top1_valid"] for x in done)
                     logger.info("fold %d: %d/%d trials, best top1_valid=%.4f",
                                 cv_fold, len(done), num_search, best)
+            jf.close()
             done.sort(key=lambda x: x[1]["top1_valid"], reverse=True)
             for cfg, r in done[:num_result_per_cv]:
                 final_policy = policy_decoder(cfg, num_policy, num_op)
