@@ -71,8 +71,12 @@ def main():
     from fast_autoaugment_amd.parallel.flat import flatten_module
     from fast_autoaugment_amd.policies import resolve_aug
 
+    model_conf = {"type": args.model}
+    if args.model == "pyramid":
+        # reference confs/pyramid272_cifar.yaml (BASELINE config 5)
+        model_conf.update({"depth": 272, "alpha": 200, "bottleneck": True})
     conf = {
-        "model": {"type": args.model}, "dataset": args.dataset,
+        "model": model_conf, "dataset": args.dataset,
         "aug": "fa_reduced_cifar10", "cutout": 16, "batch": args.batch,
         "epoch": 200, "lr": 0.1,
         "lr_schedule": {"type": "cosine", "warmup": {"multiplier": 1, "epoch": 5}},
@@ -91,7 +95,7 @@ def main():
                        train=True, mean=mean, std=std, cutout=16,
                        seed=rank, out_dtype=out_dtype, prefetch=4)
 
-    model = build_model(conf["model"], nc).to(dev).to(memory_format=torch.channels_last)
+    model = build_model(model_conf, nc).to(dev).to(memory_format=torch.channels_last)
     # pure-bf16 compute: params+grads are bf16 flat views (no autocast cast
     # kernels), fp32 master lives in the fused optimizer
     work_dtype = torch.bfloat16 if (args.dtype == "bf16" and not cpu_mode) else torch.float32
