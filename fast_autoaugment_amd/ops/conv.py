@@ -19,11 +19,12 @@ from . import ext
 
 
 # Measured dispatch rules (profiles/conv_bench_r01.txt + wrw v2 re-measure,
-# MI355X b128): the MFMA fwd/bwd-data kernels beat MIOpen for C<=128 (all
-# WRN-40-2 shapes, 1.2-4x); MIOpen keeps larger channels. The wrw v2 kernel
-# wins on the stem (tiny Cin) and the deep stages (Cin>=128), loses 1.2-2x
-# in between -> shape-conditional.
-_FAA_MAX_CH = 128
+# MI355X b128): the MFMA fwd kernel beats MIOpen up to 160 channels
+# (fwd 160x32x32->160: 249us vs 377us); bwd-data wins to 128 (loses at
+# 160: 226 vs 171); the wrw v2 kernel wins on the stem (tiny Cin) and the
+# deep stages (Cin>=128), loses 1.2-2x in between.
+_FAA_MAX_CH = 160          # fwd (gates module patching)
+_FAA_BWD_DATA_MAX = 128
 
 
 import os as _os
@@ -35,7 +36,7 @@ def _faa_wrw_wins(cin: int) -> bool:
         return True
     if mode == "torch":
         return False
-    return cin < 8 or cin >= 128
+    return cin < 8 or cin == 128
 
 
 class FaaConv2dFn(torch.autograd.Function):
@@ -56,7 +57,9 @@ class FaaConv2dFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=torch.channels_last)
         dx = dw = dbias = None
         if ctx.needs_input_grad[0]:
-            if ctx.stride == 1 and _os.environ.get("FAA_BWD_DATA", "faa") != "torch":
+            if (ctx.stride == 1 and weight.size(0) <= _FAA_BWD_DATA_MAX
+                    and weight.size(1) <= _FAA_BWD_DATA_MAX
+                    and _os.environ.get("FAA_BWD_DATA", "faa") != "torch"):
                 dx = C.conv2d_bwd_data(dy, weight, 1, ctx.padding,
                                        x.size(2), x.size(3))
             else:
