@@ -75,9 +75,11 @@ def main():
     if args.model == "pyramid":
         # reference confs/pyramid272_cifar.yaml (BASELINE config 5)
         model_conf.update({"depth": 272, "alpha": 200, "bottleneck": True})
+    aug_name = "fa_reduced_imagenet" if "imagenet" in args.dataset else "fa_reduced_cifar10"
     conf = {
         "model": model_conf, "dataset": args.dataset,
-        "aug": "fa_reduced_cifar10", "cutout": 16, "batch": args.batch,
+        "aug": aug_name, "cutout": 0 if "imagenet" in args.dataset else 16,
+        "batch": args.batch,
         "epoch": 200, "lr": 0.1,
         "lr_schedule": {"type": "cosine", "warmup": {"multiplier": 1, "epoch": 5}},
         "optimizer": {"type": "sgd", "decay": 0.0002, "nesterov": True, "ema": 0},
@@ -85,15 +87,19 @@ def main():
     C.replace(conf)
 
     nc = num_class(args.dataset)
+    imagenet = "imagenet" in args.dataset
+    img_src = 224 if imagenet else 32        # synthetic source resolution
+    out_size = 224 if imagenet else 32
     # synthetic data, random-init weights (no network on the box)
-    n_synth = int(os.environ.get("FAA_BENCH_IMGS", "50000"))
-    imgs, labels = synthetic_arrays(n_synth, 32, nc, seed=1234 + rank)
+    n_synth = int(os.environ.get("FAA_BENCH_IMGS", "10000" if imagenet else "50000"))
+    imgs, labels = synthetic_arrays(n_synth, img_src, nc, seed=1234 + rank)
     store = TensorStore(imgs, labels, device=str(dev))
     mean, std = dataset_stats(args.dataset)
     out_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
-    loader = AugLoader(store, args.batch, resolve_aug("fa_reduced_cifar10"),
-                       train=True, mean=mean, std=std, cutout=16,
-                       seed=rank, out_dtype=out_dtype, prefetch=4)
+    loader = AugLoader(store, args.batch, resolve_aug(aug_name),
+                       train=True, mean=mean, std=std, cutout=conf["cutout"],
+                       seed=rank, out_dtype=out_dtype, prefetch=4,
+                       imagenet_size=out_size if imagenet else 0)
 
     model = build_model(model_conf, nc).to(dev).to(memory_format=torch.channels_last)
     # pure-bf16 compute: params+grads are bf16 flat views (no autocast cast
@@ -132,14 +138,19 @@ def main():
     import queue as _q
     import threading
     rng = np.random.default_rng(1000 + rank)
-    policy = resolve_aug("fa_reduced_cifar10")
+    policy = resolve_aug(aug_name)
     n_imgs = len(store)
 
     def gen_host():
         sel = rng.integers(0, n_imgs, size=args.batch)
-        prog = aug_ops.compile_program_fast(policy, args.batch, 32, 32, rng)
-        post = aug_ops.compile_post_fast(args.batch, 32, 32, rng, pad=4,
-                                         cutout_len=16, train=True)
+        prog = aug_ops.compile_program_fast(policy, args.batch, img_src, img_src, rng)
+        if imagenet:
+            from fast_autoaugment_amd.aug.imagenet import compile_post_imagenet
+            post = compile_post_imagenet(args.batch, img_src, img_src, rng,
+                                         out_size, train=True)
+        else:
+            post = aug_ops.compile_post_fast(args.batch, img_src, img_src, rng,
+                                             pad=4, cutout_len=16, train=True)
         return sel, prog, post
 
     host_q: "_q.Queue" = _q.Queue(maxsize=6)
@@ -158,7 +169,8 @@ def main():
     sel_s = torch.zeros(args.batch, dtype=torch.int64, device=dev)
     prog_s = torch.zeros((args.batch, aug_ops.PROG_SLOTS, aug_ops.PROG_WIDTH),
                          dtype=torch.float32, device=dev)
-    post_s = torch.zeros((args.batch, 6), dtype=torch.float32, device=dev)
+    post_w = 18 if imagenet else 6
+    post_s = torch.zeros((args.batch, post_w), dtype=torch.float32, device=dev)
     pin = not cpu_mode
     sel_h = torch.zeros_like(sel_s, device="cpu", pin_memory=pin)
     prog_h = torch.zeros_like(prog_s, device="cpu", pin_memory=pin)
@@ -188,7 +200,12 @@ def main():
         if cpu_mode:
             data, label = make_batch_cpu()
         else:
-            data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s, mean_t, std_t, bf16)
+            if imagenet:
+                data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s, post_s,
+                                                mean_t, std_t, out_size, out_size, bf16)
+            else:
+                data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
+                                       mean_t, std_t, bf16)
             label = store.labels.index_select(0, sel_s)
         preds = model(data)
         loss = crit(preds, label)
@@ -336,7 +353,8 @@ def main():
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {"model": args.model, "global_batch": args.batch * world_size,
-                       "image": "32x32", "aug": "fa_reduced_cifar10+cutout16",
+                       "image": f"{out_size}x{out_size}",
+                       "aug": aug_name + ("" if imagenet else "+cutout16"),
                        "parallelism": f"dp{world_size}"},
         }))
 
