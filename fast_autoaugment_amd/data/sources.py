@@ -89,6 +89,11 @@ def load_dataset_arrays(dataset: str, dataroot: str, train: bool = True,
                     ex = _load_svhn(dataroot, "extra")
                     return (np.concatenate([tr[0], ex[0]]), np.concatenate([tr[1], ex[1]]))
                 return tr if train else _load_svhn(dataroot, "test")
+            if base == "imagenet":
+                from .imagenet_folder import load_imagenet_folder
+                root = os.path.join(dataroot, "imagenet-pytorch")
+                return load_imagenet_folder(root, "train" if train else "val",
+                                            resize_short=256)
         except (FileNotFoundError, OSError):
             if synthetic == "never":
                 raise

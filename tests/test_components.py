@@ -270,3 +270,29 @@ def test_trainer_ema_rmsprop_cpu(tmp_path):
         os.environ.pop("FAA_SYNTH_TEST", None)
         from fast_autoaugment_amd.data import api as data_api
         data_api._STORE_CACHE.clear()
+
+
+def test_imagenet_folder_loader(tmp_path):
+    """ImageFolder-style loading with listfile fast path (reference
+    imagenet.py:60-88 layout), PIL decode + cache."""
+    import PIL.Image
+    from fast_autoaugment_amd.data.imagenet_folder import load_imagenet_folder
+    root = tmp_path / "imagenet-pytorch"
+    rng = np.random.default_rng(0)
+    rels = []
+    for ci, wnid in enumerate(["n01440764", "n01443537"]):
+        d = root / "train" / wnid
+        d.mkdir(parents=True)
+        for j in range(2):
+            arr = rng.integers(0, 255, (80, 100, 3), dtype=np.uint8)
+            PIL.Image.fromarray(arr).save(d / f"img{j}.JPEG")
+            rels.append((f"{wnid}/img{j}.JPEG", ci))
+    with open(root / "train_cls.txt", "w") as f:
+        for rel, c in rels:
+            f.write(f"{rel} {c}\n")
+    imgs, labels = load_imagenet_folder(str(root), "train", resize_short=64)
+    assert imgs.shape == (4, 64, 64, 3)
+    assert labels.tolist() == [0, 0, 1, 1]
+    # cache hit path
+    imgs2, labels2 = load_imagenet_folder(str(root), "train", resize_short=64)
+    np.testing.assert_array_equal(imgs, imgs2)
