@@ -83,8 +83,8 @@ def run_epoch(model, loader, loss_fn, optimizer, desc_default="", epoch=0,
             loss.backward()
             if hasattr(model, "finish_gradient_sync"):
                 model.finish_gradient_sync()
-            from ..optim import FusedSGD
-            if not isinstance(optimizer, FusedSGD):
+            from ..optim import FusedSGD, FusedRMSpropTF
+            if not isinstance(optimizer, (FusedSGD, FusedRMSpropTF)):
                 # FusedSGD folds manual WD + global clip into its kernels
                 if wd > 0.0:
                     _apply_manual_wd(decay_params, wd)
@@ -157,8 +157,8 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
         multinode=(local_rank >= 0), rank=rank, world_size=world_size,
         device=device, out_dtype=out_dtype)
 
-    # Fast path: pure-bf16 flat weights + fused SGD + MFMA convs on GPU.
-    use_fast = (use_cuda and conf["optimizer"]["type"] == "sgd"
+    # Fast path: pure-bf16 flat weights + fused optimizer + MFMA convs on GPU.
+    use_fast = (use_cuda and conf["optimizer"]["type"] in ("sgd", "rmsprop")
                 and autocast_dtype == torch.bfloat16
                 and conf.get_value("pure_bf16", True))
     work_dtype = torch.bfloat16 if use_fast else None
@@ -183,11 +183,18 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
         flat = model.flat if hasattr(model, "flat") else \
             flatten_module(model, work_dtype=torch.bfloat16)
         patch_convs(model)
-        optimizer = FusedSGD(flat, lr=conf["lr"],
-                             momentum=conf["optimizer"].get("momentum", 0.9),
-                             nesterov=conf["optimizer"].get("nesterov", True),
-                             weight_decay=conf["optimizer"].get("decay", 0.0),
-                             grad_clip=conf["optimizer"].get("clip", 5.0))
+        if conf["optimizer"]["type"] == "sgd":
+            optimizer = FusedSGD(flat, lr=conf["lr"],
+                                 momentum=conf["optimizer"].get("momentum", 0.9),
+                                 nesterov=conf["optimizer"].get("nesterov", True),
+                                 weight_decay=conf["optimizer"].get("decay", 0.0),
+                                 grad_clip=conf["optimizer"].get("clip", 5.0))
+        else:
+            from ..optim import FusedRMSpropTF
+            optimizer = FusedRMSpropTF(flat, lr=conf["lr"], alpha=0.9,
+                                       momentum=0.9, eps=0.001,
+                                       weight_decay=conf["optimizer"].get("decay", 0.0),
+                                       grad_clip=conf["optimizer"].get("clip", 0.0))
         autocast_dtype = None      # the model computes natively in bf16
     else:
         optimizer = build_optimizer(conf["optimizer"], model.parameters(), conf["lr"])

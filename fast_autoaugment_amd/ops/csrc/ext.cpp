@@ -19,6 +19,10 @@ void sgd_fused_step_mixed(torch::Tensor master, torch::Tensor work, torch::Tenso
                           int64_t nesterov);
 void ema_lerp_(torch::Tensor shadow, torch::Tensor x, double mu);
 void gather_grads(torch::Tensor table, torch::Tensor flat);
+void rmsprop_fused_step_mixed(torch::Tensor master, torch::Tensor work, torch::Tensor g,
+                              torch::Tensor ms, torch::Tensor mom, torch::Tensor normsq,
+                              torch::Tensor lr_t, int64_t n_decay, double wd,
+                              double clip, double rho, double momentum, double eps);
 torch::Tensor aug_pipeline(torch::Tensor images, torch::Tensor sel, torch::Tensor prog,
                            torch::Tensor post, torch::Tensor mean, torch::Tensor std,
                            bool bf16_out);
@@ -57,6 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_fused_step_mixed", &sgd_fused_step_mixed,
         "mixed bf16-work/fp32-master fused SGD step");
   m.def("ema_lerp_", &ema_lerp_);
+  m.def("rmsprop_fused_step_mixed", &rmsprop_fused_step_mixed,
+        "fused TF-semantics RMSprop on flat mixed-precision buffers");
   m.def("gather_grads", &gather_grads,
         "pack scattered autograd grads into the flat bf16 buffer");
   m.def("aug_pipeline_imagenet", &aug_pipeline_imagenet,

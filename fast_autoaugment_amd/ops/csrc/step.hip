@@ -234,7 +234,65 @@ __global__ void gather_grads_kernel(const int64_t* __restrict__ table, int n,
   }
 }
 
+// ---- fused RMSpropTF step (reference tf_port/rmsprop.py:80-99 semantics:
+// ms initialized to ONES by the host, eps INSIDE the sqrt), mixed bf16
+// working weights + fp32 master/ms/mom, manual WD on the decay segment,
+// optional global clip via the shared norm pass.
+__global__ void rmsprop_step_mixed_kernel(float* __restrict__ p, short* __restrict__ w,
+                                          const short* __restrict__ g,
+                                          float* __restrict__ ms, float* __restrict__ mom,
+                                          const float* __restrict__ normsq,
+                                          const float* __restrict__ lr_p,
+                                          int64_t n, int64_t n_decay, float wd,
+                                          float clip, float rho, float mu, float eps) {
+  float lr = lr_p[0];
+  float coef = 1.0f;
+  if (clip > 0.0f) {
+    float norm = sqrtf(normsq[0]);
+    coef = fminf(1.0f, clip / (norm + 1e-6f));
+  }
+  int64_t i0 = (int64_t)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i = i0; i < n; i += stride) {
+    int64_t lim = min(i + 4, n);
+    for (int64_t j = i; j < lim; ++j) {
+      float gg = bits_to_f32(g[j]);
+      if (wd != 0.0f && j < n_decay) gg += wd * p[j];
+      gg *= coef;
+      float m = ms[j] + (gg * gg - ms[j]) * (1.0f - rho);
+      ms[j] = m;
+      float mo = mu * mom[j] + lr * gg * rsqrtf(m + eps);
+      mom[j] = mo;
+      p[j] -= mo;
+      __hip_bfloat16 h = __float2bfloat16(p[j]);
+      w[j] = *reinterpret_cast<short*>(&h);
+    }
+  }
+}
+
 }  // namespace
+
+void rmsprop_fused_step_mixed(torch::Tensor master, torch::Tensor work, torch::Tensor g,
+                              torch::Tensor ms, torch::Tensor mom, torch::Tensor normsq,
+                              torch::Tensor lr_t, int64_t n_decay, double wd,
+                              double clip, double rho, double momentum, double eps) {
+  int64_t n = master.numel();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const int block = 256;
+  int grid = faa_grid(n / 4 + 1, block);
+  if (clip > 0.0) {
+    hipLaunchKernelGGL(zero1_kernel, dim3(1), dim3(64), 0, stream, normsq.data_ptr<float>());
+    hipLaunchKernelGGL(wd_norm_mixed_kernel, dim3(grid), dim3(block), 0, stream,
+                       (const short*)g.data_ptr(), master.data_ptr<float>(),
+                       normsq.data_ptr<float>(), n, n_decay, (float)wd);
+  }
+  hipLaunchKernelGGL(rmsprop_step_mixed_kernel, dim3(grid), dim3(block), 0, stream,
+                     master.data_ptr<float>(), (short*)work.data_ptr(),
+                     (const short*)g.data_ptr(), ms.data_ptr<float>(),
+                     mom.data_ptr<float>(), normsq.data_ptr<float>(),
+                     lr_t.data_ptr<float>(), n, n_decay, (float)wd, (float)clip,
+                     (float)rho, (float)momentum, (float)eps);
+}
 
 void gather_grads(torch::Tensor table, torch::Tensor flat) {
   TORCH_CHECK(table.dtype() == torch::kInt64 && table.is_cuda());

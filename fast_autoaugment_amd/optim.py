@@ -167,3 +167,75 @@ def build_optimizer(conf_opt, model_params, lr: float):
         return RMSpropTF(model_params, lr=lr, weight_decay=0.0,
                          alpha=0.9, momentum=0.9, eps=0.001)
     raise ValueError(f"invalid optimizer type={typ}")
+
+
+class FusedRMSpropTF:
+    """Flat-buffer TF-semantics RMSprop (reference tf_port/rmsprop.py) on
+    bf16 working weights + fp32 master, with manual non-BN WD and optional
+    global clip — the EfficientNet training step as 1-2 HIP kernels."""
+
+    def __init__(self, flat, lr: float, alpha: float = 0.9, momentum: float = 0.9,
+                 eps: float = 1e-3, weight_decay: float = 0.0, grad_clip: float = 0.0):
+        assert flat.flat_master is not None, "FusedRMSpropTF needs bf16 flat mode"
+        self.flat = flat
+        self.lr = lr
+        self.alpha = alpha
+        self.momentum = momentum
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.grad_clip = grad_clip
+        dev = flat.flat_param.device
+        self.ms = torch.ones_like(flat.flat_master)      # TF init (rmsprop.py:80)
+        self.mom = torch.zeros_like(flat.flat_master)
+        self._normsq = torch.zeros(1, device=dev, dtype=torch.float32)
+        self._lr_t = torch.full((1,), lr, device=dev, dtype=torch.float32)
+        self._lr_host = (torch.zeros(1, dtype=torch.float32, pin_memory=True)
+                         if dev.type == "cuda" else torch.zeros(1))
+        self._last_lr = lr
+        self.param_groups = [{"lr": lr}]
+
+    def sync_lr(self):
+        lr = self.param_groups[0]["lr"]
+        if lr != self._last_lr:
+            self._lr_host[0] = lr
+            self._lr_t.copy_(self._lr_host, non_blocking=True)
+            self._last_lr = lr
+
+    def zero_grad(self, set_to_none: bool = False):
+        self.flat.flat_grad.zero_()
+
+    @torch.no_grad()
+    def step(self):
+        self.lr = self.param_groups[0]["lr"]
+        master = self.flat.flat_master
+        g = self.flat.flat_grad
+        nd = self.flat.n_decay
+        if master.is_cuda:
+            from .ops import ext
+            self.sync_lr()
+            ext().rmsprop_fused_step_mixed(master, self.flat.flat_param, g,
+                                           self.ms, self.mom, self._normsq,
+                                           self._lr_t, nd, self.weight_decay,
+                                           self.grad_clip, self.alpha,
+                                           self.momentum, self.eps)
+            return
+        # CPU reference path (defines the kernel's semantics)
+        gf = g.float()
+        if self.weight_decay > 0 and nd > 0:
+            gf[:nd].add_(master[:nd], alpha=self.weight_decay)
+        if self.grad_clip > 0:
+            coef = min(1.0, self.grad_clip / (float(gf.norm(2)) + 1e-6))
+            if coef < 1.0:
+                gf.mul_(coef)
+        self.ms.add_((gf * gf - self.ms) * (1.0 - self.alpha))
+        self.mom.mul_(self.momentum).addcdiv_(gf, (self.ms + self.eps).sqrt(),
+                                              value=self.lr)
+        master.add_(self.mom, alpha=-1.0)
+        self.flat.flat_param.copy_(master.to(self.flat.flat_param.dtype))
+
+    def state_dict(self):
+        return {"ms": self.ms, "mom": self.mom, "lr": self.lr}
+
+    def load_state_dict(self, sd):
+        self.ms.copy_(sd["ms"])
+        self.mom.copy_(sd["mom"])

@@ -152,3 +152,28 @@ def test_flat_optimizer_equivalence():
             o.step()
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p1, p2, atol=1e-6)
+
+
+def test_fused_rmsprop_tf_cpu_matches_reference():
+    """FusedRMSpropTF flat step vs the per-tensor RMSpropTF (same math)."""
+    import torch as T
+    from fast_autoaugment_amd.optim import FusedRMSpropTF, RMSpropTF
+    from fast_autoaugment_amd.parallel.flat import flatten_module
+    T.manual_seed(0)
+    m1 = T.nn.Sequential(T.nn.Conv2d(3, 8, 3, padding=1), T.nn.Conv2d(8, 4, 1))
+    m2 = T.nn.Sequential(T.nn.Conv2d(3, 8, 3, padding=1), T.nn.Conv2d(8, 4, 1))
+    m2.load_state_dict(m1.state_dict())
+    flat = flatten_module(m1, work_dtype=T.bfloat16)
+    o1 = FusedRMSpropTF(flat, lr=0.01, weight_decay=0.0, grad_clip=0.0)
+    o2 = RMSpropTF(m2.parameters(), lr=0.01, alpha=0.9, momentum=0.9, eps=1e-3)
+    x = T.randn(2, 3, 8, 8)
+    for _ in range(3):
+        o1.zero_grad()
+        m1(x.bfloat16()).float().square().mean().backward()
+        o1.step()
+        o2.zero_grad()
+        m2(x).square().mean().backward()
+        o2.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        # bf16 grads vs fp32 grads: loose tolerance
+        assert (p1.float() - p2).abs().max().item() < 5e-2
