@@ -391,3 +391,24 @@ def test_aug_pipeline_imagenet_eval_centercrop():
                                   OS, OS, False)
     got = out.permute(0, 2, 3, 1).contiguous().cpu().numpy()
     assert np.abs(got - ref).max() * 255 < 1.5
+
+
+def test_fused_rmsprop_gpu_matches_cpu():
+    from fast_autoaugment_amd.optim import FusedRMSpropTF
+    from fast_autoaugment_amd.parallel.flat import flatten_module
+    torch.manual_seed(0)
+    mk = lambda: torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3, padding=1),
+                                     torch.nn.Conv2d(8, 8, 1))
+    mg, mc = mk().to(dev()), mk()
+    mc.load_state_dict(mg.state_dict())
+    fg = flatten_module(mg, work_dtype=torch.bfloat16)
+    fc = flatten_module(mc, work_dtype=torch.bfloat16)
+    og = FusedRMSpropTF(fg, lr=0.01, weight_decay=1e-5, grad_clip=2.0)
+    oc = FusedRMSpropTF(fc, lr=0.01, weight_decay=1e-5, grad_clip=2.0)
+    x = torch.randn(4, 3, 8, 8)
+    for _ in range(3):
+        for m, o, xx in [(mg, og, x.to(dev()).bfloat16()), (mc, oc, x.bfloat16())]:
+            o.zero_grad()
+            m(xx).float().square().mean().backward()
+            o.step()
+    assert (fg.flat_master.cpu() - fc.flat_master).abs().max().item() < 5e-3
