@@ -247,6 +247,7 @@ def main():
         graph = torch.cuda.CUDAGraph()
         upload_next()
         if gather_mode:
+            saved_grads = [p_.grad for p_ in flat.params]
             for p_ in flat.params:
                 p_.grad = None          # capture assignment-mode backward
             with torch.cuda.graph(graph):
@@ -254,19 +255,41 @@ def main():
             # grads now live in the graph pool at replay-stable addresses
             base = flat.flat_param.data_ptr()
             rows = []
+            ok = True
             for p_ in flat.params:
                 g_ = p_.grad
-                assert g_ is not None and g_.dtype == torch.bfloat16
-                # layout must match the flat region's element order
+                if g_ is None or g_.dtype != torch.bfloat16:
+                    ok = False
+                    break
+                # element order must match the flat region's (channels_last
+                # for 4D; for I==1 or 1x1 kernels both layouts coincide)
                 if g_.dim() == 4:
-                    assert g_.is_contiguous(memory_format=torch.channels_last), \
-                        f"grad layout mismatch for {tuple(g_.shape)}"
+                    order_ok = (g_.is_contiguous(memory_format=torch.channels_last)
+                                or (g_.is_contiguous()
+                                    and (g_.size(1) == 1 or g_.size(2) * g_.size(3) == 1)))
                 else:
-                    assert g_.is_contiguous()
+                    order_ok = g_.is_contiguous()
+                if not order_ok:
+                    ok = False
+                    break
                 off = (p_.data.data_ptr() - base) // 2
                 rows.append([g_.data_ptr(), off, g_.numel()])
-            gather_table = torch.tensor(rows, dtype=torch.int64, device=dev)
-            flat.flat_grad.zero_()      # pad gaps stay zero forever
+            if ok:
+                gather_table = torch.tensor(rows, dtype=torch.int64, device=dev)
+                flat.flat_grad.zero_()      # pad gaps stay zero forever
+            else:
+                # layout-incompatible grads (e.g. exotic modules): fall back
+                # to flat accumulation and re-capture the full step
+                print("# gather mode unavailable (grad layout); flat mode",
+                      flush=True)
+                gather_mode = False
+                for p_, g_ in zip(flat.params, saved_grads):
+                    p_.grad = g_
+                torch.cuda.synchronize()
+                graph = torch.cuda.CUDAGraph()
+                upload_next()
+                with torch.cuda.graph(graph):
+                    step_body()
         else:
             with torch.cuda.graph(graph):
                 step_body()
