@@ -15,6 +15,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.functional import ShakeDrop
+from ..ops.modules import bn_only, bn_relu
 
 
 def _pad_add(out: torch.Tensor, shortcut: torch.Tensor) -> torch.Tensor:
@@ -45,9 +46,9 @@ class BasicBlock(nn.Module):
         self.shake_drop = ShakeDrop(p_shakedrop)
 
     def forward(self, x):
-        out = self.conv1(self.bn1(x))
-        out = self.conv2(self.relu(self.bn2(out)))
-        out = self.bn3(out)
+        out = self.conv1(bn_only(x, self.bn1))
+        out = self.conv2(bn_relu(out, self.bn2))
+        out = bn_only(out, self.bn3)
         out = self.shake_drop(out)
         shortcut = self.downsample(x) if self.downsample is not None else x
         return _pad_add(out, shortcut)
@@ -70,10 +71,10 @@ class Bottleneck(nn.Module):
         self.shake_drop = ShakeDrop(p_shakedrop)
 
     def forward(self, x):
-        out = self.conv1(self.bn1(x))
-        out = self.conv2(self.relu(self.bn2(out)))
-        out = self.conv3(self.relu(self.bn3(out)))
-        out = self.bn4(out)
+        out = self.conv1(bn_only(x, self.bn1))
+        out = self.conv2(bn_relu(out, self.bn2))
+        out = self.conv3(bn_relu(out, self.bn3))
+        out = bn_only(out, self.bn4)
         out = self.shake_drop(out)
         shortcut = self.downsample(x) if self.downsample is not None else x
         return _pad_add(out, shortcut)
@@ -135,10 +136,10 @@ class PyramidNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.bn1(self.conv1(x))
+        x = bn_only(self.conv1(x), self.bn1)
         x = self.layer1(x)
         x = self.layer2(x)
         x = self.layer3(x)
-        x = self.relu_final(self.bn_final(x))
+        x = bn_relu(x, self.bn_final)
         x = self.avgpool(x).flatten(1)
         return self.fc(x)

@@ -10,6 +10,8 @@ import math
 
 import torch.nn as nn
 
+from ..ops.modules import bn_only, bn_relu
+
 _BLOCKS = {18: "basic", 34: "basic", 50: "bottleneck", 101: "bottleneck",
            152: "bottleneck", 200: "bottleneck"}
 _LAYERS = {18: [2, 2, 2, 2], 34: [3, 4, 6, 3], 50: [3, 4, 6, 3],
@@ -29,9 +31,12 @@ class BasicBlock(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        identity = x if self.downsample is None else self.downsample(x)
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
+        if self.downsample is None:
+            identity = x
+        else:
+            identity = bn_only(self.downsample[0](x), self.downsample[1])
+        out = bn_relu(self.conv1(x), self.bn1)
+        out = bn_only(self.conv2(out), self.bn2)
         return self.relu(out + identity)
 
 
@@ -50,10 +55,13 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        identity = x if self.downsample is None else self.downsample(x)
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
+        if self.downsample is None:
+            identity = x
+        else:
+            identity = bn_only(self.downsample[0](x), self.downsample[1])
+        out = bn_relu(self.conv1(x), self.bn1)
+        out = bn_relu(self.conv2(out), self.bn2)
+        out = bn_only(self.conv3(out), self.bn3)
         return self.relu(out + identity)
 
 
@@ -111,7 +119,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.relu(self.bn1(self.conv1(x)))
+        x = bn_relu(self.conv1(x), self.bn1)
         if self.dataset == "imagenet":
             x = self.maxpool(x)
         x = self.layer1(x)

@@ -14,6 +14,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.functional import shake_shake
+from ..ops.modules import bn_only, bn_relu
 
 
 class Shortcut(nn.Module):
@@ -50,8 +51,17 @@ class ShakeBlock(nn.Module):
             nn.BatchNorm2d(out_ch),
         )
 
+    @staticmethod
+    def _run_branch(b, x):
+        # b = [ReLU, conv, BN, ReLU, conv, BN]: fuse BN(+ReLU) pairs
+        h = b[1](b[0](x))
+        h = bn_relu(h, b[2])
+        h = b[4](h)
+        return bn_only(h, b[5])
+
     def forward(self, x):
-        h = shake_shake(self.branch1(x), self.branch2(x), self.training)
+        h = shake_shake(self._run_branch(self.branch1, x),
+                        self._run_branch(self.branch2, x), self.training)
         h0 = x if self.equal_io else self.shortcut(x)
         return h + h0
 

@@ -8,23 +8,26 @@ from . import ext
 
 class FusedBNReLUFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum, eps):
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training,
+                momentum, eps, relu):
         C = ext()
         out, mean, invstd = C.bn_relu_fwd(x, gamma, beta, running_mean, running_var,
-                                          training, momentum, eps)
+                                          training, momentum, eps, relu)
         ctx.save_for_backward(x, out, mean, invstd, gamma)
         ctx.training = training
+        ctx.relu = relu
         return out
 
     @staticmethod
     def backward(ctx, dy):
         x, out, mean, invstd, gamma = ctx.saved_tensors
         C = ext()
-        dx, dgamma, dbeta = C.bn_relu_bwd(dy, x, out, mean, invstd, gamma, ctx.training)
-        return dx, dgamma, dbeta, None, None, None, None, None
+        dx, dgamma, dbeta = C.bn_relu_bwd(dy, x, out, mean, invstd, gamma,
+                                          ctx.training, ctx.relu)
+        return dx, dgamma, dbeta, None, None, None, None, None, None
 
 
-def fused_bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
+def _fused(x, bn, relu):
     training = bn.training
     if training and bn.num_batches_tracked is not None:
         # a per-call .add_(1) is a 5us GPU kernel x 37 BN sites x step; count
@@ -32,7 +35,15 @@ def fused_bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
         bn._faa_nbt_pending = getattr(bn, "_faa_nbt_pending", 0) + 1
     momentum = bn.momentum if bn.momentum is not None else 0.1
     return FusedBNReLUFn.apply(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
-                               training, momentum, bn.eps)
+                               training, momentum, bn.eps, relu)
+
+
+def fused_bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
+    return _fused(x, bn, True)
+
+
+def fused_bn(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
+    return _fused(x, bn, False)
 
 
 def sync_bn_trackers(model: torch.nn.Module) -> None:

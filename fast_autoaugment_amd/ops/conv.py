@@ -115,7 +115,68 @@ def patch_convs(model: torch.nn.Module) -> int:
         return 0
     n = 0
     for m in model.modules():
-        if isinstance(m, torch.nn.Conv2d) and _eligible(m):
+        if not isinstance(m, torch.nn.Conv2d):
+            continue
+        if _dw_eligible(m):
+            m.forward = types.MethodType(_faa_dw_forward, m)
+            n += 1
+        elif _eligible(m):
             m.forward = types.MethodType(_faa_forward, m)
             n += 1
     return n
+
+
+# ------------------------------------------------------------- depthwise
+
+class FaaDepthwiseFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, pads):
+        C = ext()
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        ctx.pads = pads            # (pl, pr, pt, pb)
+        ctx.has_bias = bias is not None
+        pl, pr, pt, pb = pads
+        b = bias if bias is not None else torch.Tensor()
+        return C.dwconv_fwd(x, weight, b, stride, pt, pb, pl, pr)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        C = ext()
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        pl, pr, pt, pb = ctx.pads
+        dx = dw = dbias = None
+        if ctx.needs_input_grad[0]:
+            dx = C.dwconv_bwd_data(dy, weight, ctx.stride, pt, pl,
+                                   x.size(2), x.size(3))
+        if ctx.needs_input_grad[1]:
+            dw = C.dwconv_bwd_weight(dy, x, ctx.stride, pt, pl,
+                                     weight.size(2), weight.size(3))
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            dbias = dy.sum(dim=(0, 2, 3))
+        return dx, dw, dbias, None, None
+
+
+def _dw_eligible(m: torch.nn.Conv2d) -> bool:
+    k = m.kernel_size
+    return (m.groups == m.in_channels == m.out_channels
+            and m.in_channels % 8 == 0
+            and k[0] == k[1] and k[0] in (3, 5)
+            and m.stride[0] == m.stride[1] and m.stride[0] in (1, 2)
+            and m.dilation == (1, 1))
+
+
+def _faa_dw_forward(self, x):
+    if x.is_cuda and x.dtype == torch.bfloat16 and self.weight.dtype == torch.bfloat16:
+        pads = getattr(self, "_pad", None)   # Conv2dSamePadding asymmetric pads
+        if pads is None:
+            p = self.padding[0]
+            pads = (p, p, p, p)
+        return FaaDepthwiseFn.apply(x, self.weight, self.bias, self.stride[0], pads)
+    # original semantics (SAME padding modules pad explicitly)
+    pads = getattr(self, "_pad", None)
+    if pads is not None and any(pads):
+        x = F.pad(x, pads)
+    return F.conv2d(x, self.weight, self.bias, self.stride, self.padding,
+                    self.dilation, self.groups)

@@ -141,7 +141,7 @@ __device__ __forceinline__ short f_to_bf16_bits(float f) {
   return *reinterpret_cast<short*>(&h);
 }
 
-template <typename T, typename GT>
+template <typename T, typename GT, bool RELU>
 __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
@@ -164,8 +164,8 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
           int cb = c0 + k + 4; while (cb >= C) cb -= C;
           float a = (bf16_bits_to_f(v0[k]) - mean[ca]) * invstd[ca] * faa_to_float(gamma[ca]) + faa_to_float(beta[ca]);
           float b = (bf16_bits_to_f(v1[k]) - mean[cb]) * invstd[cb] * faa_to_float(gamma[cb]) + faa_to_float(beta[cb]);
-          o0[k] = f_to_bf16_bits(fmaxf(a, 0.0f));
-          o1[k] = f_to_bf16_bits(fmaxf(b, 0.0f));
+          o0[k] = f_to_bf16_bits(RELU ? fmaxf(a, 0.0f) : a);
+          o1[k] = f_to_bf16_bits(RELU ? fmaxf(b, 0.0f) : b);
         }
         short* os = reinterpret_cast<short*>(out);
         *reinterpret_cast<short4v*>(os + i) = o0;
@@ -179,14 +179,14 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
       if (j < total) {
         int c = c0 + k; while (c >= C) c -= C;
         float v = (faa_to_float(x[j]) - mean[c]) * invstd[c] * faa_to_float(gamma[c]) + faa_to_float(beta[c]);
-        faa_from_float(fmaxf(v, 0.0f), &out[j]);
+        faa_from_float(RELU ? fmaxf(v, 0.0f) : v, &out[j]);
       }
     }
   }
 }
 
 // ---------------------------------------------------------------- bwd reduce
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __restrict__ out,
                                          const T* __restrict__ dy,
                                          const float* __restrict__ mean,
@@ -213,8 +213,8 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
       short4v dv1 = *reinterpret_cast<const short4v*>(ds + i + 4);
       #pragma unroll
       for (int k = 0; k < 4; ++k) {
-        float g0 = bf16_bits_to_f(ov0[k]) > 0.0f ? bf16_bits_to_f(dv0[k]) : 0.0f;
-        float g1 = bf16_bits_to_f(ov1[k]) > 0.0f ? bf16_bits_to_f(dv1[k]) : 0.0f;
+        float g0 = (!RELU || bf16_bits_to_f(ov0[k]) > 0.0f) ? bf16_bits_to_f(dv0[k]) : 0.0f;
+        float g1 = (!RELU || bf16_bits_to_f(ov1[k]) > 0.0f) ? bf16_bits_to_f(dv1[k]) : 0.0f;
         sdy[k] += g0;
         sdyx[k] += g0 * (bf16_bits_to_f(xv0[k]) - m[k]) * is[k];
         sdy[k + 4] += g1;
@@ -225,7 +225,7 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
     for (int64_t i = i0; i < total; i += stride) {
       #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = faa_to_float(out[i + k]) > 0.0f ? faa_to_float(dy[i + k]) : 0.0f;
+        float g = (!RELU || faa_to_float(out[i + k]) > 0.0f) ? faa_to_float(dy[i + k]) : 0.0f;
         sdy[k] += g;
         sdyx[k] += g * (faa_to_float(x[i + k]) - m[k]) * is[k];
       }
@@ -258,7 +258,7 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
   }
 }
 
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_bwd_reduce_scalar_kernel(const T* __restrict__ x, const T* __restrict__ out,
                                             const T* __restrict__ dy,
                                             const float* __restrict__ mean,
@@ -269,7 +269,7 @@ __global__ void bn_bwd_reduce_scalar_kernel(const T* __restrict__ x, const T* __
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = i0; i < total; i += stride) {
     int c = (int)(i % C);
-    float g = faa_to_float(out[i]) > 0.0f ? faa_to_float(dy[i]) : 0.0f;
+    float g = (!RELU || faa_to_float(out[i]) > 0.0f) ? faa_to_float(dy[i]) : 0.0f;
     atomicAdd(&scratch[c], g);
     atomicAdd(&scratch[C + c], g * (faa_to_float(x[i]) - mean[c]) * invstd[c]);
   }
@@ -294,7 +294,7 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ scratch, int nb
   }
 }
 
-template <typename T, typename GT>
+template <typename T, typename GT, bool RELU>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict__ out,
                                     const T* __restrict__ dy, T* __restrict__ dx,
                                     const float* __restrict__ mean,
@@ -323,8 +323,8 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
         for (int k = 0; k < 4; ++k) {
           int ca = c0 + k; while (ca >= C) ca -= C;
           int cb = c0 + k + 4; while (cb >= C) cb -= C;
-          float ga = bf16_bits_to_f(ov0[k]) > 0.0f ? bf16_bits_to_f(dv0[k]) : 0.0f;
-          float gb = bf16_bits_to_f(ov1[k]) > 0.0f ? bf16_bits_to_f(dv1[k]) : 0.0f;
+          float ga = (!RELU || bf16_bits_to_f(ov0[k]) > 0.0f) ? bf16_bits_to_f(dv0[k]) : 0.0f;
+          float gb = (!RELU || bf16_bits_to_f(ov1[k]) > 0.0f) ? bf16_bits_to_f(dv1[k]) : 0.0f;
           float isa = invstd[ca], isb = invstd[cb];
           float ra, rb;
           if (training) {
@@ -350,7 +350,7 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
       int64_t j = i + k;
       if (j < total) {
         int c = c0 + k; while (c >= C) c -= C;
-        float mask = faa_to_float(out[j]) > 0.0f ? 1.0f : 0.0f;
+        float mask = (!RELU || faa_to_float(out[j]) > 0.0f) ? 1.0f : 0.0f;
         float g = faa_to_float(dy[j]) * mask;
         float is = invstd[c];
         float res;
@@ -389,7 +389,7 @@ static int bn_nblocks(int C, int64_t total) {
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
-                                       double momentum, double eps) {
+                                       double momentum, double eps, bool relu) {
   TORCH_CHECK(x.dim() == 4, "bn_relu: 4-D input expected");
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   int C = xc.size(1);
@@ -443,11 +443,18 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   DISPATCH_FB(xc.scalar_type(), "bn_apply", [&] {
     using data_t = scalar_t;
     DISPATCH_FB(g.scalar_type(), "bn_apply_g", [&] {
-      hipLaunchKernelGGL((bn_apply_kernel<data_t, scalar_t>), dim3(grid1), dim3(256), 0,
-                         stream, (const data_t*)xc.data_ptr(), (data_t*)out.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         (const scalar_t*)g.data_ptr(), (const scalar_t*)bta.data_ptr(),
-                         total, C);
+      if (relu)
+        hipLaunchKernelGGL((bn_apply_kernel<data_t, scalar_t, true>), dim3(grid1), dim3(256), 0,
+                           stream, (const data_t*)xc.data_ptr(), (data_t*)out.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           (const scalar_t*)g.data_ptr(), (const scalar_t*)bta.data_ptr(),
+                           total, C);
+      else
+        hipLaunchKernelGGL((bn_apply_kernel<data_t, scalar_t, false>), dim3(grid1), dim3(256), 0,
+                           stream, (const data_t*)xc.data_ptr(), (data_t*)out.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           (const scalar_t*)g.data_ptr(), (const scalar_t*)bta.data_ptr(),
+                           total, C);
     });
   });
   return {out, mean, invstd};
@@ -456,7 +463,7 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor out, torch::Tensor mean,
                                        torch::Tensor invstd, torch::Tensor gamma,
-                                       bool training) {
+                                       bool training, bool relu) {
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   auto oc = out.contiguous(torch::MemoryFormat::ChannelsLast);
   auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
@@ -476,10 +483,16 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
     int nb = bn_nblocks(C, total);
     auto scratch = torch::empty({nb, 2 * C}, f32);
     DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce", [&] {
-      hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<scalar_t>), dim3(nb), dim3(256), 0,
-                         stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
-                         (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
+      if (relu)
+        hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<scalar_t, true>), dim3(nb), dim3(256), 0,
+                           stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
+                           (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
+      else
+        hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<scalar_t, false>), dim3(nb), dim3(256), 0,
+                           stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
+                           (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
     });
     DISPATCH_FB(g.scalar_type(), "bn_bwd_fin", [&] {
       hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
@@ -490,7 +503,13 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
     sums.zero_();
     int grid = faa_grid(total, 256, 1024);
     DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce_s", [&] {
-      hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<scalar_t>), dim3(grid), dim3(256), 0,
+      if (!relu)
+        hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<scalar_t, false>), dim3(grid), dim3(256), 0,
+                           stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
+                           (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), sums.data_ptr<float>(), total, C);
+      else
+        hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<scalar_t, true>), dim3(grid), dim3(256), 0,
                          stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
                          (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), sums.data_ptr<float>(), total, C);
@@ -502,12 +521,20 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
   DISPATCH_FB(xc.scalar_type(), "bn_bwd_apply", [&] {
     using data_t = scalar_t;
     DISPATCH_FB(g.scalar_type(), "bn_bwd_apply_g", [&] {
-      hipLaunchKernelGGL((bn_bwd_apply_kernel<data_t, scalar_t>), dim3(grid1), dim3(256), 0,
-                         stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
-                         (const data_t*)dyc.data_ptr(), (data_t*)dx.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         (const scalar_t*)g.data_ptr(), sums.data_ptr<float>(),
-                         total, C, rows, training ? 1 : 0);
+      if (relu)
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<data_t, scalar_t, true>), dim3(grid1), dim3(256), 0,
+                           stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
+                           (const data_t*)dyc.data_ptr(), (data_t*)dx.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           (const scalar_t*)g.data_ptr(), sums.data_ptr<float>(),
+                           total, C, rows, training ? 1 : 0);
+      else
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<data_t, scalar_t, false>), dim3(grid1), dim3(256), 0,
+                           stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
+                           (const data_t*)dyc.data_ptr(), (data_t*)dx.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           (const scalar_t*)g.data_ptr(), sums.data_ptr<float>(),
+                           total, C, rows, training ? 1 : 0);
     });
   });
   return {dx, dgamma, dbeta};

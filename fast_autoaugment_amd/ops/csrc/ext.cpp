@@ -33,11 +33,11 @@ torch::Tensor aug_pipeline_imagenet(torch::Tensor images, torch::Tensor sel,
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
-                                       double momentum, double eps);
+                                       double momentum, double eps, bool relu);
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor out, torch::Tensor mean,
                                        torch::Tensor invstd, torch::Tensor gamma,
-                                       bool training);
+                                       bool training, bool relu);
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                          int64_t stride, int64_t pad);
 torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w, int64_t stride,
@@ -46,6 +46,12 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                              int64_t stride, int64_t pad,
                                              int64_t KH, int64_t KW, bool want_bias);
 torch::Tensor colsum_bf16(torch::Tensor dy);
+torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pt, int64_t pb, int64_t pl, int64_t pr);
+torch::Tensor dwconv_bwd_data(torch::Tensor dy, torch::Tensor w, int64_t stride,
+                              int64_t pt, int64_t pl, int64_t H, int64_t W);
+torch::Tensor dwconv_bwd_weight(torch::Tensor dy, torch::Tensor x, int64_t stride,
+                                int64_t pt, int64_t pl, int64_t KH, int64_t KW);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_bcast", &scale_bcast, "out = x * s[b] (per-sample broadcast)");
@@ -74,5 +80,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("colsum_bf16", &colsum_bf16, "channel column-sum (bias grad)");
+  m.def("dwconv_fwd", &dwconv_fwd, "depthwise conv forward (NHWC bf16)");
+  m.def("dwconv_bwd_data", &dwconv_bwd_data);
+  m.def("dwconv_bwd_weight", &dwconv_bwd_weight);
   m.def("bn_relu_bwd", &bn_relu_bwd);
 }

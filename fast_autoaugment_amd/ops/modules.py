@@ -24,3 +24,12 @@ def bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
         from .bnrelu import fused_bn_relu
         return fused_bn_relu(x, bn)
     return F.relu(bn(x))
+
+
+def bn_only(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
+    """BatchNorm2d without activation, fused on GPU."""
+    if (x.is_cuda and has_ext()
+            and os.environ.get("FAA_NO_FUSED_BN") != "1"):
+        from .bnrelu import fused_bn
+        return fused_bn(x, bn)
+    return bn(x)

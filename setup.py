@@ -24,6 +24,7 @@ sources = [
     os.path.join(CSRC, "aug_kernels.hip"),
     os.path.join(CSRC, "bnrelu.hip"),
     os.path.join(CSRC, "conv_mfma.hip"),
+    os.path.join(CSRC, "depthwise.hip"),
 ]
 
 setup(

@@ -106,3 +106,44 @@ def test_patched_conv_module_end_to_end():
                      (m.bias.grad.float(), mref.bias.grad)]:
         scale = ref.abs().max().item() + 1e-3
         assert (got - ref).abs().max().item() / scale < 3e-2
+
+
+@pytest.mark.parametrize("C,H,k,stride", [
+    (32, 32, 3, 1), (96, 56, 3, 2), (144, 28, 5, 1), (240, 28, 5, 2),
+    (576, 14, 5, 1), (1152, 7, 5, 1),
+])
+def test_depthwise_fwd_bwd(C, H, k, stride):
+    """Depthwise kernels vs torch grouped conv (TF-SAME asymmetric pads)."""
+    import math
+    torch.manual_seed(0)
+    x = (torch.randn(8, C, H, H, device=dev()) * 0.5)
+    w = torch.randn(C, 1, k, k, device=dev()) * 0.2
+    b = torch.randn(C, device=dev()) * 0.1
+    ph = max((math.ceil(H / stride) - 1) * stride + k - H, 0)
+    pl, pr, pt, pb = ph // 2, ph - ph // 2, ph // 2, ph - ph // 2
+    xp = torch.nn.functional.pad(x, (pl, pr, pt, pb))
+    ref = torch.nn.functional.conv2d(xp, w, b, stride=stride, groups=C)
+
+    xb = x.bfloat16().contiguous(memory_format=torch.channels_last)
+    wb = w.bfloat16().contiguous(memory_format=torch.channels_last)
+    got = C_ext().dwconv_fwd(xb, wb, b.bfloat16(), stride, pt, pb, pl, pr).float()
+    scale = ref.abs().max().item() + 1e-3
+    assert (got - ref).abs().max().item() / scale < 2e-2
+
+    # bwd-data + bwd-weight vs torch
+    dy = torch.randn_like(ref) * 0.2
+    dyb = dy.bfloat16().contiguous(memory_format=torch.channels_last)
+    ref_dx = torch.nn.grad.conv2d_input(list(xp.shape), w, dy, stride=stride,
+                                        groups=C)[:, :, pt:pt + H, pl:pl + H]
+    got_dx = C_ext().dwconv_bwd_data(dyb, wb, stride, pt, pl, H, H).float()
+    s2 = ref_dx.abs().max().item() + 1e-3
+    assert (got_dx - ref_dx).abs().max().item() / s2 < 2e-2
+    ref_dw = torch.nn.grad.conv2d_weight(xp, list(w.shape), dy, stride=stride, groups=C)
+    got_dw = C_ext().dwconv_bwd_weight(dyb, xb, stride, pt, pl, k, k).float()
+    s3 = ref_dw.abs().max().item() + 1e-3
+    assert (got_dw - ref_dw).abs().max().item() / s3 < 2e-2
+
+
+def C_ext():
+    from fast_autoaugment_amd.ops import ext
+    return ext()
