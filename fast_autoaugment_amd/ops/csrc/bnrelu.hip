@@ -71,39 +71,64 @@ __global__ void bn_reduce_vec_kernel(const T* __restrict__ x, float* __restrict_
   float* out = scratch + (int64_t)blockIdx.x * 2 * C;
   const int groups = C / 8;
   const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % groups);
-  if ((int)threadIdx.x < C) {
-    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+  // strided channel loop: C can exceed blockDim (up to 2048 channels/block)
+  for (int ch = threadIdx.x; ch < C; ch += (int)blockDim.x) {
+    int oct = ch / 8, lane = ch % 8;
     int t0 = (oct - shift + groups) % groups;
     float acc = 0;
     for (int t = t0; t < (int)blockDim.x; t += groups)
       acc += lds[t * 8 + lane];
-    out[threadIdx.x] = acc;
+    out[ch] = acc;
   }
   __syncthreads();
   #pragma unroll
   for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = ss[k];
   __syncthreads();
-  if ((int)threadIdx.x < C) {
-    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+  for (int ch = threadIdx.x; ch < C; ch += (int)blockDim.x) {
+    int oct = ch / 8, lane = ch % 8;
     int t0 = (oct - shift + groups) % groups;
     float acc = 0;
     for (int t = t0; t < (int)blockDim.x; t += groups)
       acc += lds[t * 8 + lane];
-    out[C + threadIdx.x] = acc;
+    out[C + ch] = acc;
   }
 }
 
-// scalar fallback for C % 8 != 0 (rare; uses global atomics on zeroed scratch)
+// any-C scalar reduce (C % 8 != 0, e.g. PyramidNet's rounded widths): the
+// grid stride (nblocks*256) is chosen a multiple of C on the host, so each
+// thread's channel c = i0 % C is loop-invariant; per-block partials are
+// folded through LDS (no atomics) and finalized like the vec path.
 template <typename T>
-__global__ void bn_reduce_scalar_kernel(const T* __restrict__ x, float* __restrict__ scratch,
-                                        int64_t total, int C) {
+__global__ void bn_reduce_anyc_kernel(const T* __restrict__ x, float* __restrict__ scratch,
+                                      int64_t total, int C) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float s = 0, ss = 0;
   for (int64_t i = i0; i < total; i += stride) {
     float f = faa_to_float(x[i]);
-    int c = (int)(i % C);
-    atomicAdd(&scratch[c], f);
-    atomicAdd(&scratch[C + c], f * f);
+    s += f; ss += f * f;
+  }
+  // thread t owns channel (bid*256+t) % C; fold same-channel threads in LDS.
+  // Channels with no owner thread in this block get an explicit 0 partial.
+  __shared__ float lds[256];
+  lds[threadIdx.x] = s;
+  __syncthreads();
+  float* out = scratch + (int64_t)blockIdx.x * 2 * C;
+  const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % C);
+  for (int c = threadIdx.x; c < C; c += (int)blockDim.x) {
+    int t0 = (c - shift + C) % C;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += C) acc += lds[t];
+    out[c] = acc;
+  }
+  __syncthreads();
+  lds[threadIdx.x] = ss;
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += (int)blockDim.x) {
+    int t0 = (c - shift + C) % C;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += C) acc += lds[t];
+    out[C + c] = acc;
   }
 }
 
@@ -238,40 +263,63 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
   float* outp = scratch + (int64_t)blockIdx.x * 2 * C;
   const int groups = C / 8;
   const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % groups);
-  if ((int)threadIdx.x < C) {
-    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+  for (int ch = threadIdx.x; ch < C; ch += (int)blockDim.x) {
+    int oct = ch / 8, lane = ch % 8;
     int t0 = (oct - shift + groups) % groups;
     float acc = 0;
     for (int t = t0; t < (int)blockDim.x; t += groups) acc += lds[t * 8 + lane];
-    outp[threadIdx.x] = acc;
+    outp[ch] = acc;
   }
   __syncthreads();
   #pragma unroll
   for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = sdyx[k];
   __syncthreads();
-  if ((int)threadIdx.x < C) {
-    int oct = threadIdx.x / 8, lane = threadIdx.x % 8;
+  for (int ch = threadIdx.x; ch < C; ch += (int)blockDim.x) {
+    int oct = ch / 8, lane = ch % 8;
     int t0 = (oct - shift + groups) % groups;
     float acc = 0;
     for (int t = t0; t < (int)blockDim.x; t += groups) acc += lds[t * 8 + lane];
-    outp[C + threadIdx.x] = acc;
+    outp[C + ch] = acc;
   }
 }
 
+// any-C bwd reduce: same channel-invariant decomposition as bn_reduce_anyc.
 template <typename T, bool RELU>
-__global__ void bn_bwd_reduce_scalar_kernel(const T* __restrict__ x, const T* __restrict__ out,
-                                            const T* __restrict__ dy,
-                                            const float* __restrict__ mean,
-                                            const float* __restrict__ invstd,
-                                            float* __restrict__ scratch,
-                                            int64_t total, int C) {
+__global__ void bn_bwd_reduce_anyc_kernel(const T* __restrict__ x, const T* __restrict__ out,
+                                          const T* __restrict__ dy,
+                                          const float* __restrict__ mean,
+                                          const float* __restrict__ invstd,
+                                          float* __restrict__ scratch,
+                                          int64_t total, int C) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int cown = (int)(i0 % C);
+  const float m = mean[cown], is = invstd[cown];
+  float sdy = 0, sdyx = 0;
   for (int64_t i = i0; i < total; i += stride) {
-    int c = (int)(i % C);
     float g = (!RELU || faa_to_float(out[i]) > 0.0f) ? faa_to_float(dy[i]) : 0.0f;
-    atomicAdd(&scratch[c], g);
-    atomicAdd(&scratch[C + c], g * (faa_to_float(x[i]) - mean[c]) * invstd[c]);
+    sdy += g;
+    sdyx += g * (faa_to_float(x[i]) - m) * is;
+  }
+  __shared__ float lds[256];
+  lds[threadIdx.x] = sdy;
+  __syncthreads();
+  float* outp = scratch + (int64_t)blockIdx.x * 2 * C;
+  const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % C);
+  for (int c = threadIdx.x; c < C; c += (int)blockDim.x) {
+    int t0 = (c - shift + C) % C;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += C) acc += lds[t];
+    outp[c] = acc;
+  }
+  __syncthreads();
+  lds[threadIdx.x] = sdyx;
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += (int)blockDim.x) {
+    int t0 = (c - shift + C) % C;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += C) acc += lds[t];
+    outp[C + c] = acc;
   }
 }
 
@@ -386,6 +434,16 @@ static int bn_nblocks(int C, int64_t total) {
   return ((base + q - 1) / q) * q;
 }
 
+// any-C variant: grid stride is nblocks*256 (scalar elements), must be a
+// multiple of C for channel invariance
+static int bn_nblocks_anyc(int C, int64_t total) {
+  auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
+  int q = C / gcd(C, 256);
+  int64_t want = total / 256 / 32;
+  int base = (int)std::min<int64_t>(std::max<int64_t>(want, 64), 1024);
+  return ((base + q - 1) / q) * q;
+}
+
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
@@ -421,15 +479,15 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                          running_var.defined() ? running_var.data_ptr<float>() : nullptr,
                          C, rows, (float)eps, (float)momentum);
     } else {
-      auto scratch = torch::zeros({2 * C}, f32);
-      int grid = faa_grid(total, 256, 1024);
+      int nb = bn_nblocks_anyc(C, total);
+      auto scratch = torch::empty({nb, 2 * C}, f32);
       DISPATCH_FB(xc.scalar_type(), "bn_reduce_s", [&] {
-        hipLaunchKernelGGL((bn_reduce_scalar_kernel<scalar_t>), dim3(grid), dim3(256), 0,
+        hipLaunchKernelGGL((bn_reduce_anyc_kernel<scalar_t>), dim3(nb), dim3(256), 0,
                            stream, (const scalar_t*)xc.data_ptr(),
                            scratch.data_ptr<float>(), total, C);
       });
       hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
-                         scratch.data_ptr<float>(), 1, mean.data_ptr<float>(),
+                         scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
                          invstd.data_ptr<float>(),
                          running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
                          running_var.defined() ? running_var.data_ptr<float>() : nullptr,
@@ -500,22 +558,25 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                          (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
     });
   } else {
-    sums.zero_();
-    int grid = faa_grid(total, 256, 1024);
+    int nb = bn_nblocks_anyc(C, total);
+    auto scratch = torch::empty({nb, 2 * C}, f32);
     DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce_s", [&] {
       if (!relu)
-        hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<scalar_t, false>), dim3(grid), dim3(256), 0,
+        hipLaunchKernelGGL((bn_bwd_reduce_anyc_kernel<scalar_t, false>), dim3(nb), dim3(256), 0,
                            stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
                            (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), sums.data_ptr<float>(), total, C);
+                           invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
       else
-        hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<scalar_t, true>), dim3(grid), dim3(256), 0,
+        hipLaunchKernelGGL((bn_bwd_reduce_anyc_kernel<scalar_t, true>), dim3(nb), dim3(256), 0,
                          stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
                          (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), sums.data_ptr<float>(), total, C);
+                         invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
     });
-    dbeta.copy_(sums.narrow(0, 0, C));
-    dgamma.copy_(sums.narrow(0, C, C));
+    DISPATCH_FB(g.scalar_type(), "bn_bwd_fin_s", [&] {
+      hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
+                         stream, scratch.data_ptr<float>(), nb, sums.data_ptr<float>(),
+                         (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
+    });
   }
   int grid1 = faa_grid(total / 8 + 1, 256);
   DISPATCH_FB(xc.scalar_type(), "bn_bwd_apply", [&] {

@@ -257,9 +257,16 @@ def test_ema_lerp():
 # ---------------------------------------------------------------- bn_relu
 
 @pytest.mark.parametrize("dtype,tol", [(torch.float32, 2e-5), (torch.bfloat16, 3e-2)])
-def test_bn_relu_fwd_bwd_vs_torch(dtype, tol):
+@pytest.mark.parametrize("Ch", [
+    32,    # vec path, C < blockDim
+    512,   # vec path, C > blockDim (multi-pass LDS fold)
+    640,   # WRN-28-10 widest stage
+    61,    # any-C path, odd channels (PyramidNet rounded widths)
+    340,   # any-C path, C % 8 == 4, C > blockDim
+])
+def test_bn_relu_fwd_bwd_vs_torch(Ch, dtype, tol):
     torch.manual_seed(0)
-    N, Ch, H, W = 8, 32, 16, 16
+    N, H, W = 8, 16, 16
     x = torch.randn(N, Ch, H, W, device=dev(), dtype=dtype).contiguous(
         memory_format=torch.channels_last).requires_grad_(True)
     bn = torch.nn.BatchNorm2d(Ch, momentum=0.9).to(dev())
