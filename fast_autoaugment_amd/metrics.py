@@ -125,10 +125,10 @@ class SummaryWriterDummy:
 
 
 def get_summary_writer(log_dir, enabled: bool):
+    """Per-split scalar writer (reference train.py:176-181): a real
+    TensorBoard event-file writer (in-house, no tensorboard dep) when
+    enabled, a no-op dummy for non-master ranks / untagged runs."""
     if not enabled:
         return SummaryWriterDummy(log_dir)
-    try:
-        from torch.utils.tensorboard import SummaryWriter
-        return SummaryWriter(log_dir=log_dir)
-    except Exception:
-        return SummaryWriterDummy(log_dir)
+    from .tblog import SummaryWriter
+    return SummaryWriter(log_dir=log_dir)

@@ -16,7 +16,22 @@ import traceback
 from typing import Any, Callable, Dict, List, Optional
 
 
+def _set_pdeathsig():
+    """Kill this worker if the parent dies (even by SIGKILL) — the role the
+    reference's safe_shell_exec parent-death pipe plays
+    (reference safe_shell_exec.py:29-60)."""
+    try:
+        import ctypes
+        import signal
+        libc = ctypes.CDLL("libc.so.6", use_errno=True)
+        PR_SET_PDEATHSIG = 1
+        libc.prctl(PR_SET_PDEATHSIG, signal.SIGTERM)
+    except Exception:
+        pass  # non-Linux / restricted: daemon=True still covers clean exits
+
+
 def _worker_main(worker_id: int, gpu_id: Optional[int], task_q, result_q):
+    _set_pdeathsig()
     if gpu_id is not None:
         os.environ["HIP_VISIBLE_DEVICES"] = str(gpu_id)
         os.environ["CUDA_VISIBLE_DEVICES"] = str(gpu_id)

@@ -296,3 +296,25 @@ def test_imagenet_folder_loader(tmp_path):
     # cache hit path
     imgs2, labels2 = load_imagenet_folder(str(root), "train", resize_short=64)
     np.testing.assert_array_equal(imgs, imgs2)
+
+
+def test_tblog_event_file_roundtrip(tmp_path):
+    """In-house TensorBoard writer: TFRecord framing + CRCs + scalar protos."""
+    from fast_autoaugment_amd.tblog import SummaryWriter, read_scalars
+    w = SummaryWriter(str(tmp_path / "run"))
+    w.add_scalar("loss", 1.5, 1)
+    w.add_scalar("loss", 0.75, 2)
+    w.add_scalar("top1", 0.913, 2)
+    w.close()
+    files = list((tmp_path / "run").glob("events.out.tfevents.*"))
+    assert len(files) == 1
+    scalars = read_scalars(str(files[0]))
+    assert (1, "loss", 1.5) in scalars
+    assert (2, "top1",) == scalars[-1][:2] and abs(scalars[-1][2] - 0.913) < 1e-6
+    # get_summary_writer returns the real writer when enabled, dummy otherwise
+    from fast_autoaugment_amd.metrics import get_summary_writer, SummaryWriterDummy
+    assert isinstance(get_summary_writer(str(tmp_path / "r2"), False), SummaryWriterDummy)
+    real = get_summary_writer(str(tmp_path / "r2"), True)
+    real.add_scalar("x", 1.0, 0)
+    real.close()
+    assert list((tmp_path / "r2").glob("events.out.tfevents.*"))
