@@ -51,7 +51,9 @@ def main():
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    distributed = world_size > 1
+    # FAA_BENCH_FORCE_DIST=1: take the distributed path at world_size==1
+    # (single-GPU validation of init_process_group/broadcast/all_reduce)
+    distributed = world_size > 1 or os.environ.get("FAA_BENCH_FORCE_DIST") == "1"
     if distributed:
         import torch.distributed as dist
         dist.init_process_group("gloo" if cpu_mode else "nccl", init_method="env://")
@@ -105,12 +107,21 @@ def main():
     # pure-bf16 compute: params+grads are bf16 flat views (no autocast cast
     # kernels), fp32 master lives in the fused optimizer
     work_dtype = torch.bfloat16 if (args.dtype == "bf16" and not cpu_mode) else torch.float32
-    if distributed:
+    # GPU distributed mode keeps the hipGraph step: plain flat params, and one
+    # eager RCCL all-reduce(AVG) of the flat grad buffer between graph replay
+    # and the fused optimizer step (the collective stays OUTSIDE the capture).
+    # CPU/gloo (and --graphs 0) use FlatDDP's bucketed overlapped all-reduce.
+    use_graphs = bool(args.graphs) and not cpu_mode
+    dist_in_graph = distributed and use_graphs
+    if distributed and not dist_in_graph:
         from fast_autoaugment_amd.parallel.ddp import FlatDDP
         model = FlatDDP(model, work_dtype=work_dtype)
         flat = model.flat
     else:
         flat = flatten_module(model, work_dtype=work_dtype)
+        if dist_in_graph:
+            import torch.distributed as dist
+            dist.broadcast(flat.flat_param, 0)   # rank-0 weight sync
     if work_dtype == torch.bfloat16:
         from fast_autoaugment_amd.ops.conv import patch_convs
         n_patched = patch_convs(model)
@@ -127,7 +138,6 @@ def main():
     steps_per_epoch = max(len(loader), 1)
     step_idx = 0
 
-    use_graphs = bool(args.graphs) and not distributed and not cpu_mode
     from fast_autoaugment_amd.aug import ops as aug_ops
     if not cpu_mode:
         from fast_autoaugment_amd.ops import ext
@@ -195,23 +205,38 @@ def main():
         data = torch.from_numpy(out).permute(0, 3, 1, 2).contiguous()
         return data, store.labels[torch.from_numpy(np.ascontiguousarray(sel))]
 
+    def gpu_fwd_bwd():
+        """aug + forward + loss + backward on static inputs (capturable)."""
+        if imagenet:
+            data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s, post_s,
+                                            mean_t, std_t, out_size, out_size, bf16)
+        else:
+            data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
+                                   mean_t, std_t, bf16)
+        label = store.labels.index_select(0, sel_s)
+        preds = model(data)
+        loss = crit(preds, label)
+        loss.backward()
+        return loss
+
+    def allreduce_flat_grad():
+        import torch.distributed as dist
+        dist.all_reduce(flat.flat_grad, op=dist.ReduceOp.AVG)
+
     def step_body():
         opt.zero_grad()
         if cpu_mode:
             data, label = make_batch_cpu()
+            preds = model(data)
+            loss = crit(preds, label)
+            loss.backward()
         else:
-            if imagenet:
-                data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s, post_s,
-                                                mean_t, std_t, out_size, out_size, bf16)
-            else:
-                data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
-                                       mean_t, std_t, bf16)
-            label = store.labels.index_select(0, sel_s)
-        preds = model(data)
-        loss = crit(preds, label)
-        loss.backward()
+            loss = gpu_fwd_bwd()
         if distributed:
-            model.finish_gradient_sync()
+            if dist_in_graph:
+                allreduce_flat_grad()
+            else:
+                model.finish_gradient_sync()
         opt.step()
         return loss
 
@@ -221,13 +246,12 @@ def main():
     # gather kernel packs them into the flat buffer for the fused step.
     gather_mode = (args.grad_mode == "gather" and use_graphs)
 
-    def fwd_bwd_body():
-        data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s, mean_t, std_t, bf16)
-        label = store.labels.index_select(0, sel_s)
-        preds = model(data)
-        loss = crit(preds, label)
-        loss.backward()
-        return loss
+    def flat_capture_body():
+        """flat-accumulation capture body: under distributed graphs the
+        collective + optimizer run eagerly after replay, so only
+        zero-grad + fwd + bwd go into the graph."""
+        opt.zero_grad()
+        gpu_fwd_bwd()
 
     graph = None
     gather_table = None
@@ -251,7 +275,7 @@ def main():
             for p_ in flat.params:
                 p_.grad = None          # capture assignment-mode backward
             with torch.cuda.graph(graph):
-                fwd_bwd_body()
+                gpu_fwd_bwd()
             # grads now live in the graph pool at replay-stable addresses
             base = flat.flat_param.data_ptr()
             rows = []
@@ -289,10 +313,10 @@ def main():
                 graph = torch.cuda.CUDAGraph()
                 upload_next()
                 with torch.cuda.graph(graph):
-                    step_body()
+                    flat_capture_body() if dist_in_graph else step_body()
         else:
             with torch.cuda.graph(graph):
-                step_body()
+                flat_capture_body() if dist_in_graph else step_body()
 
     def one_step():
         nonlocal step_idx
@@ -303,6 +327,12 @@ def main():
             graph.replay()
             if gather_mode:
                 CX.gather_grads(gather_table, flat.flat_grad)
+                if distributed:
+                    allreduce_flat_grad()
+                opt.step()
+            elif dist_in_graph:
+                # flat fallback under distributed: graph holds zero+fwd+bwd
+                allreduce_flat_grad()
                 opt.step()
         else:
             if not cpu_mode:
