@@ -20,6 +20,16 @@ from torch import nn
 from torch.nn import functional as F
 
 from ..ops.functional import Swish, drop_connect
+from ..ops.modules import bn_only
+
+
+def _bn(bn, x):
+    """Fused BN kernel for plain BatchNorm2d (the 20%-of-step torch
+    batch_norm_collect_statistics path in profiles/step_profile_efficientnet_b0_r01.txt);
+    Tpu/Sync BN variants keep their own cross-replica forward."""
+    if type(bn) is nn.BatchNorm2d:
+        return bn_only(x, bn)
+    return bn(x)
 
 GlobalParams = collections.namedtuple("GlobalParams", [
     "batch_norm_momentum", "batch_norm_epsilon", "dropout_rate", "num_classes",
@@ -200,13 +210,13 @@ class MBConvBlock(nn.Module):
 
         x = inputs
         if self._block_args.expand_ratio != 1:
-            x = self._swish(self._bn0(conv(self._expand_conv, x)))
-        x = self._swish(self._bn1(conv(self._depthwise_conv, x)))
+            x = self._swish(_bn(self._bn0, conv(self._expand_conv, x)))
+        x = self._swish(_bn(self._bn1, conv(self._depthwise_conv, x)))
         if self.has_se:
             sq = F.adaptive_avg_pool2d(x, 1)
             sq = self._se_expand(self._swish(self._se_reduce(sq)))
             x = torch.sigmoid(sq) * x
-        x = self._bn2(conv(self._project_conv, x))
+        x = _bn(self._bn2, conv(self._project_conv, x))
         if (self.id_skip and self._block_args.stride == 1
                 and self._block_args.input_filters == self._block_args.output_filters):
             if drop_connect_rate:
@@ -251,13 +261,13 @@ class EfficientNet(nn.Module):
         self._swish = Swish()
 
     def extract_features(self, inputs):
-        x = self._swish(self._bn0(self._conv_stem(inputs)))
+        x = self._swish(_bn(self._bn0, self._conv_stem(inputs)))
         for idx, block in enumerate(self._blocks):
             rate = self._global_params.drop_connect_rate
             if rate:
                 rate *= float(idx) / len(self._blocks)
             x = block(x, drop_connect_rate=rate)
-        return self._swish(self._bn1(self._conv_head(x)))
+        return self._swish(_bn(self._bn1, self._conv_head(x)))
 
     def forward(self, inputs):
         x = self.extract_features(inputs)
