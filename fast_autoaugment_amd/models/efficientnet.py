@@ -20,7 +20,7 @@ from torch import nn
 from torch.nn import functional as F
 
 from ..ops.functional import Swish, drop_connect
-from ..ops.modules import bn_only
+from ..ops.modules import bn_only, bn_swish
 
 
 def _bn(bn, x):
@@ -30,6 +30,13 @@ def _bn(bn, x):
     if type(bn) is nn.BatchNorm2d:
         return bn_only(x, bn)
     return bn(x)
+
+
+def _bn_swish(bn, x, swish_mod):
+    """BN + swish in one kernel for plain BatchNorm2d; composed otherwise."""
+    if type(bn) is nn.BatchNorm2d:
+        return bn_swish(x, bn)
+    return swish_mod(bn(x))
 
 GlobalParams = collections.namedtuple("GlobalParams", [
     "batch_norm_momentum", "batch_norm_epsilon", "dropout_rate", "num_classes",
@@ -210,8 +217,8 @@ class MBConvBlock(nn.Module):
 
         x = inputs
         if self._block_args.expand_ratio != 1:
-            x = self._swish(_bn(self._bn0, conv(self._expand_conv, x)))
-        x = self._swish(_bn(self._bn1, conv(self._depthwise_conv, x)))
+            x = _bn_swish(self._bn0, conv(self._expand_conv, x), self._swish)
+        x = _bn_swish(self._bn1, conv(self._depthwise_conv, x), self._swish)
         if self.has_se:
             sq = F.adaptive_avg_pool2d(x, 1)
             sq = self._se_expand(self._swish(self._se_reduce(sq)))
@@ -261,13 +268,13 @@ class EfficientNet(nn.Module):
         self._swish = Swish()
 
     def extract_features(self, inputs):
-        x = self._swish(_bn(self._bn0, self._conv_stem(inputs)))
+        x = _bn_swish(self._bn0, self._conv_stem(inputs), self._swish)
         for idx, block in enumerate(self._blocks):
             rate = self._global_params.drop_connect_rate
             if rate:
                 rate *= float(idx) / len(self._blocks)
             x = block(x, drop_connect_rate=rate)
-        return self._swish(_bn(self._bn1, self._conv_head(x)))
+        return _bn_swish(self._bn1, self._conv_head(x), self._swish)
 
     def forward(self, inputs):
         x = self.extract_features(inputs)

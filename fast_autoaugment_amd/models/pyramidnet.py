@@ -12,23 +12,9 @@ import math
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
-from ..ops.functional import ShakeDrop
+from ..ops.functional import ShakeDrop, pad_add as _pad_add
 from ..ops.modules import bn_only, bn_relu
-
-
-def _pad_add(out: torch.Tensor, shortcut: torch.Tensor) -> torch.Tensor:
-    """out += zero-channel-padded shortcut (shortcut has fewer channels).
-
-    Written as narrow-add instead of the reference's zeros+cat
-    (pyramidnet.py:109-113): no zero tensor is materialized and autograd
-    produces the narrowed gradient for the shortcut directly.
-    """
-    cs = shortcut.size(1)
-    if out.size(1) == cs:
-        return out + shortcut
-    return out + F.pad(shortcut, (0, 0, 0, 0, 0, out.size(1) - cs))
 
 
 class BasicBlock(nn.Module):

@@ -6,28 +6,32 @@ import torch
 from . import ext
 
 
+# activation codes matching the HIP kernels' ACT template parameter
+ACT_NONE, ACT_RELU, ACT_SWISH = 0, 1, 2
+
+
 class FusedBNReLUFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, relu):
+                momentum, eps, act):
         C = ext()
         out, mean, invstd = C.bn_relu_fwd(x, gamma, beta, running_mean, running_var,
-                                          training, momentum, eps, relu)
-        ctx.save_for_backward(x, out, mean, invstd, gamma)
+                                          training, momentum, eps, act)
+        ctx.save_for_backward(x, out, mean, invstd, gamma, beta)
         ctx.training = training
-        ctx.relu = relu
+        ctx.act = act
         return out
 
     @staticmethod
     def backward(ctx, dy):
-        x, out, mean, invstd, gamma = ctx.saved_tensors
+        x, out, mean, invstd, gamma, beta = ctx.saved_tensors
         C = ext()
-        dx, dgamma, dbeta = C.bn_relu_bwd(dy, x, out, mean, invstd, gamma,
-                                          ctx.training, ctx.relu)
+        dx, dgamma, dbeta = C.bn_relu_bwd(dy, x, out, mean, invstd, gamma, beta,
+                                          ctx.training, ctx.act)
         return dx, dgamma, dbeta, None, None, None, None, None, None
 
 
-def _fused(x, bn, relu):
+def _fused(x, bn, act):
     training = bn.training
     if training and bn.num_batches_tracked is not None:
         # a per-call .add_(1) is a 5us GPU kernel x 37 BN sites x step; count
@@ -35,15 +39,21 @@ def _fused(x, bn, relu):
         bn._faa_nbt_pending = getattr(bn, "_faa_nbt_pending", 0) + 1
     momentum = bn.momentum if bn.momentum is not None else 0.1
     return FusedBNReLUFn.apply(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
-                               training, momentum, bn.eps, relu)
+                               training, momentum, bn.eps, act)
 
 
 def fused_bn_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
-    return _fused(x, bn, True)
+    return _fused(x, bn, ACT_RELU)
 
 
 def fused_bn(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
-    return _fused(x, bn, False)
+    return _fused(x, bn, ACT_NONE)
+
+
+def fused_bn_swish(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
+    """BN followed by swish x*sigmoid(x) in one kernel (EfficientNet's
+    bn->swish pattern, reference model.py:203-204 + utils.py:38-54)."""
+    return _fused(x, bn, ACT_SWISH)
 
 
 def sync_bn_trackers(model: torch.nn.Module) -> None:

@@ -166,7 +166,32 @@ __device__ __forceinline__ short f_to_bf16_bits(float f) {
   return *reinterpret_cast<short*>(&h);
 }
 
-template <typename T, typename GT, bool RELU>
+// fused activation after the affine transform: 0 = none, 1 = relu, 2 = swish
+template <int ACT>
+__device__ __forceinline__ float faa_bn_act(float z) {
+  if constexpr (ACT == 1) {
+    return fmaxf(z, 0.0f);
+  } else if constexpr (ACT == 2) {
+    return z / (1.0f + __expf(-z));
+  }
+  return z;
+}
+
+// dz/dy for the fused activation. relu gates on the saved output; swish
+// recomputes the pre-activation z = (x-mean)*invstd*gamma+beta.
+template <int ACT>
+__device__ __forceinline__ float faa_bn_act_grad(float g, float out_v, float z) {
+  if constexpr (ACT == 1) {
+    return out_v > 0.0f ? g : 0.0f;
+  } else if constexpr (ACT == 2) {
+    float s = 1.0f / (1.0f + __expf(-z));
+    return g * s * (1.0f + z * (1.0f - s));
+  }
+  (void)out_v; (void)z;
+  return g;
+}
+
+template <typename T, typename GT, int ACT>
 __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
@@ -189,8 +214,8 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
           int cb = c0 + k + 4; while (cb >= C) cb -= C;
           float a = (bf16_bits_to_f(v0[k]) - mean[ca]) * invstd[ca] * faa_to_float(gamma[ca]) + faa_to_float(beta[ca]);
           float b = (bf16_bits_to_f(v1[k]) - mean[cb]) * invstd[cb] * faa_to_float(gamma[cb]) + faa_to_float(beta[cb]);
-          o0[k] = f_to_bf16_bits(RELU ? fmaxf(a, 0.0f) : a);
-          o1[k] = f_to_bf16_bits(RELU ? fmaxf(b, 0.0f) : b);
+          o0[k] = f_to_bf16_bits(faa_bn_act<ACT>(a));
+          o1[k] = f_to_bf16_bits(faa_bn_act<ACT>(b));
         }
         short* os = reinterpret_cast<short*>(out);
         *reinterpret_cast<short4v*>(os + i) = o0;
@@ -204,25 +229,31 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ out,
       if (j < total) {
         int c = c0 + k; while (c >= C) c -= C;
         float v = (faa_to_float(x[j]) - mean[c]) * invstd[c] * faa_to_float(gamma[c]) + faa_to_float(beta[c]);
-        faa_from_float(RELU ? fmaxf(v, 0.0f) : v, &out[j]);
+        faa_from_float(faa_bn_act<ACT>(v), &out[j]);
       }
     }
   }
 }
 
 // ---------------------------------------------------------------- bwd reduce
-template <typename T, bool RELU>
+template <typename T, typename GT, int ACT>
 __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __restrict__ out,
                                          const T* __restrict__ dy,
                                          const float* __restrict__ mean,
                                          const float* __restrict__ invstd,
+                                         const GT* __restrict__ gamma,
+                                         const GT* __restrict__ beta,
                                          float* __restrict__ scratch,
                                          int64_t total, int C) {
   float sdy[8] = {0}, sdyx[8] = {0};
   const int c0 = (int)((((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8) % C);
-  float m[8], is[8];
+  float m[8], is[8], ga[8], be[8];
   #pragma unroll
-  for (int k = 0; k < 8; ++k) { int c = (c0 + k) % C; m[k] = mean[c]; is[k] = invstd[c]; }
+  for (int k = 0; k < 8; ++k) {
+    int c = (c0 + k) % C;
+    m[k] = mean[c]; is[k] = invstd[c];
+    ga[k] = faa_to_float(gamma[c]); be[k] = faa_to_float(beta[c]);
+  }
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   if constexpr (sizeof(T) == 2) {
@@ -238,21 +269,27 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
       short4v dv1 = *reinterpret_cast<const short4v*>(ds + i + 4);
       #pragma unroll
       for (int k = 0; k < 4; ++k) {
-        float g0 = (!RELU || bf16_bits_to_f(ov0[k]) > 0.0f) ? bf16_bits_to_f(dv0[k]) : 0.0f;
-        float g1 = (!RELU || bf16_bits_to_f(ov1[k]) > 0.0f) ? bf16_bits_to_f(dv1[k]) : 0.0f;
+        float xh0 = (bf16_bits_to_f(xv0[k]) - m[k]) * is[k];
+        float xh1 = (bf16_bits_to_f(xv1[k]) - m[k + 4]) * is[k + 4];
+        float g0 = faa_bn_act_grad<ACT>(bf16_bits_to_f(dv0[k]), bf16_bits_to_f(ov0[k]),
+                                        xh0 * ga[k] + be[k]);
+        float g1 = faa_bn_act_grad<ACT>(bf16_bits_to_f(dv1[k]), bf16_bits_to_f(ov1[k]),
+                                        xh1 * ga[k + 4] + be[k + 4]);
         sdy[k] += g0;
-        sdyx[k] += g0 * (bf16_bits_to_f(xv0[k]) - m[k]) * is[k];
+        sdyx[k] += g0 * xh0;
         sdy[k + 4] += g1;
-        sdyx[k + 4] += g1 * (bf16_bits_to_f(xv1[k]) - m[k + 4]) * is[k + 4];
+        sdyx[k + 4] += g1 * xh1;
       }
     }
   } else {
     for (int64_t i = i0; i < total; i += stride) {
       #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = (!RELU || faa_to_float(out[i + k]) > 0.0f) ? faa_to_float(dy[i + k]) : 0.0f;
+        float xh = (faa_to_float(x[i + k]) - m[k]) * is[k];
+        float g = faa_bn_act_grad<ACT>(faa_to_float(dy[i + k]), faa_to_float(out[i + k]),
+                                       xh * ga[k] + be[k]);
         sdy[k] += g;
-        sdyx[k] += g * (faa_to_float(x[i + k]) - m[k]) * is[k];
+        sdyx[k] += g * xh;
       }
     }
   }
@@ -284,22 +321,27 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
 }
 
 // any-C bwd reduce: same channel-invariant decomposition as bn_reduce_anyc.
-template <typename T, bool RELU>
+template <typename T, typename GT, int ACT>
 __global__ void bn_bwd_reduce_anyc_kernel(const T* __restrict__ x, const T* __restrict__ out,
                                           const T* __restrict__ dy,
                                           const float* __restrict__ mean,
                                           const float* __restrict__ invstd,
+                                          const GT* __restrict__ gamma,
+                                          const GT* __restrict__ beta,
                                           float* __restrict__ scratch,
                                           int64_t total, int C) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const int cown = (int)(i0 % C);
   const float m = mean[cown], is = invstd[cown];
+  const float ga = faa_to_float(gamma[cown]), be = faa_to_float(beta[cown]);
   float sdy = 0, sdyx = 0;
   for (int64_t i = i0; i < total; i += stride) {
-    float g = (!RELU || faa_to_float(out[i]) > 0.0f) ? faa_to_float(dy[i]) : 0.0f;
+    float xh = (faa_to_float(x[i]) - m) * is;
+    float g = faa_bn_act_grad<ACT>(faa_to_float(dy[i]), faa_to_float(out[i]),
+                                   xh * ga + be);
     sdy += g;
-    sdyx += g * (faa_to_float(x[i]) - m) * is;
+    sdyx += g * xh;
   }
   __shared__ float lds[256];
   lds[threadIdx.x] = sdy;
@@ -342,12 +384,13 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ scratch, int nb
   }
 }
 
-template <typename T, typename GT, bool RELU>
+template <typename T, typename GT, int ACT>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict__ out,
                                     const T* __restrict__ dy, T* __restrict__ dx,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     const GT* __restrict__ gamma,
+                                    const GT* __restrict__ beta,
                                     const float* __restrict__ sums,
                                     int64_t total, int C, int64_t count, int training) {
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
@@ -371,18 +414,21 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
         for (int k = 0; k < 4; ++k) {
           int ca = c0 + k; while (ca >= C) ca -= C;
           int cb = c0 + k + 4; while (cb >= C) cb -= C;
-          float ga = (!RELU || bf16_bits_to_f(ov0[k]) > 0.0f) ? bf16_bits_to_f(dv0[k]) : 0.0f;
-          float gb = (!RELU || bf16_bits_to_f(ov1[k]) > 0.0f) ? bf16_bits_to_f(dv1[k]) : 0.0f;
           float isa = invstd[ca], isb = invstd[cb];
+          float gca = faa_to_float(gamma[ca]), gcb = faa_to_float(gamma[cb]);
+          float xa = (bf16_bits_to_f(xv0[k]) - mean[ca]) * isa;
+          float xb = (bf16_bits_to_f(xv1[k]) - mean[cb]) * isb;
+          float ga = faa_bn_act_grad<ACT>(bf16_bits_to_f(dv0[k]), bf16_bits_to_f(ov0[k]),
+                                          xa * gca + faa_to_float(beta[ca]));
+          float gb = faa_bn_act_grad<ACT>(bf16_bits_to_f(dv1[k]), bf16_bits_to_f(ov1[k]),
+                                          xb * gcb + faa_to_float(beta[cb]));
           float ra, rb;
           if (training) {
-            float xa = (bf16_bits_to_f(xv0[k]) - mean[ca]) * isa;
-            float xb = (bf16_bits_to_f(xv1[k]) - mean[cb]) * isb;
-            ra = faa_to_float(gamma[ca]) * isa * (ga - sums[ca] * inv_count - xa * sums[C + ca] * inv_count);
-            rb = faa_to_float(gamma[cb]) * isb * (gb - sums[cb] * inv_count - xb * sums[C + cb] * inv_count);
+            ra = gca * isa * (ga - sums[ca] * inv_count - xa * sums[C + ca] * inv_count);
+            rb = gcb * isb * (gb - sums[cb] * inv_count - xb * sums[C + cb] * inv_count);
           } else {
-            ra = faa_to_float(gamma[ca]) * isa * ga;
-            rb = faa_to_float(gamma[cb]) * isb * gb;
+            ra = gca * isa * ga;
+            rb = gcb * isb * gb;
           }
           r0[k] = f_to_bf16_bits(ra);
           r1[k] = f_to_bf16_bits(rb);
@@ -398,15 +444,16 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
       int64_t j = i + k;
       if (j < total) {
         int c = c0 + k; while (c >= C) c -= C;
-        float mask = (!RELU || faa_to_float(out[j]) > 0.0f) ? 1.0f : 0.0f;
-        float g = faa_to_float(dy[j]) * mask;
         float is = invstd[c];
+        float gc = faa_to_float(gamma[c]);
+        float xhat = (faa_to_float(x[j]) - mean[c]) * is;
+        float g = faa_bn_act_grad<ACT>(faa_to_float(dy[j]), faa_to_float(out[j]),
+                                       xhat * gc + faa_to_float(beta[c]));
         float res;
         if (training) {
-          float xhat = (faa_to_float(x[j]) - mean[c]) * is;
-          res = faa_to_float(gamma[c]) * is * (g - sums[c] * inv_count - xhat * sums[C + c] * inv_count);
+          res = gc * is * (g - sums[c] * inv_count - xhat * sums[C + c] * inv_count);
         } else {
-          res = faa_to_float(gamma[c]) * is * g;
+          res = gc * is * g;
         }
         faa_from_float(res, &dx[j]);
       }
@@ -434,6 +481,14 @@ static int bn_nblocks(int C, int64_t total) {
   return ((base + q - 1) / q) * q;
 }
 
+// 3-way activation dispatch: f receives integral_constant<int, ACT>
+template <typename F>
+static void act_dispatch(int act, F&& f) {
+  if (act == 1) f(std::integral_constant<int, 1>{});
+  else if (act == 2) f(std::integral_constant<int, 2>{});
+  else f(std::integral_constant<int, 0>{});
+}
+
 // any-C variant: grid stride is nblocks*256 (scalar elements), must be a
 // multiple of C for channel invariance
 static int bn_nblocks_anyc(int C, int64_t total) {
@@ -447,7 +502,7 @@ static int bn_nblocks_anyc(int C, int64_t total) {
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
-                                       double momentum, double eps, bool relu) {
+                                       double momentum, double eps, int64_t act) {
   TORCH_CHECK(x.dim() == 4, "bn_relu: 4-D input expected");
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   int C = xc.size(1);
@@ -501,18 +556,14 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   DISPATCH_FB(xc.scalar_type(), "bn_apply", [&] {
     using data_t = scalar_t;
     DISPATCH_FB(g.scalar_type(), "bn_apply_g", [&] {
-      if (relu)
-        hipLaunchKernelGGL((bn_apply_kernel<data_t, scalar_t, true>), dim3(grid1), dim3(256), 0,
+      act_dispatch((int)act, [&](auto A) {
+        hipLaunchKernelGGL((bn_apply_kernel<data_t, scalar_t, decltype(A)::value>),
+                           dim3(grid1), dim3(256), 0,
                            stream, (const data_t*)xc.data_ptr(), (data_t*)out.data_ptr(),
                            mean.data_ptr<float>(), invstd.data_ptr<float>(),
                            (const scalar_t*)g.data_ptr(), (const scalar_t*)bta.data_ptr(),
                            total, C);
-      else
-        hipLaunchKernelGGL((bn_apply_kernel<data_t, scalar_t, false>), dim3(grid1), dim3(256), 0,
-                           stream, (const data_t*)xc.data_ptr(), (data_t*)out.data_ptr(),
-                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                           (const scalar_t*)g.data_ptr(), (const scalar_t*)bta.data_ptr(),
-                           total, C);
+      });
     });
   });
   return {out, mean, invstd};
@@ -521,7 +572,7 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor out, torch::Tensor mean,
                                        torch::Tensor invstd, torch::Tensor gamma,
-                                       bool training, bool relu) {
+                                       torch::Tensor beta, bool training, int64_t act) {
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   auto oc = out.contiguous(torch::MemoryFormat::ChannelsLast);
   auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
@@ -531,71 +582,58 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
   auto f32 = xc.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(xc);
   auto g = gamma.contiguous();
+  auto bta = beta.contiguous();
+  TORCH_CHECK(g.scalar_type() == bta.scalar_type(), "gamma/beta dtype mismatch");
   auto sums = torch::empty({2 * C}, f32);
   auto dgamma = torch::empty({C}, xc.options().dtype(g.scalar_type()));
   auto dbeta = torch::empty({C}, xc.options().dtype(g.scalar_type()));
   auto stream = at::hip::getCurrentHIPStream().stream();
 
   bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
-  if (vec) {
-    int nb = bn_nblocks(C, total);
-    auto scratch = torch::empty({nb, 2 * C}, f32);
-    DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce", [&] {
-      if (relu)
-        hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<scalar_t, true>), dim3(nb), dim3(256), 0,
-                           stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
-                           (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
-      else
-        hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<scalar_t, false>), dim3(nb), dim3(256), 0,
-                           stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
-                           (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
+  int nb = vec ? bn_nblocks(C, total) : bn_nblocks_anyc(C, total);
+  auto scratch = torch::empty({nb, 2 * C}, f32);
+  DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce", [&] {
+    using data_t = scalar_t;
+    DISPATCH_FB(g.scalar_type(), "bn_bwd_reduce_g", [&] {
+      act_dispatch((int)act, [&](auto A) {
+        constexpr int kAct = decltype(A)::value;
+        if (vec)
+          hipLaunchKernelGGL((bn_bwd_reduce_vec_kernel<data_t, scalar_t, kAct>),
+                             dim3(nb), dim3(256), 0,
+                             stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
+                             (const data_t*)dyc.data_ptr(), mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), (const scalar_t*)g.data_ptr(),
+                             (const scalar_t*)bta.data_ptr(), scratch.data_ptr<float>(),
+                             total, C);
+        else
+          hipLaunchKernelGGL((bn_bwd_reduce_anyc_kernel<data_t, scalar_t, kAct>),
+                             dim3(nb), dim3(256), 0,
+                             stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
+                             (const data_t*)dyc.data_ptr(), mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), (const scalar_t*)g.data_ptr(),
+                             (const scalar_t*)bta.data_ptr(), scratch.data_ptr<float>(),
+                             total, C);
+      });
     });
-    DISPATCH_FB(g.scalar_type(), "bn_bwd_fin", [&] {
-      hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
-                         stream, scratch.data_ptr<float>(), nb, sums.data_ptr<float>(),
-                         (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
-    });
-  } else {
-    int nb = bn_nblocks_anyc(C, total);
-    auto scratch = torch::empty({nb, 2 * C}, f32);
-    DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce_s", [&] {
-      if (!relu)
-        hipLaunchKernelGGL((bn_bwd_reduce_anyc_kernel<scalar_t, false>), dim3(nb), dim3(256), 0,
-                           stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
-                           (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
-      else
-        hipLaunchKernelGGL((bn_bwd_reduce_anyc_kernel<scalar_t, true>), dim3(nb), dim3(256), 0,
-                         stream, (const scalar_t*)xc.data_ptr(), (const scalar_t*)oc.data_ptr(),
-                         (const scalar_t*)dyc.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), scratch.data_ptr<float>(), total, C);
-    });
-    DISPATCH_FB(g.scalar_type(), "bn_bwd_fin_s", [&] {
-      hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
-                         stream, scratch.data_ptr<float>(), nb, sums.data_ptr<float>(),
-                         (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
-    });
-  }
+  });
+  DISPATCH_FB(g.scalar_type(), "bn_bwd_fin", [&] {
+    hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
+                       stream, scratch.data_ptr<float>(), nb, sums.data_ptr<float>(),
+                       (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
+  });
   int grid1 = faa_grid(total / 8 + 1, 256);
   DISPATCH_FB(xc.scalar_type(), "bn_bwd_apply", [&] {
     using data_t = scalar_t;
     DISPATCH_FB(g.scalar_type(), "bn_bwd_apply_g", [&] {
-      if (relu)
-        hipLaunchKernelGGL((bn_bwd_apply_kernel<data_t, scalar_t, true>), dim3(grid1), dim3(256), 0,
+      act_dispatch((int)act, [&](auto A) {
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<data_t, scalar_t, decltype(A)::value>),
+                           dim3(grid1), dim3(256), 0,
                            stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
                            (const data_t*)dyc.data_ptr(), (data_t*)dx.data_ptr(),
                            mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                           (const scalar_t*)g.data_ptr(), sums.data_ptr<float>(),
-                           total, C, rows, training ? 1 : 0);
-      else
-        hipLaunchKernelGGL((bn_bwd_apply_kernel<data_t, scalar_t, false>), dim3(grid1), dim3(256), 0,
-                           stream, (const data_t*)xc.data_ptr(), (const data_t*)oc.data_ptr(),
-                           (const data_t*)dyc.data_ptr(), (data_t*)dx.data_ptr(),
-                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                           (const scalar_t*)g.data_ptr(), sums.data_ptr<float>(),
-                           total, C, rows, training ? 1 : 0);
+                           (const scalar_t*)g.data_ptr(), (const scalar_t*)bta.data_ptr(),
+                           sums.data_ptr<float>(), total, C, rows, training ? 1 : 0);
+      });
     });
   });
   return {dx, dgamma, dbeta};

@@ -33,11 +33,11 @@ torch::Tensor aug_pipeline_imagenet(torch::Tensor images, torch::Tensor sel,
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
-                                       double momentum, double eps, bool relu);
+                                       double momentum, double eps, int64_t act);
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor out, torch::Tensor mean,
                                        torch::Tensor invstd, torch::Tensor gamma,
-                                       bool training, bool relu);
+                                       torch::Tensor beta, bool training, int64_t act);
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                          int64_t stride, int64_t pad);
 torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w, int64_t stride,

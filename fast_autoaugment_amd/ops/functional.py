@@ -196,3 +196,31 @@ def mixup(data: torch.Tensor, indices: torch.Tensor, lam: float) -> torch.Tensor
     if C is not None:
         return C.mixup_fwd(data, indices, lam)
     return data * lam + data[indices] * (1 - lam)
+
+
+# ------------------------------------------------------------- pad-add
+
+class PadAddFn(torch.autograd.Function):
+    """out + zero-channel-padded shortcut (PyramidNet residual,
+    reference pyramidnet.py:109-113) as one NHWC kernel: no zeros tensor,
+    no F.pad copy pair; backward for the shortcut is a zero-copy channel
+    narrow of the incoming gradient."""
+
+    @staticmethod
+    def forward(ctx, out: torch.Tensor, shortcut: torch.Tensor):
+        ctx.cs = shortcut.size(1)
+        C = require_ext_for(out)
+        if C is not None:
+            return C.pad_add(out, shortcut)
+        import torch.nn.functional as F
+        return out + F.pad(shortcut, (0, 0, 0, 0, 0, out.size(1) - shortcut.size(1)))
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        return grad_out, grad_out.narrow(1, 0, ctx.cs)
+
+
+def pad_add(out: torch.Tensor, shortcut: torch.Tensor) -> torch.Tensor:
+    if out.size(1) == shortcut.size(1):
+        return out + shortcut
+    return PadAddFn.apply(out, shortcut)

@@ -33,3 +33,13 @@ def bn_only(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
         from .bnrelu import fused_bn
         return fused_bn(x, bn)
     return bn(x)
+
+
+def bn_swish(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
+    """BatchNorm2d followed by swish, fused on GPU (EfficientNet)."""
+    if (x.is_cuda and has_ext()
+            and os.environ.get("FAA_NO_FUSED_BN") != "1"):
+        from .bnrelu import fused_bn_swish
+        return fused_bn_swish(x, bn)
+    from .functional import swish
+    return swish(bn(x))

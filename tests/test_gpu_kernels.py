@@ -419,3 +419,60 @@ def test_fused_rmsprop_gpu_matches_cpu():
             m(xx).float().square().mean().backward()
             o.step()
     assert (fg.flat_master.cpu() - fc.flat_master).abs().max().item() < 5e-3
+
+
+def test_pad_add_fwd_bwd_vs_torch():
+    """Fused channel-pad residual add (PyramidNet) vs F.pad reference."""
+    import torch.nn.functional as F
+    from fast_autoaugment_amd.ops.functional import pad_add
+    torch.manual_seed(0)
+    for dtype, tol in [(torch.float32, 1e-6), (torch.bfloat16, 1e-2)]:
+        out = torch.randn(4, 37, 8, 8, device=dev(), dtype=dtype).contiguous(
+            memory_format=torch.channels_last).requires_grad_(True)
+        sc = torch.randn(4, 21, 8, 8, device=dev(), dtype=dtype).contiguous(
+            memory_format=torch.channels_last).requires_grad_(True)
+        oref = out.detach().clone().requires_grad_(True)
+        sref = sc.detach().clone().requires_grad_(True)
+        y = pad_add(out, sc)
+        yr = oref + F.pad(sref, (0, 0, 0, 0, 0, 37 - 21))
+        assert (y - yr).abs().max().item() < tol
+        g = torch.randn_like(yr)
+        y.backward(g)
+        yr.backward(g)
+        assert (out.grad - oref.grad).abs().max().item() < tol
+        assert (sc.grad - sref.grad).abs().max().item() < tol
+
+
+@pytest.mark.parametrize("Ch", [32, 61])
+def test_bn_swish_fwd_bwd_vs_torch(Ch):
+    """Fused BN+swish (EfficientNet pattern) vs fp32 torch BN + x*sigmoid."""
+    torch.manual_seed(1)
+    dtype, tol = torch.bfloat16, 3e-2
+    x = torch.randn(8, Ch, 14, 14, device=dev(), dtype=dtype).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    bn = torch.nn.BatchNorm2d(Ch, momentum=0.1, eps=1e-3).to(dev())
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.data.uniform_(-0.5, 0.5)
+    xref = x.detach().float().clone().requires_grad_(True)
+    bnref = torch.nn.BatchNorm2d(Ch, momentum=0.1, eps=1e-3).to(dev())
+    bnref.load_state_dict(bn.state_dict())
+    z = bnref(xref)
+    ref = z * torch.sigmoid(z)
+
+    from fast_autoaugment_amd.ops.bnrelu import fused_bn_swish
+    bn.train()
+    out = fused_bn_swish(x, bn)
+    assert (out.float() - ref).abs().max().item() < tol
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    out.backward(g.to(dtype))
+    assert (x.grad.float() - xref.grad).abs().max().item() < tol * 4
+    assert (bn.weight.grad.float() - bnref.weight.grad).abs().max().item() < tol * 10
+    assert (bn.bias.grad.float() - bnref.bias.grad).abs().max().item() < tol * 10
+    # eval mode
+    bn.eval(); bnref.eval()
+    xe = torch.randn(4, Ch, 7, 7, device=dev(), dtype=dtype).contiguous(
+        memory_format=torch.channels_last)
+    ze = bnref(xe.float())
+    oute = fused_bn_swish(xe, bn)
+    assert (oute.float() - ze * torch.sigmoid(ze)).abs().max().item() < tol
