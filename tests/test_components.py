@@ -318,3 +318,24 @@ def test_tblog_event_file_roundtrip(tmp_path):
     real.add_scalar("x", 1.0, 0)
     real.close()
     assert list((tmp_path / "r2").glob("events.out.tfevents.*"))
+
+
+@pytest.mark.parametrize("conf,nc,millions", [
+    ({"type": "wresnet40_2"}, 10, 2.25),
+    ({"type": "wresnet28_10"}, 100, 36.55),
+    ({"type": "shakeshake26_2x96d"}, 10, 26.19),
+    ({"type": "pyramid", "depth": 272, "alpha": 200, "bottleneck": True}, 10, 26.21),
+    ({"type": "resnet50"}, 1000, 25.56),
+    ({"type": "resnet200"}, 1000, 64.67),
+    ({"type": "efficientnet-b0"}, 1000, 5.29),
+    ({"type": "efficientnet-b1"}, 1000, 7.79),
+    ({"type": "efficientnet-b4"}, 1000, 19.34),
+    ({"type": "shakeshake26_2x96d_next"}, 10, 22.72),
+])
+def test_model_zoo_param_counts(conf, nc, millions):
+    """Architecture parity: parameter counts match the reference models
+    (e.g. WRN-40-2 2.2M, ResNet-50 25.6M, EfficientNet-B0 5.3M)."""
+    from fast_autoaugment_amd.models import build_model
+    m = build_model(conf, nc)
+    n = sum(p.numel() for p in m.parameters())
+    assert abs(n / 1e6 - millions) < 0.02, f"{conf['type']}: {n/1e6:.2f}M"
