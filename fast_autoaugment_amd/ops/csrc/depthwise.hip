@@ -73,6 +73,110 @@ __global__ void dw_fwd_kernel(const short* __restrict__ X, const short* __restri
   }
 }
 
+// Specialized fwd: compile-time K/stride, channel-octet-invariant grid
+// (host aligns gridDim*256 to a multiple of C/8) so each thread's 8-channel
+// weight tap set loads ONCE into registers; interior pixels skip bounds
+// checks; for stride 1 each thread produces PW=2 adjacent outputs from one
+// sliding window (column loads shared between the pair).
+template <bool HAS_BIAS, int KH, int KW, int S, int PW>
+__global__ void dw_fwd_tpl_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
+                                  const short* __restrict__ bias, short* __restrict__ Y,
+                                  DwGeom g) {
+  const int octs = g.C >> 3;
+  const int wop = (g.Wo + PW - 1) / PW;
+  const int64_t items = (int64_t)g.B * g.Ho * wop * octs;
+  int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+  const int oct = (int)(i0 % octs);       // loop-invariant: gstride % octs == 0
+  const int c0 = oct << 3;
+
+  bf16v8 wv[KH * KW];
+  #pragma unroll
+  for (int t = 0; t < KH * KW; ++t) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      wv[t][j] = Wt[(int64_t)(c0 + j) * KH * KW + t];
+  }
+  float bv[8] = {};
+  if (HAS_BIAS) {
+    bf16v8 b8 = *reinterpret_cast<const bf16v8*>(bias + c0);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) bv[j] = dw_b2f(b8[j]);
+  }
+
+  for (int64_t i = i0; i < items; i += gstride) {
+    int64_t m = i / octs;
+    int wo = (int)(m % wop) * PW;
+    int64_t t2 = m / wop;
+    int ho = (int)(t2 % g.Ho);
+    int b = (int)(t2 / g.Ho);
+    int hi0 = ho * S - g.pt;
+    int wi0 = wo * S - g.pl;
+    float acc0[8] = {}, acc1[8] = {};
+    bool pair = (PW == 2) && (wo + 1 < g.Wo);
+    bool interior = hi0 >= 0 && hi0 + KH <= g.H && wi0 >= 0
+                    && wi0 + KW + (PW - 1) * S <= g.W && (PW == 1 || pair);
+    if (interior) {
+      const short* base = X + (((int64_t)b * g.H + hi0) * g.W + wi0) * g.C + c0;
+      #pragma unroll
+      for (int kh = 0; kh < KH; ++kh) {
+        #pragma unroll
+        for (int col = 0; col < KW + (PW - 1) * S; ++col) {
+          bf16v8 xv = *reinterpret_cast<const bf16v8*>(
+              base + ((int64_t)kh * g.W + col) * g.C);
+          if (col < KW) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+              acc0[j] += dw_b2f(xv[j]) * dw_b2f(wv[kh * KW + col][j]);
+          }
+          if (PW == 2 && col >= S && col - S < KW) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+              acc1[j] += dw_b2f(xv[j]) * dw_b2f(wv[kh * KW + col - S][j]);
+          }
+        }
+      }
+    } else {
+      #pragma unroll
+      for (int kh = 0; kh < KH; ++kh) {
+        int hi = hi0 + kh;
+        if (hi < 0 || hi >= g.H) continue;
+        #pragma unroll
+        for (int kw = 0; kw < KW; ++kw) {
+          int wi = wi0 + kw;
+          if (wi >= 0 && wi < g.W) {
+            bf16v8 xv = *reinterpret_cast<const bf16v8*>(
+                X + (((int64_t)b * g.H + hi) * g.W + wi) * g.C + c0);
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+              acc0[j] += dw_b2f(xv[j]) * dw_b2f(wv[kh * KW + kw][j]);
+          }
+          if (PW == 2 && pair) {
+            int wi1 = wi + S;
+            if (wi1 >= 0 && wi1 < g.W) {
+              bf16v8 xv = *reinterpret_cast<const bf16v8*>(
+                  X + (((int64_t)b * g.H + hi) * g.W + wi1) * g.C + c0);
+              #pragma unroll
+              for (int j = 0; j < 8; ++j)
+                acc1[j] += dw_b2f(xv[j]) * dw_b2f(wv[kh * KW + kw][j]);
+            }
+          }
+        }
+      }
+    }
+    short* out = Y + (((int64_t)b * g.Ho + ho) * g.Wo + wo) * g.C + c0;
+    bf16v8 ov;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) ov[j] = dw_f2b(acc0[j] + (HAS_BIAS ? bv[j] : 0.0f));
+    *reinterpret_cast<bf16v8*>(out) = ov;
+    if (PW == 2 && pair) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) ov[j] = dw_f2b(acc1[j] + (HAS_BIAS ? bv[j] : 0.0f));
+      *reinterpret_cast<bf16v8*>(out + g.C) = ov;
+    }
+  }
+}
+
 __global__ void dw_bwd_data_kernel(const short* __restrict__ dY, const short* __restrict__ Wt,
                                    short* __restrict__ dX, DwGeom g) {
   int64_t total = (int64_t)g.B * g.H * g.W * g.C;
@@ -208,12 +312,39 @@ torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   auto y = torch::empty({g.B, g.C, Ho, Wo},
                         xc.options().memory_format(torch::MemoryFormat::ChannelsLast));
   int64_t total = (int64_t)g.B * Ho * Wo * g.C;
-  int grid = faa_grid(total / 8 + 1, 256);
   auto stream = at::hip::getCurrentHIPStream().stream();
   bool hb = bias.defined() && bias.numel() > 0;
   torch::Tensor bc;
   const short* bp = nullptr;
   if (hb) { bc = bias.contiguous(); bp = (const short*)bc.data_ptr(); }
+
+  // specialized path: K in {3,5}, stride in {1,2}; octet-invariant grid
+  bool tpl = (KH == KW) && (KH == 3 || KH == 5) && (stride == 1 || stride == 2);
+  if (tpl) {
+    auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
+    int octs = g.C / 8;
+    int pw = (stride == 1) ? 2 : 1;
+    int64_t items = (int64_t)g.B * Ho * ((Wo + pw - 1) / pw) * octs;
+    int q = octs / gcd(octs, 256);
+    int base = (int)std::min<int64_t>(std::max<int64_t>((items + 255) / 256, 1), 2048);
+    int nb = ((base + q - 1) / q) * q;
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(nb), dim3(256), 0, stream,
+                         (const short*)xc.data_ptr(), (const short*)wc.data_ptr(), bp,
+                         (short*)y.data_ptr(), g);
+    };
+    if (KH == 3 && stride == 1) launch(hb ? dw_fwd_tpl_kernel<true, 3, 3, 1, 2>
+                                          : dw_fwd_tpl_kernel<false, 3, 3, 1, 2>);
+    else if (KH == 3)           launch(hb ? dw_fwd_tpl_kernel<true, 3, 3, 2, 1>
+                                          : dw_fwd_tpl_kernel<false, 3, 3, 2, 1>);
+    else if (stride == 1)       launch(hb ? dw_fwd_tpl_kernel<true, 5, 5, 1, 2>
+                                          : dw_fwd_tpl_kernel<false, 5, 5, 1, 2>);
+    else                        launch(hb ? dw_fwd_tpl_kernel<true, 5, 5, 2, 1>
+                                          : dw_fwd_tpl_kernel<false, 5, 5, 2, 1>);
+    return y;
+  }
+
+  int grid = faa_grid(total / 8 + 1, 256);
   if (hb)
     hipLaunchKernelGGL((dw_fwd_kernel<true>), dim3(grid), dim3(256), 0, stream,
                        (const short*)xc.data_ptr(), (const short*)wc.data_ptr(), bp,
