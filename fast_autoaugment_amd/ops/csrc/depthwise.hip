@@ -318,8 +318,15 @@ torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   const short* bp = nullptr;
   if (hb) { bc = bias.contiguous(); bp = (const short*)bc.data_ptr(); }
 
-  // specialized path: K in {3,5}, stride in {1,2}; octet-invariant grid
-  bool tpl = (KH == KW) && (KH == 3 || KH == 5) && (stride == 1 || stride == 2);
+  // specialized path: K in {3,5}, stride in {1,2}; octet-invariant grid.
+  // k=5 keeps 25 bf16x8 weight vectors in registers (~100 VGPRs) which cuts
+  // occupancy below what the latency-bound loop needs — measured slower on
+  // EfficientNet-B0 — so k=5 stays on the generic kernel unless FAA_DW_TPL5=1.
+  static const bool tpl5 = []() {
+    const char* e = getenv("FAA_DW_TPL5");
+    return e && e[0] == '1';
+  }();
+  bool tpl = (KH == KW) && (KH == 3 || (KH == 5 && tpl5)) && (stride == 1 || stride == 2);
   if (tpl) {
     auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
     int octs = g.C / 8;
