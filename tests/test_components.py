@@ -339,3 +339,32 @@ def test_model_zoo_param_counts(conf, nc, millions):
     m = build_model(conf, nc)
     n = sum(p.numel() for p in m.parameters())
     assert abs(n / 1e6 - millions) < 0.02, f"{conf['type']}: {n/1e6:.2f}M"
+
+
+def test_trainer_only_eval_cpu(tmp_path):
+    """--only-eval path (reference train.py:228-246): load a saved
+    checkpoint and re-evaluate without training."""
+    os.environ["FAA_SYNTH_TRAIN"] = "64"
+    os.environ["FAA_SYNTH_TEST"] = "32"
+    try:
+        from fast_autoaugment_amd.data import api as data_api
+        from fast_autoaugment_amd.engine import train_and_eval
+        data_api._STORE_CACHE.clear()
+        conf = {
+            "model": {"type": "wresnet40_2"}, "dataset": "cifar10",
+            "aug": "default", "cutout": 0, "batch": 32, "epoch": 1,
+            "lr": 0.01,
+            "lr_schedule": {"type": "cosine", "warmup": {"multiplier": 1, "epoch": 0}},
+            "optimizer": {"type": "sgd", "decay": 1e-4, "nesterov": True, "ema": 0},
+        }
+        C.replace(conf)
+        p = str(tmp_path / "m.pth")
+        train_and_eval("", "./data", save_path=p, evaluation_interval=1)
+        r = train_and_eval("", "./data", save_path=p, only_eval=True,
+                           evaluation_interval=1)
+        assert "top1_test" in r or "top1_valid" in r or "top1_train" in r
+    finally:
+        os.environ.pop("FAA_SYNTH_TRAIN", None)
+        os.environ.pop("FAA_SYNTH_TEST", None)
+        from fast_autoaugment_amd.data import api as data_api
+        data_api._STORE_CACHE.clear()
