@@ -28,7 +28,13 @@ logger = get_logger("faa_amd.search")
 
 
 def _model_ckpt_dir() -> str:
-    d = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "..", "models")
+    """Child checkpoints live in <repo>/models like the reference's
+    FastAutoAugment/models (reference search.py:56-57); FAA_MODEL_DIR
+    overrides (tests, scratch runs)."""
+    d = os.environ.get("FAA_MODEL_DIR")
+    if not d:
+        d = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                         "..", "models")
     d = os.path.abspath(d)
     os.makedirs(d, exist_ok=True)
     return d

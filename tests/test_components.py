@@ -368,3 +368,20 @@ def test_trainer_only_eval_cpu(tmp_path):
         os.environ.pop("FAA_SYNTH_TEST", None)
         from fast_autoaugment_amd.data import api as data_api
         data_api._STORE_CACHE.clear()
+
+
+def test_search_phase12_cpu_smoke(tmp_path):
+    """End-to-end phase 1+2 of the search driver on CPU workers (spawn,
+    tiny synthetic set, 2 TPE trials): scheduler, child training, TPE,
+    density-matching eval and policy decoding all wired together."""
+    import subprocess
+    import sys
+    env = dict(os.environ, FAA_SYNTH_TRAIN="300", FAA_SYNTH_TEST="64",
+               FAA_MODEL_DIR=str(tmp_path / "models"))
+    r = subprocess.run(
+        [sys.executable, "tools/search_smoke.py", "--until", "2", "--workers", "1",
+         "--num-search", "2", "--cv-num", "1", "--num-policy", "2", "--batch", "32"],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        env=env, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "search keys:" in r.stdout and "n_pol:" in r.stdout
