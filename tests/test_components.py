@@ -385,3 +385,27 @@ def test_search_phase12_cpu_smoke(tmp_path):
         env=env, capture_output=True, text=True, timeout=420)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "search keys:" in r.stdout and "n_pol:" in r.stdout
+
+
+def test_tpe_sampler_concentrates():
+    """TPE semantics: uniform sampling during startup, then suggestions
+    concentrate on the low-loss region (hyperopt-style gamma split +
+    adaptive Parzen; reference used hyperopt via Ray Tune, search.py:230)."""
+    import numpy as np
+    from fast_autoaugment_amd.search.tpe import TPESampler, policy_search_space
+    space = policy_search_space(num_policy=1, num_op=1, n_ops=8)
+    samp = TPESampler(space, seed=0, n_startup=20)
+    # loss landscape: op index 3 with prob near 0.8 is best
+    def loss(cfg):
+        op = cfg["policy_0_0"]
+        pr = cfg["prob_0_0"]
+        return (0.0 if op == 3 else 1.0) + abs(pr - 0.8)
+    rng = np.random.default_rng(1)
+    for _ in range(80):
+        c = samp.suggest()
+        samp.observe(c, loss(c) + rng.normal(0, 0.01))
+    sugg = [samp.suggest() for _ in range(50)]
+    frac_best_op = sum(1 for s in sugg if s["policy_0_0"] == 3) / len(sugg)
+    assert frac_best_op > 0.5, frac_best_op          # uniform would be 1/8
+    probs = [s["prob_0_0"] for s in sugg if s["policy_0_0"] == 3]
+    assert abs(float(np.median(probs)) - 0.8) < 0.25
