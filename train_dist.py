@@ -50,7 +50,12 @@ def main():
     args, rest = parser.parse_known_args()
 
     if not args.hosts:
-        sys.exit(launch_local(args.np, args.master_addr, args.master_port, 0, 1, rest))
+        # NODE_RANK/NNODES arrive via env when this is the per-host inner
+        # launcher of an ssh fan-out (see below)
+        node_rank = int(os.environ.get("NODE_RANK", "0"))
+        nnodes = int(os.environ.get("NNODES", "1"))
+        sys.exit(launch_local(args.np, args.master_addr, args.master_port,
+                              node_rank, nnodes, rest))
 
     hosts = [h.split(":") for h in args.hosts.split(",")]
     procs = []
