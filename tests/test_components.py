@@ -409,3 +409,21 @@ def test_tpe_sampler_concentrates():
     assert frac_best_op > 0.5, frac_best_op          # uniform would be 1/8
     probs = [s["prob_0_0"] for s in sugg if s["policy_0_0"] == 3]
     assert abs(float(np.median(probs)) - 0.8) < 0.25
+
+
+def test_split_fallback_sparse_classes():
+    """Stratified splits degrade to plain shuffle when a class has < 2
+    members (tiny synthetic reduced_imagenet caps); real-data path stays
+    stratified with random_state=0 like the reference (data.py:192-203)."""
+    import numpy as np
+    from fast_autoaugment_amd.data.split import cv_split, stratified_split
+    # 200 samples over 120 classes -> many singleton classes
+    labels = np.arange(200, dtype=np.int64) % 120
+    tr, va = cv_split(labels, 0.4, 2)
+    assert len(tr) + len(va) == 200 and len(set(tr) & set(va)) == 0
+    tr2, rest = stratified_split(labels, test_size=50)
+    assert len(tr2) == 150 and len(rest) == 50
+    # balanced labels stay stratified: every class represented in train
+    labels_b = np.arange(500, dtype=np.int64) % 10
+    tr3, va3 = cv_split(labels_b, 0.2, 0)
+    assert len(np.unique(labels_b[tr3])) == 10
