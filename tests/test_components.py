@@ -427,3 +427,23 @@ def test_split_fallback_sparse_classes():
     labels_b = np.arange(500, dtype=np.int64) % 10
     tr3, va3 = cv_split(labels_b, 0.2, 0)
     assert len(np.unique(labels_b[tr3])) == 10
+
+
+def test_all_archives_compile_and_execute():
+    """Every shipped policy archive (FA + AutoAugment-compat tables,
+    reference archive.py:281-293 + 59-87) compiles into op programs and
+    executes through the CPU pipeline."""
+    from fast_autoaugment_amd import policies
+    from fast_autoaugment_amd.aug import ops as aug_ops, cpu_exec
+    rng = np.random.default_rng(0)
+    imgs = rng.integers(0, 256, (8, 32, 32, 3), dtype=np.uint8)
+    mean = np.zeros(3, np.float32)
+    std = np.ones(3, np.float32)
+    for name in ["fa_reduced_cifar10", "fa_reduced_svhn", "fa_resnet50_rimagenet",
+                 "arsaug_policy", "autoaug_policy", "autoaug_paper_cifar10"]:
+        pol = policies.get_archive(name)
+        prog = aug_ops.compile_program_fast(pol, 8, 32, 32, rng)
+        post = aug_ops.compile_post_fast(8, 32, 32, rng, pad=4, cutout_len=16,
+                                         train=True)
+        out = cpu_exec.run_pipeline_cpu(imgs, prog, post, mean, std)
+        assert out.shape == (8, 32, 32, 3) and np.isfinite(out).all(), name
