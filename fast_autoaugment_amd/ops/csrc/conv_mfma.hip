@@ -67,11 +67,15 @@ __device__ __forceinline__ bf16x8 load_x8(const short* __restrict__ X,
 
 // ------------------------------------------------------------------- fwd
 
-template <bool HAS_BIAS, int BMT, int BNT>
+// SPLIT=false: each workgroup owns a full K loop and writes bf16 Y directly.
+// SPLIT=true (occupancy aid for deep stages whose tile grid underfills the
+// 256 CUs): blockIdx.y partitions the K tiles; partial C-tiles accumulate
+// into the fp32 workspace Yacc with atomics and a bias+cast kernel finishes.
+template <bool HAS_BIAS, int BMT, int BNT, bool SPLIT = false>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
                      const short* __restrict__ bias, short* __restrict__ Y,
-                     ConvGeom g, int M, int grid_m) {
+                     ConvGeom g, int M, int grid_m, float* __restrict__ Yacc = nullptr) {
   // 4 waves in a fixed 2x2 grid; per-wave sub-tile (BMT/2) x (BNT/2),
   // i.e. MFRAG=BMT/32 x NFRAG=BNT/32 fragments of 16x16. Smaller tiles
   // multiply the workgroup count for shapes that underfill 256 CUs
@@ -116,6 +120,13 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
   const int n_g = bn * BNT + b_row;
 
   const int nk = g.kpad / BK;
+  int kt_lo = 0, kt_hi = nk;
+  if (SPLIT) {
+    int per = (nk + gridDim.y - 1) / gridDim.y;
+    kt_lo = blockIdx.y * per;
+    kt_hi = min(nk, kt_lo + per);
+    if (kt_lo >= kt_hi) return;
+  }
   f32x4 acc[MFRAG][NFRAG] = {};
   const int CELLS = g.KH * g.KW;
 
@@ -156,16 +167,16 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
       *reinterpret_cast<bf16x8*>(&ldsB[buf][b_row * LDSP + a_kc * 8]) = vb;
   };
 
-  write_lds(0, load_a(0), load_b(0));
+  write_lds(0, load_a(kt_lo), load_b(kt_lo));
   __syncthreads();
 
   const int fr = lane & 15;
   const int kq = (lane >> 4) * 8;
 
-  for (int kt = 0; kt < nk; ++kt) {
-    int buf = kt & 1;
+  for (int kt = kt_lo; kt < kt_hi; ++kt) {
+    int buf = (kt - kt_lo) & 1;
     bf16x8 na = {}, nb = {};
-    if (kt + 1 < nk) {      // issue next-tile global loads BEFORE the MFMAs
+    if (kt + 1 < kt_hi) {   // issue next-tile global loads BEFORE the MFMAs
       na = load_a(kt + 1);
       nb = load_b(kt + 1);
     }
@@ -182,7 +193,7 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
       for (int fn = 0; fn < NFRAG; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[fm], bfrag[fn],
                                                               acc[fm][fn], 0, 0, 0);
-    if (kt + 1 < nk) {
+    if (kt + 1 < kt_hi) {
       __syncthreads();
       write_lds(buf ^ 1, na, nb);
     }
@@ -201,11 +212,27 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
         int n = bn * BNT + wc + fn * 16 + fr;
         if (n >= g.Cout) continue;
         float v = acc[fm][fn][r];
-        if (HAS_BIAS) v += b2f(bias[n]);
-        Y[(int64_t)m * g.Cout + n] = f2b(v);
+        if (SPLIT) {
+          atomicAdd(&Yacc[(int64_t)m * g.Cout + n], v);
+        } else {
+          if (HAS_BIAS) v += b2f(bias[n]);
+          Y[(int64_t)m * g.Cout + n] = f2b(v);
+        }
       }
     }
   }
+}
+
+// bias-add + fp32 -> bf16 cast for the split-K path
+template <bool HAS_BIAS>
+__global__ void conv_splitk_cast_kernel(const float* __restrict__ Yacc,
+                                        const short* __restrict__ bias,
+                                        short* __restrict__ Y, int64_t total, int Cout) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  float v = Yacc[i];
+  if (HAS_BIAS) v += b2f(bias[(int)(i % Cout)]);
+  Y[i] = f2b(v);
 }
 
 // ------------------------------------------------- weight repack (bwd-data)
@@ -415,6 +442,45 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     bc = bias.contiguous();
     TORCH_CHECK(bc.scalar_type() == torch::kBFloat16);
     bptr = (const short*)bc.data_ptr();
+  }
+  // FAA_CONV_SPLITK=1: partition K across blockIdx.y when the tile grid
+  // underfills the chip (deep 8x8 stages run ~256 WGs = 1/CU; PMC showed
+  // 9% occupancy there). fp32-atomic workspace + bias/cast epilogue.
+  // Env-gated default-off pending round-2 measurement.
+  {
+    const char* e = getenv("FAA_CONV_SPLITK");
+    int nk = g.kpad / BK;
+    int blocks = grid_m * grid_n;
+    if (e && e[0] == '1' && blocks < 1024 && nk >= 2) {
+      int sk = std::min(std::min((1024 + blocks - 1) / blocks, nk), 8);
+      if (sk > 1) {
+        auto yacc = torch::zeros({(int64_t)M, (int64_t)g.Cout},
+                                 xc.options().dtype(torch::kFloat32));
+        dim3 grid2(grid_m * grid_n, sk);
+        #define CF_LAUNCH_SK(BMT_, BNT_)                                         \
+          hipLaunchKernelGGL((conv_fwd_kernel<false, BMT_, BNT_, true>), grid2,  \
+                             dim3(256), 0, stream, (const short*)xc.data_ptr(),  \
+                             (const short*)wc.data_ptr(), nullptr,               \
+                             (short*)y.data_ptr(), g, M, grid_m,                 \
+                             yacc.data_ptr<float>())
+        if (bmt == 32 && bnt == 32) CF_LAUNCH_SK(32, 32);
+        else if (bmt == 32) CF_LAUNCH_SK(32, 64);
+        else if (bnt == 32) CF_LAUNCH_SK(64, 32);
+        else CF_LAUNCH_SK(64, 64);
+        #undef CF_LAUNCH_SK
+        int64_t total = (int64_t)M * g.Cout;
+        dim3 cgrid((unsigned)((total + 255) / 256));
+        if (has_bias)
+          hipLaunchKernelGGL((conv_splitk_cast_kernel<true>), cgrid, dim3(256), 0,
+                             stream, yacc.data_ptr<float>(), bptr,
+                             (short*)y.data_ptr(), total, g.Cout);
+        else
+          hipLaunchKernelGGL((conv_splitk_cast_kernel<false>), cgrid, dim3(256), 0,
+                             stream, yacc.data_ptr<float>(), nullptr,
+                             (short*)y.data_ptr(), total, g.Cout);
+        return y;
+      }
+    }
   }
   #define CF_LAUNCH(HB, BMT_, BNT_)                                            \
     hipLaunchKernelGGL((conv_fwd_kernel<HB, BMT_, BNT_>), grid, dim3(256), 0,  \
