@@ -147,3 +147,26 @@ def test_depthwise_fwd_bwd(C, H, k, stride):
 def C_ext():
     from fast_autoaugment_amd.ops import ext
     return ext()
+
+
+def test_conv_fwd_splitk_matches():
+    """FAA_CONV_SPLITK=1 K-partitioned fwd (fp32-atomic workspace) vs torch
+    (validated in round 1, tools/splitk_check.py; default-off pending
+    round-2 measurement)."""
+    import os
+    os.environ["FAA_CONV_SPLITK"] = "1"
+    try:
+        torch.manual_seed(0)
+        for B, Cin, H, Cout, k, s in [(128, 64, 8, 64, 3, 1), (8, 32, 8, 64, 3, 2)]:
+            x = torch.randn(B, Cin, H, H, device=dev()) * 0.5
+            w = torch.randn(Cout, Cin, k, k, device=dev()) * 0.05
+            b = torch.randn(Cout, device=dev()) * 0.1
+            ref = torch.nn.functional.conv2d(x, w, b, stride=s, padding=k // 2)
+            got = C_ext().conv2d_fwd(
+                x.bfloat16().contiguous(memory_format=torch.channels_last),
+                w.bfloat16().contiguous(memory_format=torch.channels_last),
+                b.bfloat16(), s, k // 2).float()
+            err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+            assert err < 2e-2, f"splitk rel err {err}"
+    finally:
+        os.environ.pop("FAA_CONV_SPLITK", None)
