@@ -431,6 +431,15 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     };
     if (blocks(64, bnt) < 1024 && blocks(32, bnt) <= 4096) bmt = 32;
   }
+  // FAA_CONV_TILE=BMxBN forces a tile pair (tuning sweeps, tools/conv_tune.py)
+  if (const char* e = getenv("FAA_CONV_TILE")) {
+    int fm = 0, fn = 0;
+    if (sscanf(e, "%dx%d", &fm, &fn) == 2 && (fm == 32 || fm == 64)
+        && (fn == 32 || fn == 64)) {
+      bmt = fm;
+      bnt = fn;
+    }
+  }
   int grid_m = (M + bmt - 1) / bmt;
   int grid_n = (g.Cout + bnt - 1) / bnt;
   dim3 grid(grid_m * grid_n);

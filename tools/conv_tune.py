@@ -1,0 +1,65 @@
+#!/usr/bin/env python3
+"""Round-2 conv tuning sweep: per-shape fwd time across tile forcings and
+split-K, vs MIOpen. Run on a GPU box:
+
+  python tools/conv_tune.py            # flagship + deep-stage shapes
+  python tools/conv_tune.py --big      # WRN-28-10 / shake C>=256 shapes
+"""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, ".")
+import torch
+
+
+def bench(fn, iters=50):
+    for _ in range(5):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--big", action="store_true")
+    ap.add_argument("--iters", type=int, default=50)
+    args = ap.parse_args()
+    from fast_autoaugment_amd.ops import ext
+    C = ext()
+    shapes = ([(128, 256, 16, 256, 3, 1), (128, 384, 8, 384, 3, 1),
+               (128, 640, 8, 640, 3, 1), (128, 320, 16, 320, 3, 1)]
+              if args.big else
+              [(128, 16, 32, 32, 3, 1), (128, 32, 32, 32, 3, 1),
+               (128, 64, 16, 64, 3, 1), (128, 128, 8, 128, 3, 1),
+               (128, 64, 16, 128, 3, 2)])
+    print(f"{'shape':<28} {'miopen':>8} {'auto':>8} "
+          + " ".join(f"{t:>8}" for t in ["32x32", "32x64", "64x32", "64x64"])
+          + f" {'auto+sk':>8}")
+    for B, Cin, H, Cout, k, s in shapes:
+        x = (torch.randn(B, Cin, H, H, device="cuda") * 0.5).bfloat16() \
+            .contiguous(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, k, k, device="cuda") * 0.05).bfloat16() \
+            .contiguous(memory_format=torch.channels_last)
+        row = [f"{B}x{Cin}x{H}^2->{Cout} k{k}s{s}"]
+        row.append(f"{bench(lambda: torch.nn.functional.conv2d(x, w, stride=s, padding=k//2), args.iters):8.1f}")
+        os.environ.pop("FAA_CONV_TILE", None)
+        os.environ.pop("FAA_CONV_SPLITK", None)
+        row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
+        for tile in ["32x32", "32x64", "64x32", "64x64"]:
+            os.environ["FAA_CONV_TILE"] = tile
+            row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
+        os.environ.pop("FAA_CONV_TILE", None)
+        os.environ["FAA_CONV_SPLITK"] = "1"
+        row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
+        os.environ.pop("FAA_CONV_SPLITK", None)
+        print(" ".join(row))
+
+
+if __name__ == "__main__":
+    main()
