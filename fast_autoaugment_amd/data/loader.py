@@ -132,20 +132,35 @@ class AugLoader:
             import queue as _q
             import threading
             q: "_q.Queue" = _q.Queue(maxsize=self.prefetch)
+            stop = threading.Event()
 
             def producer():
                 for b in range(nb):
-                    q.put(self._gen_host(idx, rng, b))
+                    item = self._gen_host(idx, rng, b)
+                    while not stop.is_set():
+                        try:
+                            q.put(item, timeout=0.5)
+                            break
+                        except _q.Full:
+                            continue
+                    if stop.is_set():
+                        return
                 q.put(None)
 
             t = threading.Thread(target=producer, daemon=True)
             t.start()
-            while True:
-                item = q.get()
-                if item is None:
-                    break
-                yield self._make_batch(*item)
-            t.join()
+            try:
+                while True:
+                    item = q.get()
+                    if item is None:
+                        break
+                    yield self._make_batch(*item)
+                t.join()
+            finally:
+                # consumer abandoned the epoch (exception / early break):
+                # unblock and reap the producer instead of leaking it
+                stop.set()
+                t.join(timeout=2)
         else:
             for b in range(nb):
                 yield self._make_batch(*self._gen_host(idx, rng, b))
