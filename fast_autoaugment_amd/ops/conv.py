@@ -1,11 +1,17 @@
-"""Custom MFMA conv dispatch.
+"""Custom conv dispatch (the reference delegates all conv to cuDNN,
+SURVEY.md §2.6).
 
-`patch_convs(model)` rebinds every eligible nn.Conv2d's forward to the
-in-house CDNA4 implicit-GEMM kernels (csrc/conv_mfma.hip): bf16 NHWC,
-square kernel 1 or 3 with SAME-style padding, groups=1, dilation=1.
-Ineligible convs (grouped/depthwise, exotic kernels, non-bf16) keep the
-torch/MIOpen path. The stride-2 backward-data piece also falls back to
-torch.nn.grad.conv2d_input until the scatter variant lands.
+`patch_convs(model)` rebinds eligible nn.Conv2d forwards to in-house
+CDNA4 kernels:
+  * depthwise (groups == Cin == Cout, k in {3,5}, C % 8 == 0) ->
+    csrc/depthwise.hip streaming kernels (EfficientNet's MBConv);
+  * dense bf16 NHWC, square kernel 1/3, groups=1, dilation=1 ->
+    csrc/conv_mfma.hip implicit-GEMM MFMA kernels.
+The dense dispatch thresholds below are MEASURED per shape
+(profiles/conv_bench_r01.txt); everything else — including stride-2
+backward-data — keeps the torch/MIOpen path where it currently wins.
+Env knobs: FAA_NO_PATCH, FAA_WRW, FAA_BWD_DATA, FAA_CONV_SPLITK,
+FAA_CONV_TILE, FAA_DW_TPL5.
 """
 from __future__ import annotations
 
