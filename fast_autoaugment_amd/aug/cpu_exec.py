@@ -8,7 +8,7 @@ tests/test_aug_pil_golden.py.
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 

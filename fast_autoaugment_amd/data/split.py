@@ -6,7 +6,7 @@ when run on the real datasets.
 """
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import Tuple
 
 import numpy as np
 from sklearn.model_selection import ShuffleSplit, StratifiedShuffleSplit

@@ -9,7 +9,6 @@ optimizer's param_groups.
 from __future__ import annotations
 
 import math
-from typing import List, Optional, Sequence
 
 
 class _Schedule:

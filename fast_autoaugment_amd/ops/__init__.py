@@ -13,7 +13,6 @@ for the kernels.
 from __future__ import annotations
 
 import importlib
-import os
 
 import torch
 

@@ -16,8 +16,6 @@ FAA_CONV_TILE, FAA_DW_TPL5.
 from __future__ import annotations
 
 import types
-from typing import Optional
-
 import torch
 import torch.nn.functional as F
 

@@ -15,8 +15,6 @@ plain torch, so fixed-seed comparisons validate the kernels.
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
-
 import torch
 
 from . import require_ext_for

@@ -16,8 +16,6 @@ No host sync anywhere, so the whole step is hipGraph-capturable.
 """
 from __future__ import annotations
 
-from typing import Iterable, Optional
-
 import torch
 from torch.optim.optimizer import Optimizer
 

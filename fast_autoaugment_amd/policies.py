@@ -17,7 +17,7 @@ from __future__ import annotations
 import json
 import os
 from functools import lru_cache
-from typing import Dict, List, Sequence, Tuple
+from typing import Dict, List, Tuple
 
 SubPolicy = List[Tuple[str, float, float]]
 Policy = List[SubPolicy]

@@ -14,7 +14,6 @@ from __future__ import annotations
 import copy
 import json
 import os
-import time
 from typing import Dict, List, Optional
 
 from ..common import Stopwatch, get_logger
