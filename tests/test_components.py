@@ -447,3 +447,29 @@ def test_all_archives_compile_and_execute():
                                          train=True)
         out = cpu_exec.run_pipeline_cpu(imgs, prog, post, mean, std)
         assert out.shape == (8, 32, 32, 3) and np.isfinite(out).all(), name
+
+
+def test_program_compilers_distribution_equivalent():
+    """Scalar and vectorized program compilers draw from the same
+    distribution (RNG call order differs by design; op frequencies and
+    deterministic-level parameters must agree)."""
+    from fast_autoaugment_amd import policies
+    from fast_autoaugment_amd.aug import ops as aug_ops
+    pol = policies.get_archive("fa_reduced_cifar10")[:50]
+    B, N = 256, 20
+    def freqs(fn):
+        cs = [fn(pol, B, 32, 32, np.random.default_rng(1000 + s))[:, :, 0].ravel()
+              for s in range(N)]
+        c = np.concatenate(cs).astype(int)
+        return np.bincount(c, minlength=13) / len(c)
+    fa = freqs(aug_ops.compile_program)
+    fb = freqs(aug_ops.compile_program_fast)
+    assert np.abs(fa - fb).max() < 0.01
+    # deterministic-level ops: params must match exactly in distribution
+    for opname, want in [("Solarize", 179.2), ("Posterize", 6.0), ("Contrast", 1.54)]:
+        for fn in (aug_ops.compile_program, aug_ops.compile_program_fast):
+            ps = np.concatenate([fn([[(opname, 1.0, 0.7 if opname == "Solarize" else
+                                       0.5 if opname == "Posterize" else 0.8)]],
+                                    128, 32, 32, np.random.default_rng(s))[:, 0, 1]
+                                 for s in range(10)])
+            assert abs(ps.mean() - want) < 0.05, (opname, fn.__name__, ps.mean())
