@@ -228,9 +228,14 @@ _COMPILED_CACHE = {}
 
 
 def _compiled(policy) -> CompiledPolicy:
-    key = id(policy)
+    # keyed on CONTENT: id(policy) is unsafe — TPE trials decode fresh
+    # equal-length policy lists and CPython recycles ids, which silently
+    # evaluated a stale policy (caught by the compiler-equivalence test)
+    key = tuple(tuple((n, float(p), float(l)) for (n, p, l) in sub) for sub in policy)
     cp = _COMPILED_CACHE.get(key)
-    if cp is None or cp.n_sub != len(policy):
+    if cp is None:
+        if len(_COMPILED_CACHE) > 512:
+            _COMPILED_CACHE.clear()
         cp = CompiledPolicy(policy)
         _COMPILED_CACHE[key] = cp
     return cp
