@@ -28,12 +28,63 @@ __device__ __forceinline__ float bf16_bits_to_f(short u) {
   return c.f;
 }
 
+// ------------------------------------------------- last-block finalize
+// FAA_BN_LASTBLOCK=1 (staged for round 2, default off): the LAST reduce
+// block to finish performs the finalize inline instead of a separate
+// 4.7us kernel launch per BN site (~66 launches/step on WRN-40-2).
+// Pattern: write partials -> __threadfence -> atomicAdd(counter); the last
+// arrival re-reduces scratch and resets the counter, so the persistent
+// counter never needs per-call zeroing.
+__device__ __forceinline__ void bn_finalize_inblock(
+    const float* __restrict__ scratch, int nblocks, float* __restrict__ mean,
+    float* __restrict__ invstd, float* __restrict__ running_mean,
+    float* __restrict__ running_var, int C, int64_t count, float eps,
+    float momentum) {
+  for (int c = threadIdx.x; c < C; c += (int)blockDim.x) {
+    float s = 0, ss = 0;
+    for (int b = 0; b < nblocks; ++b) {
+      s += scratch[(int64_t)b * 2 * C + c];
+      ss += scratch[(int64_t)b * 2 * C + C + c];
+    }
+    float m = s / count;
+    float var = fmaxf(ss / count - m * m, 0.0f);
+    mean[c] = m;
+    invstd[c] = rsqrtf(var + eps);
+    if (running_mean != nullptr) {
+      float unbiased = count > 1 ? var * (float)count / (float)(count - 1) : var;
+      running_mean[c] = (1.0f - momentum) * running_mean[c] + momentum * m;
+      running_var[c] = (1.0f - momentum) * running_var[c] + momentum * unbiased;
+    }
+  }
+}
+
+struct BnFinTail {          // nullptr counter = disabled
+  unsigned* counter;
+  float* mean; float* invstd; float* running_mean; float* running_var;
+  int64_t count; float eps; float momentum;
+};
+
+__device__ __forceinline__ void bn_fin_tail_run(const BnFinTail& t,
+                                                const float* scratch, int C) {
+  if (t.counter == nullptr) return;
+  __threadfence();
+  __shared__ unsigned order;
+  if (threadIdx.x == 0) order = atomicAdd(t.counter, 1u);
+  __syncthreads();
+  if (order == gridDim.x - 1) {
+    bn_finalize_inblock(scratch, gridDim.x, t.mean, t.invstd, t.running_mean,
+                        t.running_var, C, t.count, t.eps, t.momentum);
+    __syncthreads();
+    if (threadIdx.x == 0) *t.counter = 0;
+  }
+}
+
 // ---------------------------------------------------------------- fwd reduce
 // block: 256 threads; each thread owns channels [c0, c0+8) with
 // c0 = (tid*8) % C. partials: scratch[blockIdx.x*2C + {c, C+c}]
 template <typename T>
 __global__ void bn_reduce_vec_kernel(const T* __restrict__ x, float* __restrict__ scratch,
-                                     int64_t total, int C) {
+                                     int64_t total, int C, BnFinTail tail = {}) {
   // host guarantees (gridDim*blockDim*8) % C == 0, so each thread's channel
   // octet c0 is FIXED across its grid-stride loop.
   float s[8] = {0}, ss[8] = {0};
@@ -92,6 +143,7 @@ __global__ void bn_reduce_vec_kernel(const T* __restrict__ x, float* __restrict_
       acc += lds[t * 8 + lane];
     out[C + ch] = acc;
   }
+  bn_fin_tail_run(tail, scratch, C);
 }
 
 // any-C scalar reduce (C % 8 != 0, e.g. PyramidNet's rounded widths): the
@@ -100,7 +152,7 @@ __global__ void bn_reduce_vec_kernel(const T* __restrict__ x, float* __restrict_
 // folded through LDS (no atomics) and finalized like the vec path.
 template <typename T>
 __global__ void bn_reduce_anyc_kernel(const T* __restrict__ x, float* __restrict__ scratch,
-                                      int64_t total, int C) {
+                                      int64_t total, int C, BnFinTail tail = {}) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   float s = 0, ss = 0;
@@ -130,6 +182,7 @@ __global__ void bn_reduce_anyc_kernel(const T* __restrict__ x, float* __restrict
     for (int t = t0; t < (int)blockDim.x; t += C) acc += lds[t];
     out[C + c] = acc;
   }
+  bn_fin_tail_run(tail, scratch, C);
 }
 
 // one 64-lane wave per channel: lanes stride the partial blocks in parallel
@@ -472,6 +525,22 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict
 
 // partial-sum block count: the grid stride (nblocks*256*8) must be a
 // multiple of C so each thread's channel octet is loop-invariant.
+// persistent device counter for the last-block finalize (allocated on
+// first eager use, i.e. before any graph capture; reset by the last block)
+static unsigned* bn_lastblock_counter() {
+  static unsigned* p = nullptr;
+  if (p == nullptr) {
+    (void)hipMalloc(&p, sizeof(unsigned));
+    (void)hipMemset(p, 0, sizeof(unsigned));
+  }
+  return p;
+}
+
+static bool bn_lastblock_enabled() {
+  const char* e = getenv("FAA_BN_LASTBLOCK");
+  return e && e[0] == '1';
+}
+
 static int bn_nblocks(int C, int64_t total) {
   auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
   int q = C / gcd(C, 2048);
@@ -519,34 +588,48 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
 
   if (training) {
     bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
+    BnFinTail tail{};
+    bool lastblock = bn_lastblock_enabled();
+    if (lastblock) {
+      tail.counter = bn_lastblock_counter();
+      tail.mean = mean.data_ptr<float>();
+      tail.invstd = invstd.data_ptr<float>();
+      tail.running_mean = running_mean.defined() ? running_mean.data_ptr<float>() : nullptr;
+      tail.running_var = running_var.defined() ? running_var.data_ptr<float>() : nullptr;
+      tail.count = rows;
+      tail.eps = (float)eps;
+      tail.momentum = (float)momentum;
+    }
     if (vec) {
       int nb = bn_nblocks(C, total);
       auto scratch = torch::empty({nb, 2 * C}, f32);
       DISPATCH_FB(xc.scalar_type(), "bn_reduce", [&] {
         hipLaunchKernelGGL((bn_reduce_vec_kernel<scalar_t>), dim3(nb), dim3(256), 0,
                            stream, (const scalar_t*)xc.data_ptr(),
-                           scratch.data_ptr<float>(), total, C);
+                           scratch.data_ptr<float>(), total, C, tail);
       });
-      hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
-                         scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(),
-                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
-                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
-                         C, rows, (float)eps, (float)momentum);
+      if (!lastblock)
+        hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
+                           scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(),
+                           running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                           running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                           C, rows, (float)eps, (float)momentum);
     } else {
       int nb = bn_nblocks_anyc(C, total);
       auto scratch = torch::empty({nb, 2 * C}, f32);
       DISPATCH_FB(xc.scalar_type(), "bn_reduce_s", [&] {
         hipLaunchKernelGGL((bn_reduce_anyc_kernel<scalar_t>), dim3(nb), dim3(256), 0,
                            stream, (const scalar_t*)xc.data_ptr(),
-                           scratch.data_ptr<float>(), total, C);
+                           scratch.data_ptr<float>(), total, C, tail);
       });
-      hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
-                         scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(),
-                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
-                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
-                         C, rows, (float)eps, (float)momentum);
+      if (!lastblock)
+        hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
+                           scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(),
+                           running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                           running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                           C, rows, (float)eps, (float)momentum);
     }
   } else {
     mean.copy_(running_mean);

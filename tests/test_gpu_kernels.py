@@ -476,3 +476,4 @@ def test_bn_swish_fwd_bwd_vs_torch(Ch):
     ze = bnref(xe.float())
     oute = fused_bn_swish(xe, bn)
     assert (oute.float() - ze * torch.sigmoid(ze)).abs().max().item() < tol
+
