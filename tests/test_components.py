@@ -473,3 +473,18 @@ def test_program_compilers_distribution_equivalent():
                                     128, 32, 32, np.random.default_rng(s))[:, 0, 1]
                                  for s in range(10)])
             assert abs(ps.mean() - want) < 0.05, (opname, fn.__name__, ps.mean())
+
+
+def test_xcd_swizzle_bijective():
+    """The conv kernels' blockIdx->tile XCD swizzle (conv_mfma.hip) must be
+    a bijection for EVERY grid size or tiles would be dropped/duplicated.
+    Python mirror of the device arithmetic, exhaustive over realistic grids."""
+    def swizzle(wg, nwg):
+        q, r = divmod(nwg, 8)
+        xcd, idx = wg % 8, wg // 8
+        if q > 0:
+            return xcd * (q + 1) + idx if xcd < r else r * (q + 1) + (xcd - r) * q + idx
+        return wg
+    for nwg in list(range(1, 600)) + [1024, 2047, 2048, 4095, 4096, 8191]:
+        seen = {swizzle(w, nwg) for w in range(nwg)}
+        assert len(seen) == nwg and min(seen) == 0 and max(seen) == nwg - 1, nwg
