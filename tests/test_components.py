@@ -488,3 +488,40 @@ def test_xcd_swizzle_bijective():
     for nwg in list(range(1, 600)) + [1024, 2047, 2048, 4095, 4096, 8191]:
         seen = {swizzle(w, nwg) for w in range(nwg)}
         assert len(seen) == nwg and min(seen) == 0 and max(seen) == nwg - 1, nwg
+
+
+def test_bn_fold_ownership_math():
+    """CPU mirror of the BN reduce LDS-fold indexing (bnrelu.hip): for every
+    channel, the fold loop's thread set must equal the set of threads whose
+    (channel-invariant) octet/channel owns it — the exact invariant whose
+    violation caused the C>256 partial-fold bug."""
+    def check_vec(C, bid, blockDim=256):
+        groups = C // 8
+        t = np.arange(blockDim)
+        own_oct = ((bid * blockDim + t) * 8 % C) // 8
+        shift = (bid * blockDim) % groups
+        for ch in range(C):
+            oct_ = ch // 8
+            t0 = (oct_ - shift + groups) % groups
+            fold = set(range(t0, blockDim, groups))
+            owners = set(np.nonzero(own_oct == oct_)[0].tolist())
+            if fold != owners:
+                return False
+        return True
+
+    def check_anyc(C, bid, blockDim=256):
+        t = np.arange(blockDim)
+        own_c = (bid * blockDim + t) % C
+        shift = (bid * blockDim) % C
+        for ch in range(C):
+            t0 = (ch - shift + C) % C
+            if set(range(t0, blockDim, C)) != set(np.nonzero(own_c == ch)[0].tolist()):
+                return False
+        return True
+
+    for C in list(range(8, 129, 8)) + [256, 512, 640, 2048]:
+        for bid in (0, 1, 13):
+            assert check_vec(C, bid), ("vec", C, bid)
+    for C in [3, 21, 61, 340, 485, 850]:
+        for bid in (0, 1, 13):
+            assert check_anyc(C, bid), ("anyc", C, bid)
