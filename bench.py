@@ -51,6 +51,11 @@ def main():
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if args.gpus > 1 and world_size == 1:
+        raise SystemExit(
+            f"--gpus {args.gpus} needs a launcher: python -m torch.distributed.run "
+            f"--nnodes=1 --nproc-per-node {args.gpus} --master-addr 127.0.0.1 "
+            f"bench.py --gpus {args.gpus} ...")
     # FAA_BENCH_FORCE_DIST=1: take the distributed path at world_size==1
     # (single-GPU validation of init_process_group/broadcast/all_reduce)
     distributed = world_size > 1 or os.environ.get("FAA_BENCH_FORCE_DIST") == "1"
