@@ -79,6 +79,34 @@ __device__ __forceinline__ void bn_fin_tail_run(const BnFinTail& t,
   }
 }
 
+template <typename GT>
+struct BnBwdFinTail {      // nullptr counter = disabled
+  unsigned* counter;
+  float* sums; GT* dbeta; GT* dgamma;
+};
+
+template <typename GT>
+__device__ __forceinline__ void bn_bwd_fin_tail_run(const BnBwdFinTail<GT>& t,
+                                                    const float* scratch, int C) {
+  if (t.counter == nullptr) return;
+  __threadfence();
+  __shared__ unsigned order;
+  if (threadIdx.x == 0) order = atomicAdd(t.counter, 1u);
+  __syncthreads();
+  if (order == gridDim.x - 1) {
+    for (int c = threadIdx.x; c < 2 * C; c += (int)blockDim.x) {
+      float s = 0;
+      for (int b = 0; b < (int)gridDim.x; ++b)
+        s += scratch[(int64_t)b * 2 * C + c];
+      t.sums[c] = s;
+      if (c < C) faa_from_float(s, &t.dbeta[c]);
+      else faa_from_float(s, &t.dgamma[c - C]);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) *t.counter = 0;
+  }
+}
+
 // ---------------------------------------------------------------- fwd reduce
 // block: 256 threads; each thread owns channels [c0, c0+8) with
 // c0 = (tid*8) % C. partials: scratch[blockIdx.x*2C + {c, C+c}]
@@ -297,7 +325,8 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
                                          const GT* __restrict__ gamma,
                                          const GT* __restrict__ beta,
                                          float* __restrict__ scratch,
-                                         int64_t total, int C) {
+                                         int64_t total, int C,
+                                         BnBwdFinTail<GT> tail = {}) {
   float sdy[8] = {0}, sdyx[8] = {0};
   const int c0 = (int)((((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8) % C);
   float m[8], is[8], ga[8], be[8];
@@ -371,6 +400,7 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ x, const T* __res
     for (int t = t0; t < (int)blockDim.x; t += groups) acc += lds[t * 8 + lane];
     outp[C + ch] = acc;
   }
+  bn_bwd_fin_tail_run(tail, scratch, C);
 }
 
 // any-C bwd reduce: same channel-invariant decomposition as bn_reduce_anyc.
@@ -382,7 +412,8 @@ __global__ void bn_bwd_reduce_anyc_kernel(const T* __restrict__ x, const T* __re
                                           const GT* __restrict__ gamma,
                                           const GT* __restrict__ beta,
                                           float* __restrict__ scratch,
-                                          int64_t total, int C) {
+                                          int64_t total, int C,
+                                          BnBwdFinTail<GT> tail = {}) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const int cown = (int)(i0 % C);
@@ -416,6 +447,7 @@ __global__ void bn_bwd_reduce_anyc_kernel(const T* __restrict__ x, const T* __re
     for (int t = t0; t < (int)blockDim.x; t += C) acc += lds[t];
     outp[C + c] = acc;
   }
+  bn_bwd_fin_tail_run(tail, scratch, C);
 }
 
 template <typename GT>
@@ -675,9 +707,17 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
   bool vec = (C % 8 == 0) && (total % 8 == 0) && (C <= 2048);
   int nb = vec ? bn_nblocks(C, total) : bn_nblocks_anyc(C, total);
   auto scratch = torch::empty({nb, 2 * C}, f32);
+  bool lastblock = bn_lastblock_enabled();
   DISPATCH_FB(xc.scalar_type(), "bn_bwd_reduce", [&] {
     using data_t = scalar_t;
     DISPATCH_FB(g.scalar_type(), "bn_bwd_reduce_g", [&] {
+      BnBwdFinTail<scalar_t> tail{};
+      if (lastblock) {
+        tail.counter = bn_lastblock_counter();
+        tail.sums = sums.data_ptr<float>();
+        tail.dbeta = (scalar_t*)dbeta.data_ptr();
+        tail.dgamma = (scalar_t*)dgamma.data_ptr();
+      }
       act_dispatch((int)act, [&](auto A) {
         constexpr int kAct = decltype(A)::value;
         if (vec)
@@ -687,7 +727,7 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                              (const data_t*)dyc.data_ptr(), mean.data_ptr<float>(),
                              invstd.data_ptr<float>(), (const scalar_t*)g.data_ptr(),
                              (const scalar_t*)bta.data_ptr(), scratch.data_ptr<float>(),
-                             total, C);
+                             total, C, tail);
         else
           hipLaunchKernelGGL((bn_bwd_reduce_anyc_kernel<data_t, scalar_t, kAct>),
                              dim3(nb), dim3(256), 0,
@@ -695,15 +735,16 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                              (const data_t*)dyc.data_ptr(), mean.data_ptr<float>(),
                              invstd.data_ptr<float>(), (const scalar_t*)g.data_ptr(),
                              (const scalar_t*)bta.data_ptr(), scratch.data_ptr<float>(),
-                             total, C);
+                             total, C, tail);
       });
     });
   });
-  DISPATCH_FB(g.scalar_type(), "bn_bwd_fin", [&] {
-    hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
-                       stream, scratch.data_ptr<float>(), nb, sums.data_ptr<float>(),
-                       (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
-  });
+  if (!lastblock)
+    DISPATCH_FB(g.scalar_type(), "bn_bwd_fin", [&] {
+      hipLaunchKernelGGL((bn_bwd_finalize_kernel<scalar_t>), dim3(2 * C), dim3(64), 0,
+                         stream, scratch.data_ptr<float>(), nb, sums.data_ptr<float>(),
+                         (scalar_t*)dbeta.data_ptr(), (scalar_t*)dgamma.data_ptr(), C);
+    });
   int grid1 = faa_grid(total / 8 + 1, 256);
   DISPATCH_FB(xc.scalar_type(), "bn_bwd_apply", [&] {
     using data_t = scalar_t;
