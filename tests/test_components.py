@@ -525,3 +525,28 @@ def test_bn_fold_ownership_math():
     for C in [3, 21, 61, 340, 485, 850]:
         for bid in (0, 1, 13):
             assert check_anyc(C, bid), ("anyc", C, bid)
+
+
+@pytest.mark.parametrize("conf,nc,probe", [
+    ({"type": "resnet50"}, 1000,
+     ["conv1.weight", "layer1.0.conv3.weight", "layer1.0.downsample.0.weight",
+      "layer4.2.bn3.running_var", "fc.weight"]),
+    ({"type": "pyramid", "depth": 272, "alpha": 200, "bottleneck": True}, 10,
+     ["conv1.weight", "layer1.0.bn1.weight", "layer1.0.conv2.weight",
+      "layer3.29.bn4.running_mean", "bn_final.weight", "fc.bias"]),
+    ({"type": "shakeshake26_2x96d"}, 10,
+     ["c_in.weight", "layer1.0.branch1.1.weight", "layer1.0.branch2.5.running_mean",
+      "layer2.0.shortcut.conv1.weight", "fc_out.weight"]),
+    ({"type": "efficientnet-b0"}, 1000,
+     ["_conv_stem.weight", "_bn0.weight", "_blocks.0._depthwise_conv.weight",
+      "_blocks.1._expand_conv.weight", "_blocks.15._se_reduce.bias",
+      "_conv_head.weight", "_fc.weight"]),
+])
+def test_state_dict_key_parity_all_families(conf, nc, probe):
+    """Module-path parity with the reference model trees (resnet.py,
+    pyramidnet.py, shake_resnet.py, efficientnet model.py) so released
+    .pth checkpoints interchange across the whole zoo."""
+    from fast_autoaugment_amd.models import build_model
+    keys = set(build_model(conf, nc).state_dict().keys())
+    for p in probe:
+        assert p in keys, p
