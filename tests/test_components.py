@@ -550,3 +550,78 @@ def test_state_dict_key_parity_all_families(conf, nc, probe):
     keys = set(build_model(conf, nc).state_dict().keys())
     for p in probe:
         assert p in keys, p
+
+
+def test_dw_tpl_branch_logic_vs_reference():
+    """CPU mirror of dw_fwd_tpl_kernel's interior/guarded/pairing branches
+    (depthwise.hip) fuzzed over odd widths and TF-SAME pads that the GPU
+    parametrized shapes don't cover."""
+    import math
+    rng = np.random.default_rng(0)
+
+    def sim(X, Wt, stride, pt, pb, pl, pr, K, PW):
+        B, H, Wd, C = X.shape
+        Ho = (H + pt + pb - K) // stride + 1
+        Wo = (Wd + pl + pr - K) // stride + 1
+        Y = np.full((B, Ho, Wo, C), np.nan, np.float32)
+        for b in range(B):
+            for ho in range(Ho):
+                for wp in range((Wo + PW - 1) // PW):
+                    wo = wp * PW
+                    hi0, wi0 = ho * stride - pt, wo * stride - pl
+                    pair = (PW == 2) and (wo + 1 < Wo)
+                    interior = (hi0 >= 0 and hi0 + K <= H and wi0 >= 0
+                                and wi0 + K + (PW - 1) * stride <= Wd
+                                and (PW == 1 or pair))
+                    acc0 = np.zeros(C, np.float32)
+                    acc1 = np.zeros(C, np.float32)
+                    if interior:
+                        for kh in range(K):
+                            for col in range(K + (PW - 1) * stride):
+                                xv = X[b, hi0 + kh, wi0 + col]
+                                if col < K:
+                                    acc0 += xv * Wt[:, kh, col]
+                                if PW == 2 and col >= stride and col - stride < K:
+                                    acc1 += xv * Wt[:, kh, col - stride]
+                    else:
+                        for kh in range(K):
+                            hi = hi0 + kh
+                            if hi < 0 or hi >= H:
+                                continue
+                            for kw in range(K):
+                                wi = wi0 + kw
+                                if 0 <= wi < Wd:
+                                    acc0 += X[b, hi, wi] * Wt[:, kh, kw]
+                                if PW == 2 and pair and 0 <= wi + stride < Wd:
+                                    acc1 += X[b, hi, wi + stride] * Wt[:, kh, kw]
+                    Y[b, ho, wo] = acc0
+                    if PW == 2 and pair:
+                        Y[b, ho, wo + 1] = acc1
+        return Y
+
+    def ref(X, Wt, stride, pt, pb, pl, pr, K):
+        B, H, Wd, C = X.shape
+        Xp = np.pad(X, ((0, 0), (pt, pb), (pl, pr), (0, 0)))
+        Ho = (H + pt + pb - K) // stride + 1
+        Wo = (Wd + pl + pr - K) // stride + 1
+        Y = np.zeros((B, Ho, Wo, C), np.float32)
+        for ho in range(Ho):
+            for wo in range(Wo):
+                win = Xp[:, ho * stride:ho * stride + K, wo * stride:wo * stride + K]
+                Y[:, ho, wo] = np.einsum("bhwc,chw->bc", win, Wt)
+        return Y
+
+    for _ in range(12):
+        K = int(rng.choice([3, 5]))
+        s = int(rng.choice([1, 2]))
+        H = int(rng.integers(K, 11))
+        Wd = int(rng.integers(K, 11))
+        ph = max((math.ceil(H / s) - 1) * s + K - H, 0)
+        pw = max((math.ceil(Wd / s) - 1) * s + K - Wd, 0)
+        X = rng.standard_normal((2, H, Wd, 8)).astype(np.float32)
+        Wt = rng.standard_normal((8, K, K)).astype(np.float32)
+        got = sim(X, Wt, s, ph // 2, ph - ph // 2, pw // 2, pw - pw // 2, K,
+                  2 if s == 1 else 1)
+        want = ref(X, Wt, s, ph // 2, ph - ph // 2, pw // 2, pw - pw // 2, K)
+        assert not np.isnan(got).any()
+        assert np.abs(got - want).max() < 1e-4, (K, s, H, Wd)
