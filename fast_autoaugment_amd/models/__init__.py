@@ -38,7 +38,8 @@ def build_model(conf, num_class: int = 10) -> nn.Module:
         return PyramidNet("cifar10", depth=conf["depth"], alpha=conf["alpha"],
                           num_classes=num_class, bottleneck=conf["bottleneck"])
     if "efficientnet" in name:
-        model = EfficientNet.from_name(name, condconv_num_expert=conf.get("condconv_num_expert", 1))
+        model = EfficientNet.from_name(name, {"num_classes": num_class},
+                                       condconv_num_expert=conf.get("condconv_num_expert", 1))
         _tf_style_init(model)
         return model
     raise NameError(f"no model named {name}")

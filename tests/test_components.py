@@ -625,3 +625,17 @@ def test_dw_tpl_branch_logic_vs_reference():
         want = ref(X, Wt, s, ph // 2, ph - ph // 2, pw // 2, pw - pw // 2, K)
         assert not np.isnan(got).any()
         assert np.abs(got - want).max() < 1e-4, (K, s, H, Wd)
+
+
+def test_num_classes_plumbs_to_classifier():
+    """Every family's classifier must honor build_model's num_class
+    (EfficientNet silently kept 1000 until round 1's audit)."""
+    import torch as _t
+    from fast_autoaugment_amd.models import build_model
+    for conf in [{"type": "wresnet40_2"}, {"type": "resnet50"},
+                 {"type": "shakeshake26_2x96d"},
+                 {"type": "pyramid", "depth": 32, "alpha": 20, "bottleneck": True},
+                 {"type": "efficientnet-b0"}]:
+        m = build_model(conf, 7)
+        head = [mod for mod in m.modules() if isinstance(mod, _t.nn.Linear)][-1]
+        assert head.out_features == 7, (conf["type"], head.out_features)
