@@ -639,3 +639,34 @@ def test_num_classes_plumbs_to_classifier():
         m = build_model(conf, 7)
         head = [mod for mod in m.modules() if isinstance(mod, _t.nn.Linear)][-1]
         assert head.out_features == 7, (conf["type"], head.out_features)
+
+
+def test_trainer_epoch_resume(tmp_path):
+    """Epoch-based checkpoint resume (reference train.py:191-218): extending
+    conf.epoch continues from the saved epoch instead of restarting."""
+    os.environ["FAA_SYNTH_TRAIN"] = "32"
+    os.environ["FAA_SYNTH_TEST"] = "16"
+    try:
+        from fast_autoaugment_amd.data import api as data_api
+        from fast_autoaugment_amd.engine import train_and_eval
+        data_api._STORE_CACHE.clear()
+        base = {
+            "model": {"type": "wresnet40_2"}, "dataset": "cifar10",
+            "aug": "default", "cutout": 0, "batch": 16, "epoch": 1,
+            "lr": 0.01,
+            "lr_schedule": {"type": "cosine", "warmup": {"multiplier": 1, "epoch": 0}},
+            "optimizer": {"type": "sgd", "decay": 1e-4, "nesterov": True, "ema": 0},
+        }
+        C.replace(dict(base))
+        p = str(tmp_path / "resume.pth")
+        train_and_eval("", "./data", save_path=p, evaluation_interval=1)
+        assert torch.load(p, weights_only=False)["epoch"] == 1
+        base["epoch"] = 2
+        C.replace(dict(base))
+        train_and_eval("", "./data", save_path=p, evaluation_interval=2)
+        assert torch.load(p, weights_only=False)["epoch"] == 2
+    finally:
+        os.environ.pop("FAA_SYNTH_TRAIN", None)
+        os.environ.pop("FAA_SYNTH_TEST", None)
+        from fast_autoaugment_amd.data import api as data_api
+        data_api._STORE_CACHE.clear()
