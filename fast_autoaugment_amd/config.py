@@ -87,11 +87,31 @@ class ConfigArgumentParser(argparse.ArgumentParser):
         self._override_keys.append((dest, key))
 
     def parse_args(self, args=None, namespace=None):
-        ns, _ = super().parse_known_args(args, namespace)
+        ns, extra = super().parse_known_args(args, namespace)
         if ns.config:
             Config.load(ns.config)
-        ns = super().parse_args(args, namespace)
         conf = Config.get()
+        # theconf-style generic overrides: any leftover `--key value` (dotted
+        # keys allowed) that names an EXISTING conf entry overrides it, with
+        # YAML type coercion; unknown keys still error via the strict parse.
+        pairs = self._apply_generic_overrides(conf, extra)
+        if args is None:
+            import sys
+            args = list(sys.argv[1:])
+        remaining = []
+        pending = list(pairs)
+        i = 0
+        while i < len(args):
+            hit = next((p for p in pending
+                        if args[i] == p[0] and i + 1 < len(args) and args[i + 1] == p[1]),
+                       None)
+            if hit is not None:
+                pending.remove(hit)
+                i += 2
+                continue
+            remaining.append(args[i])
+            i += 1
+        ns = super().parse_args(remaining, namespace)
         for dest, key in self._override_keys:
             val = getattr(ns, dest, None)
             if val is not None:
@@ -101,3 +121,28 @@ class ConfigArgumentParser(argparse.ArgumentParser):
                     node = node.setdefault(p, {})
                 node[parts[-1]] = val
         return ns
+
+    @staticmethod
+    def _apply_generic_overrides(conf, extra):
+        pairs = []
+        i = 0
+        while i < len(extra):
+            tok = extra[i]
+            if tok.startswith("--") and i + 1 < len(extra):
+                key = tok[2:].replace("-", "_")
+                parts = key.split(".")
+                node = conf.conf
+                ok = True
+                for p in parts[:-1]:
+                    if isinstance(node, dict) and p in node:
+                        node = node[p]
+                    else:
+                        ok = False
+                        break
+                if ok and isinstance(node, dict) and parts[-1] in node:
+                    node[parts[-1]] = yaml.safe_load(extra[i + 1])
+                    pairs.append((tok, extra[i + 1]))
+                    i += 2
+                    continue
+            i += 1
+        return pairs

@@ -670,3 +670,25 @@ def test_trainer_epoch_resume(tmp_path):
         os.environ.pop("FAA_SYNTH_TEST", None)
         from fast_autoaugment_amd.data import api as data_api
         data_api._STORE_CACHE.clear()
+
+
+def test_config_generic_cli_overrides(tmp_path):
+    """theconf-style generic `--key value` CLI overrides of existing YAML
+    keys (dotted keys included), with YAML type coercion."""
+    import yaml as _yaml
+    from fast_autoaugment_amd.config import Config, ConfigArgumentParser
+    conf_file = tmp_path / "c.yaml"
+    conf_file.write_text(_yaml.safe_dump({
+        "epoch": 200, "batch": 128, "lr": 0.1,
+        "optimizer": {"type": "sgd", "decay": 0.0005},
+    }))
+    Config.clear()
+    p = ConfigArgumentParser()
+    p.add_argument("--tag", type=str, default="")
+    ns = p.parse_args(["-c", str(conf_file), "--epoch", "3", "--tag", "x",
+                       "--optimizer.decay", "0.01"])
+    conf = Config.get()
+    assert conf["epoch"] == 3 and isinstance(conf["epoch"], int)
+    assert conf["optimizer"]["decay"] == 0.01
+    assert conf["batch"] == 128 and ns.tag == "x"
+    Config.clear()
