@@ -97,6 +97,11 @@ def main():
     imagenet = "imagenet" in args.dataset
     img_src = 224 if imagenet else 32        # synthetic source resolution
     out_size = 224 if imagenet else 32
+    if imagenet and args.model.startswith("efficientnet"):
+        # compound-scaled input resolutions (B0 224 ... B4 380)
+        from fast_autoaugment_amd.models.efficientnet import _SCALING
+        out_size = _SCALING[args.model][2]
+        img_src = max(out_size, 224)
     # synthetic data, random-init weights (no network on the box)
     n_synth = int(os.environ.get("FAA_BENCH_IMGS", "10000" if imagenet else "50000"))
     imgs, labels = synthetic_arrays(n_synth, img_src, nc, seed=1234 + rank)
