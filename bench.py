@@ -211,7 +211,11 @@ def main():
     def make_batch_cpu():
         from fast_autoaugment_amd.aug import cpu_exec
         sel, prog, post2 = host_q.get()
-        out = cpu_exec.run_pipeline_cpu(store.images_np[sel], prog, post2, mean, std)
+        if imagenet:
+            out = cpu_exec.run_pipeline_imagenet_cpu(store.images_np[sel], prog, post2,
+                                                     mean, std, out_size, out_size)
+        else:
+            out = cpu_exec.run_pipeline_cpu(store.images_np[sel], prog, post2, mean, std)
         data = torch.from_numpy(out).permute(0, 3, 1, 2).contiguous()
         return data, store.labels[torch.from_numpy(np.ascontiguousarray(sel))]
 
