@@ -221,3 +221,19 @@ def test_train_dist_launcher_node_rank_env(tmp_path, monkeypatch):
     except SystemExit as e:
         assert e.code == 0
     assert captured == {"node_rank": 2, "nnodes": 4, "np": 3}
+
+
+def test_bench_cpu_dry_run_imagenet(tmp_path):
+    """bench.py CPU dry-run must exercise the ImageNet pipeline branch
+    (run_pipeline_imagenet_cpu + 18-wide post), not just CIFAR."""
+    import subprocess
+    import sys as _sys
+    import os as _os
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    env = dict(_os.environ, FAA_BENCH_CPU="1", FAA_SYNTH_TRAIN="16", FAA_SYNTH_TEST="8")
+    r = subprocess.run(
+        [_sys.executable, "bench.py", "--model", "resnet50", "--dataset", "imagenet",
+         "--batch", "2", "--steps", "1", "--warmup", "0"],
+        cwd=repo, env=env, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert '"image": "224x224"' in r.stdout
