@@ -386,8 +386,20 @@ def main():
             sel = torch.arange(min(args.batch, len(store)), device=dev)
             zero_prog = torch.zeros_like(prog_s)
             zero_post = torch.zeros_like(post_s)
-            data = (CX.aug_pipeline(store.images, sel, zero_prog, zero_post,
-                                    mean_t, std_t, bf16) if not cpu_mode else None)
+            n_dbg = int(sel.numel())
+            if cpu_mode:
+                data = None
+            elif imagenet:
+                from fast_autoaugment_amd.aug.imagenet import compile_post_imagenet
+                post_np = compile_post_imagenet(n_dbg, img_src, img_src,
+                                                np.random.default_rng(0), out_size,
+                                                train=False)   # center crop
+                data = CX.aug_pipeline_imagenet(store.images, sel, zero_prog[:n_dbg],
+                                                torch.from_numpy(post_np).to(dev),
+                                                mean_t, std_t, out_size, out_size, bf16)
+            else:
+                data = CX.aug_pipeline(store.images, sel, zero_prog[:n_dbg],
+                                       zero_post[:n_dbg], mean_t, std_t, bf16)
             if data is not None:
                 model.eval()
                 preds = model(data)
