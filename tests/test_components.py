@@ -692,3 +692,28 @@ def test_config_generic_cli_overrides(tmp_path):
     assert conf["optimizer"]["decay"] == 0.01
     assert conf["batch"] == 128 and ns.tag == "x"
     Config.clear()
+
+
+def test_loader_seed_determinism():
+    """Same (seed, epoch) -> identical batches; different epoch -> different
+    shuffle (AugLoader mirrors DistributedSampler's set_epoch contract)."""
+    from fast_autoaugment_amd.data.loader import AugLoader, TensorStore
+    rng = np.random.default_rng(0)
+    imgs = rng.integers(0, 255, (64, 32, 32, 3), dtype=np.uint8)
+    st = TensorStore(imgs, np.arange(64, dtype=np.int64))
+    mean = np.zeros(3, np.float32)
+    std = np.ones(3, np.float32)
+
+    def first_batch(seed, epoch):
+        ld = AugLoader(st, 16, [], train=True, mean=mean, std=std, seed=seed,
+                       prefetch=0)
+        ld.set_epoch(epoch)
+        data, label = next(iter(ld))
+        return data.numpy(), label.numpy()
+
+    d1, l1 = first_batch(3, 1)
+    d2, l2 = first_batch(3, 1)
+    np.testing.assert_array_equal(d1, d2)
+    np.testing.assert_array_equal(l1, l2)
+    _, l3 = first_batch(3, 2)
+    assert not np.array_equal(l1, l3)
