@@ -116,7 +116,6 @@ __global__ void bn_reduce_vec_kernel(const T* __restrict__ x, float* __restrict_
   // host guarantees (gridDim*blockDim*8) % C == 0, so each thread's channel
   // octet c0 is FIXED across its grid-stride loop.
   float s[8] = {0}, ss[8] = {0};
-  const int c0 = (int)((((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8) % C);
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   if constexpr (sizeof(T) == 2) {

@@ -32,7 +32,7 @@ __device__ __forceinline__ float faa_block_reduce_sum(float v, float* lds) {
   if (lane == 0) lds[wid] = v;
   __syncthreads();
   int nw = blockDim.x / FAA_WAVE;
-  v = (threadIdx.x < nw) ? lds[threadIdx.x] : 0.0f;
+  v = ((int)threadIdx.x < nw) ? lds[threadIdx.x] : 0.0f;
   if (wid == 0) v = faa_warp_reduce_sum(v);
   return v;  // valid in thread 0
 }
