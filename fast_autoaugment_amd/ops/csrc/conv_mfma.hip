@@ -29,7 +29,7 @@ namespace {
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-constexpr int BM = 64, BN = 64, BK = 32;
+constexpr int BK = 32;   // K-slab depth (BM/BN are per-launch templates)
 constexpr int LDSP = 40;             // padded LDS row stride (elems)
 
 __device__ __forceinline__ float b2f(short u) {

@@ -395,7 +395,6 @@ torch::Tensor dwconv_bwd_weight(torch::Tensor dy, torch::Tensor x, int64_t strid
   auto f32 = xc.options().dtype(torch::kFloat32);
   auto acc = torch::zeros({cells, C}, f32);
   auto stream = at::hip::getCurrentHIPStream().stream();
-  int cmain = std::min(C, 256);
   {
     // stride-aligned x-grid so each thread's channel octet is fixed
     auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
