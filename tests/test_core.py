@@ -1,8 +1,6 @@
 """CPU unit tests: policies codec, schedules, optimizers, EMA, metrics, flat params."""
 import math
 
-import numpy as np
-import pytest
 import torch
 
 from fast_autoaugment_amd import policies

@@ -7,7 +7,6 @@ whole trace) and prints time share, counts and mean duration.
 """
 import argparse
 import sqlite3
-import sys
 
 
 def main():
