@@ -79,7 +79,12 @@ class EMA:
         sd = module.state_dict()
         new_keys = [k for k in sd if k not in self.shadow]
         for k in new_keys:
-            self.shadow[k] = sd[k].detach().clone()
+            # Shadow is kept in fp32 regardless of the param dtype: under the
+            # pure-bf16 fast path, a bf16 shadow at mu=0.9999 has a per-step
+            # increment (1-mu)*delta below bf16 ULP and silently freezes
+            # (reference params are fp32 so common.py:44-51 never hit this).
+            t = sd[k].detach()
+            self.shadow[k] = t.float().clone() if t.dtype.is_floating_point else t.clone()
         live = [k for k in sd if k not in new_keys]
         if not live:
             return

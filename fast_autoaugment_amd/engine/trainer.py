@@ -300,10 +300,16 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
         if math.isnan(rs["train"]["loss"]):
             raise Exception("train loss is NaN.")
 
-        if (ema is not None and conf["optimizer"].get("ema_interval", -1) > 0
+        # EMA->model periodic sync (reference train.py:262-270). ema exists
+        # only on the master rank, so EVERY rank must enter the broadcast
+        # collective or the job deadlocks — gate on the config alone and let
+        # non-masters participate receive-only.
+        if (conf["optimizer"].get("ema", 0) > 0
+                and conf["optimizer"].get("ema_interval", -1) > 0
                 and epoch % conf["optimizer"]["ema_interval"] == 0):
             raw = model.module if hasattr(model, "module") else model
-            raw.load_state_dict(ema.state_dict())
+            if ema is not None:
+                raw.load_state_dict(ema.state_dict())
             if local_rank >= 0:
                 import torch.distributed as dist
                 for _, x in raw.state_dict().items():

@@ -9,6 +9,7 @@ optimizer's param_groups.
 from __future__ import annotations
 
 import math
+from typing import Sequence
 
 
 class _Schedule:

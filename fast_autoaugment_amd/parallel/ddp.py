@@ -110,6 +110,13 @@ class FlatDDP(nn.Module):
         the optimizer wrapper or trainer hook."""
         if self.world_size <= 1 or not self._sync_enabled:
             return
+        # Unused-parameter safety: a param whose branch never ran leaves its
+        # bucket counter non-zero and the bucket un-reduced — ranks would
+        # silently diverge. Reduce any leftover buckets here (their grad
+        # segment is whatever accumulated, zeros for fully-unused params).
+        for bi, pending in enumerate(self._pending):
+            if pending > 0:
+                self._launch_bucket(bi)
         for w in self._works:
             w.wait()
         self._works.clear()

@@ -26,10 +26,17 @@ def main():
     parser.add_argument("--decay", type=float, default=-1)
     parser.add_argument("--workers", type=int, default=None,
                         help="scheduler workers (default: one per GPU)")
-    parser.add_argument("--per-class", action="store_true")
-    parser.add_argument("--resume", action="store_true")
+    parser.add_argument("--per-class", action="store_true",
+                        help="accepted for reference-CLI parity; not implemented "
+                             "(the reference parses and ignores it too, search.py:151)")
+    parser.add_argument("--resume", action="store_true",
+                        help="reuse existing fold checkpoints and trial journals")
     parser.add_argument("--smoke-test", action="store_true")
     args = parser.parse_args()
+
+    if args.per_class:
+        logger.warning("--per-class is parsed for CLI parity but has no effect "
+                       "(unimplemented in the reference as well)")
 
     if args.decay > 0:
         C.get()["optimizer"]["decay"] = args.decay
@@ -42,7 +49,7 @@ def main():
     results = run_search(args.dataroot, until=args.until, num_op=args.num_op,
                          num_policy=args.num_policy, num_search=args.num_search,
                          cv_ratio=args.cv_ratio, smoke_test=args.smoke_test,
-                         n_workers=args.workers, resume=True)
+                         n_workers=args.workers, resume=args.resume)
     if "final_policy_set" in results:
         logger.info("final policies:\n%s", json.dumps(results["final_policy_set"]))
     logger.info("done. search_gpu_hours=%.4f", results.get("search_gpu_hours", 0.0))

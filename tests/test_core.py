@@ -88,6 +88,34 @@ def test_ema_warmup_and_lerp():
     assert torch.allclose(ema.shadow["weight"], expect, atol=1e-6)
 
 
+def test_ema_fp32_shadow_moves_on_bf16_weights():
+    """At mu=0.9999 the per-step increment (1-mu)*delta is below bf16 ULP; the
+    shadow must be fp32 so the EMA keeps integrating (ADVICE r1 high,
+    reference common.py:44-51 never hit this because its params are fp32)."""
+    m = torch.nn.Linear(16, 16).bfloat16()
+    ema = EMA(0.9999)
+    ema(m)  # no step arg -> mu stays 0.9999 (initial copy)
+    assert ema.shadow["weight"].dtype == torch.float32
+    w0 = ema.shadow["weight"].clone()
+    # 200 tiny updates: each (1-mu)*delta ~ 1e-6, invisible at bf16 ULP (~0.008
+    # at magnitude 1) but must accumulate in fp32
+    with torch.no_grad():
+        m.weight.fill_(1.0)
+    for _ in range(200):
+        ema(m)
+    moved = (ema.shadow["weight"] - w0).abs().max().item()
+    # after 200 steps shadow should have moved ~ (1 - mu^200)*(1 - w0) ~ 2% of gap
+    assert moved > 1e-3, f"EMA shadow frozen (moved {moved})"
+    # integer buffers still copied verbatim
+    sd = {"num_batches_tracked": torch.tensor(3)}
+    class _M(torch.nn.Module):
+        def state_dict(self, *a, **k):
+            return sd
+    ema2 = EMA(0.9999)
+    ema2(_M())
+    assert ema2.shadow["num_batches_tracked"].dtype == torch.int64
+
+
 def test_label_smooth_ce_matches_manual():
     torch.manual_seed(0)
     logits = torch.randn(8, 10)
