@@ -43,6 +43,20 @@ def _faa_wrw_wins(cin: int) -> bool:
     return cin < 8 or cin == 128
 
 
+def _dbias(dy: torch.Tensor) -> torch.Tensor:
+    """Bias gradient = column sum of dy over (B,H,W). Default: the replay-safe
+    colsum v2 HIP kernel (no atomics/memset -> hipGraph-capture clean);
+    FAA_DBIAS=torch keeps at::reduce, FAA_DBIAS=legacy uses the round-1
+    atomic kernel (bisect only, corrupts ~50% of graph replays)."""
+    mode = _os.environ.get("FAA_DBIAS", "colsum")
+    C = dy.size(1)
+    if mode == "colsum" and C % 8 == 0:
+        return ext().colsum_bf16(dy)
+    if mode == "legacy" and C % 8 == 0:
+        return ext().colsum_bf16_legacy(dy)
+    return dy.sum(dim=(0, 2, 3))
+
+
 class FaaConv2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
@@ -75,19 +89,13 @@ class FaaConv2dFn(torch.autograd.Function):
                 dw, _ = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
                                             weight.size(2), weight.size(3),
                                             False)
-                if ctx.has_bias and ctx.needs_input_grad[2]:
-                    dbias = dy.sum(dim=(0, 2, 3))
-            else:
-                if ctx.needs_input_grad[1]:
-                    dw = torch.nn.grad.conv2d_weight(
-                        x, list(weight.shape), dy, stride=ctx.stride,
-                        padding=ctx.padding).contiguous(
-                            memory_format=torch.channels_last)
-                if ctx.has_bias and ctx.needs_input_grad[2]:
-                    # NOTE: colsum_bf16 is replay-clean standalone but was
-                    # implicated in hipGraph-replay corruption inside the full
-                    # captured step (tools/nan_flake.py bisection) -> torch sum
-                    dbias = dy.sum(dim=(0, 2, 3))
+            elif ctx.needs_input_grad[1]:
+                dw = torch.nn.grad.conv2d_weight(
+                    x, list(weight.shape), dy, stride=ctx.stride,
+                    padding=ctx.padding).contiguous(
+                        memory_format=torch.channels_last)
+            if ctx.has_bias and ctx.needs_input_grad[2]:
+                dbias = _dbias(dy)
         return dx, dw, dbias, None, None
 
 
@@ -158,7 +166,7 @@ class FaaDepthwiseFn(torch.autograd.Function):
             dw = C.dwconv_bwd_weight(dy, x, ctx.stride, pt, pl,
                                      weight.size(2), weight.size(3))
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            dbias = dy.sum(dim=(0, 2, 3))
+            dbias = _dbias(dy)
         return dx, dw, dbias, None, None
 
 

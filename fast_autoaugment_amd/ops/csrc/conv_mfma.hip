@@ -396,6 +396,51 @@ __global__ void cast_f32_bf16_kernel(const float* __restrict__ in, short* __rest
   if (i < n) out[i] = f2b(in[i]);
 }
 
+// colsum v2 — replay-safe column sum (dbias) for hipGraph capture.
+// The legacy colsum_kernel (atomicAdd into a torch::zeros workspace) was
+// implicated in hipGraph-replay corruption (tools/nan_flake.py bisection,
+// round 1). v2 removes BOTH suspects: per-block partials are plain stores
+// into a fully-overwritten torch::empty workspace (no memset node in the
+// graph, no atomics), and a second tiny kernel reduces nb x C -> C and
+// casts to bf16 in one pass (also removing the separate cast launch).
+__global__ void colsum_partial_kernel(const short* __restrict__ dY,
+                                      float* __restrict__ partial,
+                                      int64_t M, int C) {
+  float s[8] = {};
+  int64_t total = M * C;
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t i = i0; i < total; i += stride) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(dY + i);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) s[j] += b2f(v[j]);
+  }
+  __shared__ float lds[256 * 8];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) lds[threadIdx.x * 8 + j] = s[j];
+  __syncthreads();
+  const int groups = C / 8;
+  const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % groups);
+  // each thread folds one or more output channels (supports C > blockDim)
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    int oct = c / 8, ln = c % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups)
+      acc += lds[t * 8 + ln];
+    partial[(int64_t)blockIdx.x * C + c] = acc;
+  }
+}
+
+__global__ void colsum_finish_kernel(const float* __restrict__ partial,
+                                     short* __restrict__ out, int nb, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float acc = 0;
+  for (int b = 0; b < nb; ++b) acc += partial[(int64_t)b * C + c];
+  out[c] = f2b(acc);
+}
+
 }  // namespace
 
 static ConvGeom make_geom(const torch::Tensor& x, const torch::Tensor& w,
@@ -531,7 +576,31 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w,
 }
 
 torch::Tensor colsum_bf16(torch::Tensor dy) {
-  // dbias[n] = sum over rows of a [*, C] bf16 channels-last tensor
+  // dbias[n] = sum over rows of a [*, C] bf16 channels-last tensor.
+  // v2 two-kernel scheme: no atomics, no memset — every byte read was
+  // written earlier in the same stream/replay (hipGraph-safe).
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  int C = dyc.size(1);
+  int64_t M = dyc.numel() / C;
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  TORCH_CHECK(C % 8 == 0, "colsum_bf16: C % 8 == 0 expected");
+  auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
+  int q = C / gcd(C, 2048);
+  int nb = ((64 + q - 1) / q) * q;   // multiple of q => per-thread channel set
+                                     // is invariant across grid-stride iters
+  auto part = torch::empty({nb, C}, dyc.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(colsum_partial_kernel, dim3(nb), dim3(256), 0, stream,
+                     (const short*)dyc.data_ptr(), part.data_ptr<float>(),
+                     M, C);
+  auto out = torch::empty({C}, dyc.options());
+  hipLaunchKernelGGL(colsum_finish_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     stream, part.data_ptr<float>(), (short*)out.data_ptr(), nb, C);
+  return out;
+}
+
+torch::Tensor colsum_bf16_legacy(torch::Tensor dy) {
+  // round-1 atomic+zeros implementation, kept ONLY for the hipGraph
+  // corruption bisect (tools/nan_flake.py); not used on any dispatch path.
   auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
   int C = dyc.size(1);
   int64_t M = dyc.numel() / C;
@@ -585,18 +654,8 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
 
   torch::Tensor dbias;
   if (want_bias) {
-    auto part = torch::zeros({Cout}, f32);
     if (Cout % 8 == 0) {
-      // stride-aligned colsum (channel set fixed per thread modulo C)
-      auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
-      int q = Cout / gcd(Cout, 2048);
-      int nb = ((64 + q - 1) / q) * q;
-      hipLaunchKernelGGL(colsum_kernel, dim3(nb), dim3(256), 0, stream,
-                         (const short*)dyc.data_ptr(), part.data_ptr<float>(),
-                         (int64_t)M, Cout);
-      dbias = torch::empty({Cout}, xc.options());
-      hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3((Cout + 255) / 256), dim3(256), 0,
-                         stream, part.data_ptr<float>(), (short*)dbias.data_ptr(), Cout);
+      dbias = colsum_bf16(dyc);
     } else {
       dbias = dyc.sum(/*dim=*/{0, 2, 3}).to(xc.scalar_type());
     }

@@ -170,3 +170,20 @@ def test_conv_fwd_splitk_matches():
             assert err < 2e-2, f"splitk rel err {err}"
     finally:
         os.environ.pop("FAA_CONV_SPLITK", None)
+
+
+@pytest.mark.parametrize("M,Ch", [(128 * 32 * 32, 16), (128 * 8 * 8, 128),
+                                  (128 * 16 * 16, 160), (64 * 8 * 8, 320),
+                                  (128 * 8 * 8, 640), (1000, 8)])
+def test_colsum_v2_matches_reference(M, Ch):
+    """Replay-safe colsum v2 (dbias) vs fp32 torch sum, incl. C > blockDim
+    shapes for the big-config coverage (C up to 640)."""
+    torch.manual_seed(3)
+    dy = (torch.randn(M, Ch, device=dev()) * 0.5).bfloat16()
+    # view as NHWC channels-last image so the C++ side sees a 4D tensor
+    dy4 = dy.view(M, 1, 1, Ch).permute(0, 3, 1, 2).contiguous(
+        memory_format=torch.channels_last)
+    ref = dy.float().sum(dim=0)
+    got = C.colsum_bf16(dy4).float()
+    err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert err < 1e-2, f"colsum rel err {err}"
