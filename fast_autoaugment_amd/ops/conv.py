@@ -43,17 +43,46 @@ def _faa_wrw_wins(cin: int) -> bool:
     return cin < 8 or cin == 128
 
 
+_dbias_static = {}
+
+
 def _dbias(dy: torch.Tensor) -> torch.Tensor:
-    """Bias gradient = column sum of dy over (B,H,W). Default: the replay-safe
-    colsum v2 HIP kernel (no atomics/memset -> hipGraph-capture clean);
-    FAA_DBIAS=torch keeps at::reduce, FAA_DBIAS=legacy uses the round-1
-    atomic kernel (bisect only, corrupts ~50% of graph replays)."""
+    """Bias gradient = column sum of dy over (B,H,W).
+
+    FAA_DBIAS modes (tools/nan_hunt.py hipGraph bisect):
+      colsum  (default) replay-safe v2 HIP kernel
+      torch   at::reduce (round-1 quarantine default)
+      legacy  round-1 atomic kernel
+      dummy   at::reduce result + a DISCARDED colsum launch (perturbation
+              control: flakes here => colsum is innocent)
+      static  v2 into non-pool buffers cached at first (eager) call, then
+              cloned into the pool
+    """
     mode = _os.environ.get("FAA_DBIAS", "colsum")
     C = dy.size(1)
-    if mode == "colsum" and C % 8 == 0:
+    if C % 8 != 0:
+        return dy.sum(dim=(0, 2, 3))
+    if mode == "colsum":
         return ext().colsum_bf16(dy)
-    if mode == "legacy" and C % 8 == 0:
+    if mode == "legacy":
         return ext().colsum_bf16_legacy(dy)
+    if mode == "dummy":
+        ext().colsum_bf16(dy)
+        return dy.sum(dim=(0, 2, 3))
+    if mode == "static":
+        import math
+        key = (C, dy.device.index)
+        buf = _dbias_static.get(key)
+        if buf is None:
+            if torch.cuda.is_current_stream_capturing():
+                raise RuntimeError("FAA_DBIAS=static needs an eager warmup pass")
+            q = C // math.gcd(C, 2048)
+            nb = -(-64 // q) * q
+            buf = (torch.empty(nb, C, dtype=torch.float32, device=dy.device),
+                   torch.empty(C, dtype=torch.bfloat16, device=dy.device))
+            _dbias_static[key] = buf
+        ext().colsum_bf16_ws(dy, buf[0], buf[1])
+        return buf[1].clone()
     return dy.sum(dim=(0, 2, 3))
 
 

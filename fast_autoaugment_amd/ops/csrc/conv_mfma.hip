@@ -223,6 +223,155 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
   }
 }
 
+// ------------------------------------------------- direct tiled 3x3 fwd
+// conv3x3_tile_kernel: stride-1 pad-1 3x3 conv, NHWC bf16, for the CIFAR
+// geometries (W in {8,16,32}, H % 8 == 0, any Cin >= 16, any Cout).
+//
+// Rationale vs the im2col conv_fwd_kernel above (profiles r01: 24us at
+// 2.4% MfmaUtil): implicit GEMM re-reads every input pixel 9x through
+// global/L2 and spends the latency-bound inner loop on per-load integer
+// division. Here each block stages its (TH+2)x(TW+2) input tile (+halo)
+// in LDS ONCE per cin-slab and the 9 taps become compile-time LDS
+// offsets; weights live in registers (one bf16x8 per (tap, nfrag) lane
+// slot). Per cin-slab a wave issues 9*MFRAG*2 MFMAs between two
+// syncthreads vs 2 in the im2col loop.
+//
+//  * m-tile = IB*TH*TW pixels (= 64*MFRAG, 4 waves stacked on m)
+//  * n-tile = 32 couts (NFRAG=2), grid_n = ceil(Cout/32)
+//  * K loop = cin slabs of CS=32 (zero-padded tail), 9 taps each
+//  * LDS pixel stride PS=40 shorts: 16B-aligned ds_read_b128 and 2-way-
+//    max bank aliasing (fr*20 mod 32 covers 8 banks x 2 lanes = free)
+template <int TH, int TW, int IB, int MFRAG, bool HAS_BIAS>
+__global__ __launch_bounds__(256)
+void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
+                         const short* __restrict__ bias, short* __restrict__ Y,
+                         ConvGeom g, int tiles_h, int grid_n) {
+  constexpr int CS = 32;             // cin slab depth (one MFMA k)
+  constexpr int PS = 40;             // per-pixel LDS stride in shorts
+  constexpr int XR = TH + 2;
+  constexpr int XC = TW + 2;
+  static_assert(IB * TH * TW == 64 * MFRAG, "m tile mismatch");
+
+  __shared__ short ldsX[IB * XR * XC * PS];
+
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {  // XCD-aware remap: consecutive n-tiles land on one XCD's L2
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, idx = wg / 8;
+    if (q > 0) wg = (xcd < r) ? (xcd * (q + 1) + idx) : (r * (q + 1) + (xcd - r) * q + idx);
+  }
+  const int nt = wg % grid_n;
+  int t = wg / grid_n;
+  const int ty = t % tiles_h;
+  const int img0 = (t / tiles_h) * IB;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int fr = lane & 15;
+  const int kq = (lane >> 4) * 8;
+  const int n0 = nt * 32;
+
+  int aoff[MFRAG];
+  #pragma unroll
+  for (int mf = 0; mf < MFRAG; ++mf) {
+    int m_loc = wave * (MFRAG * 16) + mf * 16 + fr;
+    int ib = m_loc / (TH * TW);
+    int pix = m_loc % (TH * TW);
+    int py = pix / TW, px = pix % TW;
+    aoff[mf] = ((ib * XR + py) * XC + px) * PS + kq;
+  }
+
+  f32x4 acc[MFRAG][2] = {};
+  const int row0 = ty * TH - 1;      // pad = 1
+
+  for (int cs = 0; cs < g.Cin; cs += CS) {
+    // stage the X tile slab (borders/tails zero-filled)
+    constexpr int NV = IB * XR * XC * 4;     // bf16x8 stores
+    for (int v = tid; v < NV; v += 256) {
+      int kc = v & 3;
+      int cell = v >> 2;
+      int col = cell % XC;
+      int rowt = cell / XC;
+      int row = rowt % XR;
+      int ib = rowt / XR;
+      int hi = row0 + row;
+      int wi = col - 1;
+      int ci0 = cs + kc * 8;
+      bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.Wd && ci0 < g.Cin) {
+        const short* p = X + ((((int64_t)(img0 + ib)) * g.H + hi) * g.Wd + wi) * g.Cin + ci0;
+        if (ci0 + 8 <= g.Cin) {
+          val = *reinterpret_cast<const bf16x8*>(p);
+        } else {
+          for (int j = 0; j < g.Cin - ci0; ++j) val[j] = p[j];
+        }
+      }
+      *reinterpret_cast<bf16x8*>(&ldsX[cell * PS + kc * 8]) = val;
+    }
+    // weights slab -> registers (per-lane B fragments, no LDS round-trip)
+    bf16x8 wreg[9][2];
+    #pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      #pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int n = n0 + nf * 16 + fr;
+        int ci0 = cs + kq;
+        bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (n < g.Cout && ci0 < g.Cin) {
+          const short* p = Wt + ((int64_t)n * 9 + tap) * g.Cin + ci0;
+          if (ci0 + 8 <= g.Cin) {
+            v = *reinterpret_cast<const bf16x8*>(p);
+          } else {
+            for (int j = 0; j < g.Cin - ci0; ++j) v[j] = p[j];
+          }
+        }
+        wreg[tap][nf] = v;
+      }
+    }
+    __syncthreads();
+    #pragma unroll
+    for (int kh = 0; kh < 3; ++kh) {
+      #pragma unroll
+      for (int kw = 0; kw < 3; ++kw) {
+        const int toff = (kh * XC + kw) * PS;   // compile-time per (kh,kw)
+        #pragma unroll
+        for (int mf = 0; mf < MFRAG; ++mf) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(&ldsX[aoff[mf] + toff]);
+          acc[mf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, wreg[kh * 3 + kw][0],
+                                                               acc[mf][0], 0, 0, 0);
+          acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, wreg[kh * 3 + kw][1],
+                                                               acc[mf][1], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  #pragma unroll
+  for (int mf = 0; mf < MFRAG; ++mf) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m_loc = wave * (MFRAG * 16) + mf * 16 + (lane >> 4) * 4 + r;
+      int ib = m_loc / (TH * TW);
+      int pix = m_loc % (TH * TW);
+      int py = pix / TW, px = pix % TW;
+      int ho = ty * TH + py;
+      int64_t base = ((((int64_t)(img0 + ib)) * g.Ho + ho) * g.Wo + px) * g.Cout;
+      #pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int n = n0 + nf * 16 + fr;
+        if (n < g.Cout) {
+          float v = acc[mf][nf][r];
+          if (HAS_BIAS) v += b2f(bias[n]);
+          Y[base + n] = f2b(v);
+        }
+      }
+    }
+  }
+}
+
 // bias-add + fp32 -> bf16 cast for the split-K path
 template <bool HAS_BIAS>
 __global__ void conv_splitk_cast_kernel(const float* __restrict__ Yacc,
@@ -497,6 +646,40 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     TORCH_CHECK(bc.scalar_type() == torch::kBFloat16);
     bptr = (const short*)bc.data_ptr();
   }
+
+  // direct tiled 3x3 path: CIFAR geometries, any channel count.
+  // FAA_CONV_DIRECT=0 falls back to the im2col kernel below.
+  {
+    const char* de = getenv("FAA_CONV_DIRECT");
+    bool want = !(de && de[0] == '0');
+    bool geom_ok = (g.KH == 3 && g.KW == 3 && g.stride == 1 && g.pad == 1
+                    && g.Cin >= 16
+                    && ((g.Wd == 32 && g.H % 8 == 0)
+                        || (g.Wd == 16 && g.H % 8 == 0)
+                        || (g.Wd == 8 && g.H == 8 && g.B % 2 == 0)));
+    if (want && geom_ok) {
+      int tiles_h = g.H / 8;
+      int gn = (g.Cout + 31) / 32;
+      int ib = (g.Wd == 8) ? 2 : 1;
+      dim3 dgrid((unsigned)((g.B / ib) * tiles_h * gn));
+      #define CT_LAUNCH(TW_, IB_, MF_, HB_)                                        \
+        hipLaunchKernelGGL((conv3x3_tile_kernel<8, TW_, IB_, MF_, HB_>), dgrid,    \
+                           dim3(256), 0, stream, (const short*)xc.data_ptr(),      \
+                           (const short*)wc.data_ptr(), bptr,                      \
+                           (short*)y.data_ptr(), g, tiles_h, gn)
+      if (has_bias) {
+        if (g.Wd == 32) CT_LAUNCH(32, 1, 4, true);
+        else if (g.Wd == 16) CT_LAUNCH(16, 1, 2, true);
+        else CT_LAUNCH(8, 2, 2, true);
+      } else {
+        if (g.Wd == 32) CT_LAUNCH(32, 1, 4, false);
+        else if (g.Wd == 16) CT_LAUNCH(16, 1, 2, false);
+        else CT_LAUNCH(8, 2, 2, false);
+      }
+      #undef CT_LAUNCH
+      return y;
+    }
+  }
   // FAA_CONV_SPLITK=1: partition K across blockIdx.y when the tile grid
   // underfills the chip (deep 8x8 stages run ~256 WGs = 1/CU; PMC showed
   // 9% occupancy there). fp32-atomic workspace + bias/cast epilogue.
@@ -596,6 +779,26 @@ torch::Tensor colsum_bf16(torch::Tensor dy) {
   hipLaunchKernelGGL(colsum_finish_kernel, dim3((C + 255) / 256), dim3(256), 0,
                      stream, part.data_ptr<float>(), (short*)out.data_ptr(), nb, C);
   return out;
+}
+
+void colsum_bf16_ws(torch::Tensor dy, torch::Tensor part, torch::Tensor out) {
+  // colsum v2 into caller-provided buffers (hipGraph bisect variant E:
+  // workspace + out live OUTSIDE the capture pool).
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  int C = dyc.size(1);
+  int64_t M = dyc.numel() / C;
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  TORCH_CHECK(C % 8 == 0, "colsum_bf16_ws: C % 8 == 0 expected");
+  auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
+  int q = C / gcd(C, 2048);
+  int nb = ((64 + q - 1) / q) * q;
+  TORCH_CHECK(part.numel() >= (int64_t)nb * C && part.scalar_type() == torch::kFloat32,
+              "colsum_bf16_ws: workspace too small");
+  TORCH_CHECK(out.numel() == C && out.scalar_type() == torch::kBFloat16);
+  hipLaunchKernelGGL(colsum_partial_kernel, dim3(nb), dim3(256), 0, stream,
+                     (const short*)dyc.data_ptr(), part.data_ptr<float>(), M, C);
+  hipLaunchKernelGGL(colsum_finish_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     stream, part.data_ptr<float>(), (short*)out.data_ptr(), nb, C);
 }
 
 torch::Tensor colsum_bf16_legacy(torch::Tensor dy) {

@@ -31,6 +31,10 @@ SHAPES = [
     (8, 64, 16, 128, 1, 2),      # 1x1 stride 2
     (4, 160, 16, 160, 3, 1),     # WRN-28-10 non-pow2 channels
     (3, 48, 9, 40, 3, 1),        # odd sizes (M not multiple of 64)
+    # big-config shapes (direct tiled kernel, cin-slab loop, C >= 256)
+    (4, 256, 16, 256, 3, 1),
+    (4, 320, 8, 320, 3, 1),
+    (2, 640, 8, 640, 3, 1),
 ]
 
 

@@ -38,9 +38,9 @@ def main():
               [(128, 16, 32, 32, 3, 1), (128, 32, 32, 32, 3, 1),
                (128, 64, 16, 64, 3, 1), (128, 128, 8, 128, 3, 1),
                (128, 64, 16, 128, 3, 2)])
-    print(f"{'shape':<28} {'miopen':>8} {'auto':>8} "
+    print(f"{'shape':<28} {'miopen':>8} {'direct':>8} {'im2col':>8} "
           + " ".join(f"{t:>8}" for t in ["32x32", "32x64", "64x32", "64x64"])
-          + f" {'auto+sk':>8}")
+          + f" {'i2c+sk':>8}")
     for B, Cin, H, Cout, k, s in shapes:
         x = (torch.randn(B, Cin, H, H, device="cuda") * 0.5).bfloat16() \
             .contiguous(memory_format=torch.channels_last)
@@ -50,6 +50,9 @@ def main():
         row.append(f"{bench(lambda: torch.nn.functional.conv2d(x, w, stride=s, padding=k//2), args.iters):8.1f}")
         os.environ.pop("FAA_CONV_TILE", None)
         os.environ.pop("FAA_CONV_SPLITK", None)
+        os.environ.pop("FAA_CONV_DIRECT", None)     # default: direct when eligible
+        row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
+        os.environ["FAA_CONV_DIRECT"] = "0"
         row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
         for tile in ["32x32", "32x64", "64x32", "64x64"]:
             os.environ["FAA_CONV_TILE"] = tile
@@ -58,6 +61,7 @@ def main():
         os.environ["FAA_CONV_SPLITK"] = "1"
         row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
         os.environ.pop("FAA_CONV_SPLITK", None)
+        os.environ.pop("FAA_CONV_DIRECT", None)
         print(" ".join(row))
 
 
