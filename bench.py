@@ -192,14 +192,33 @@ def main():
     post_w = 18 if imagenet else 6
     post_s = torch.zeros((args.batch, post_w), dtype=torch.float32, device=dev)
     pin = not cpu_mode
-    sel_h = torch.zeros_like(sel_s, device="cpu", pin_memory=pin)
-    prog_h = torch.zeros_like(prog_s, device="cpu", pin_memory=pin)
-    post_h = torch.zeros_like(post_s, device="cpu", pin_memory=pin)
+    # Ring of pinned staging buffer sets. A single reused pinned buffer with
+    # non-blocking H2D is a host-vs-DMA race: when the CPU enqueues ahead of
+    # the GPU, the host-side copy_ overwrites the buffer while the previous
+    # step's DMA is still in flight, and a TORN copy mixes bytes of two
+    # floats into arbitrary bit patterns (incl. NaN) that flow into the aug
+    # program params. (Root cause of the round-1 "colsum hipGraph
+    # corruption": the colsum dbias path merely enqueued faster than
+    # at::sum, opening the window — tools/nan_hunt.py, call3 A/B/C.)
+    N_STAGE = 4
+    stage = [(torch.zeros_like(sel_s, device="cpu", pin_memory=pin),
+              torch.zeros_like(prog_s, device="cpu", pin_memory=pin),
+              torch.zeros_like(post_s, device="cpu", pin_memory=pin),
+              torch.cuda.Event() if not cpu_mode else None)
+             for _ in range(N_STAGE)]
+    stage_i = [0]
 
     sync_upload = os.environ.get("FAA_BENCH_SYNC_UPLOAD", "0") == "1"
+    # FAA_BENCH_UNSAFE_UPLOAD=1 reinstates the round-1 racy single-buffer
+    # behavior (no DMA-done wait) for the regression repro in nan_flake.
+    unsafe_upload = os.environ.get("FAA_BENCH_UNSAFE_UPLOAD", "0") == "1"
 
     def upload_next():
         sel, prog, post = host_q.get()
+        sel_h, prog_h, post_h, ev = stage[0 if unsafe_upload else stage_i[0]]
+        stage_i[0] = (stage_i[0] + 1) % N_STAGE
+        if ev is not None and not unsafe_upload:
+            ev.synchronize()     # previous DMA from THIS buffer set is done
         sel_h.copy_(torch.from_numpy(sel))
         prog_h.copy_(torch.from_numpy(prog))
         post_h.copy_(torch.from_numpy(post))
@@ -207,6 +226,8 @@ def main():
         sel_s.copy_(sel_h, non_blocking=nb)
         prog_s.copy_(prog_h, non_blocking=nb)
         post_s.copy_(post_h, non_blocking=nb)
+        if ev is not None:
+            ev.record()
 
     def make_batch_cpu():
         from fast_autoaugment_amd.aug import cpu_exec

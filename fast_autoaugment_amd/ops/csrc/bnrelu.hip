@@ -67,11 +67,12 @@ struct BnFinTail {          // nullptr counter = disabled
 __device__ __forceinline__ void bn_fin_tail_run(const BnFinTail& t,
                                                 const float* scratch, int C) {
   if (t.counter == nullptr) return;
-  __threadfence();
+  __threadfence();                      // release: partials visible before tick
   __shared__ unsigned order;
   if (threadIdx.x == 0) order = atomicAdd(t.counter, 1u);
   __syncthreads();
   if (order == gridDim.x - 1) {
+    __threadfence();                    // acquire: see every block's partials
     bn_finalize_inblock(scratch, gridDim.x, t.mean, t.invstd, t.running_mean,
                         t.running_var, C, t.count, t.eps, t.momentum);
     __syncthreads();
@@ -94,6 +95,7 @@ __device__ __forceinline__ void bn_bwd_fin_tail_run(const BnBwdFinTail<GT>& t,
   if (threadIdx.x == 0) order = atomicAdd(t.counter, 1u);
   __syncthreads();
   if (order == gridDim.x - 1) {
+    __threadfence();                    // acquire: see every block's partials
     for (int c = threadIdx.x; c < 2 * C; c += (int)blockDim.x) {
       float s = 0;
       for (int b = 0; b < (int)gridDim.x; ++b)

@@ -236,12 +236,15 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
 // slot). Per cin-slab a wave issues 9*MFRAG*2 MFMAs between two
 // syncthreads vs 2 in the im2col loop.
 //
-//  * m-tile = IB*TH*TW pixels (= 64*MFRAG, 4 waves stacked on m)
-//  * n-tile = 32 couts (NFRAG=2), grid_n = ceil(Cout/32)
+//  * m-tile = IB*TH*TW pixels (= WM*MR*16); wave grid WM x WN over (m, n)
+//  * n-tile = WN*NR*16 couts, grid_n = ceil(Cout / n-tile)
+//    small-C config: WN=1 NR=2 (BN=32, max blocks); big-C config: WN=2
+//    NR=2 (BN=64, halves per-block weight traffic, measured win at C>64
+//    when the grid still fills the 256 CUs)
 //  * K loop = cin slabs of CS=32 (zero-padded tail), 9 taps each
 //  * LDS pixel stride PS=40 shorts: 16B-aligned ds_read_b128 and 2-way-
 //    max bank aliasing (fr*20 mod 32 covers 8 banks x 2 lanes = free)
-template <int TH, int TW, int IB, int MFRAG, bool HAS_BIAS>
+template <int TH, int TW, int IB, int WN, int MR, int NR, bool HAS_BIAS>
 __global__ __launch_bounds__(256)
 void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
                          const short* __restrict__ bias, short* __restrict__ Y,
@@ -250,7 +253,8 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
   constexpr int PS = 40;             // per-pixel LDS stride in shorts
   constexpr int XR = TH + 2;
   constexpr int XC = TW + 2;
-  static_assert(IB * TH * TW == 64 * MFRAG, "m tile mismatch");
+  constexpr int WM = 4 / WN;         // wave grid: WM (m) x WN (n)
+  static_assert(IB * TH * TW == WM * MR * 16, "m tile mismatch");
 
   __shared__ short ldsX[IB * XR * XC * PS];
 
@@ -269,21 +273,23 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  const int wm = wave / WN;
+  const int wn = wave % WN;
   const int fr = lane & 15;
   const int kq = (lane >> 4) * 8;
-  const int n0 = nt * 32;
+  const int n0 = nt * (WN * NR * 16) + wn * (NR * 16);
 
-  int aoff[MFRAG];
+  int aoff[MR];
   #pragma unroll
-  for (int mf = 0; mf < MFRAG; ++mf) {
-    int m_loc = wave * (MFRAG * 16) + mf * 16 + fr;
+  for (int mf = 0; mf < MR; ++mf) {
+    int m_loc = wm * (MR * 16) + mf * 16 + fr;
     int ib = m_loc / (TH * TW);
     int pix = m_loc % (TH * TW);
     int py = pix / TW, px = pix % TW;
     aoff[mf] = ((ib * XR + py) * XC + px) * PS + kq;
   }
 
-  f32x4 acc[MFRAG][2] = {};
+  f32x4 acc[MR][NR] = {};
   const int row0 = ty * TH - 1;      // pad = 1
 
   for (int cs = 0; cs < g.Cin; cs += CS) {
@@ -311,11 +317,11 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
       *reinterpret_cast<bf16x8*>(&ldsX[cell * PS + kc * 8]) = val;
     }
     // weights slab -> registers (per-lane B fragments, no LDS round-trip)
-    bf16x8 wreg[9][2];
+    bf16x8 wreg[9][NR];
     #pragma unroll
     for (int tap = 0; tap < 9; ++tap) {
       #pragma unroll
-      for (int nf = 0; nf < 2; ++nf) {
+      for (int nf = 0; nf < NR; ++nf) {
         int n = n0 + nf * 16 + fr;
         int ci0 = cs + kq;
         bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -337,12 +343,12 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
       for (int kw = 0; kw < 3; ++kw) {
         const int toff = (kh * XC + kw) * PS;   // compile-time per (kh,kw)
         #pragma unroll
-        for (int mf = 0; mf < MFRAG; ++mf) {
+        for (int mf = 0; mf < MR; ++mf) {
           bf16x8 a = *reinterpret_cast<const bf16x8*>(&ldsX[aoff[mf] + toff]);
-          acc[mf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, wreg[kh * 3 + kw][0],
-                                                               acc[mf][0], 0, 0, 0);
-          acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, wreg[kh * 3 + kw][1],
-                                                               acc[mf][1], 0, 0, 0);
+          #pragma unroll
+          for (int nf = 0; nf < NR; ++nf)
+            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, wreg[kh * 3 + kw][nf], acc[mf][nf], 0, 0, 0);
         }
       }
     }
@@ -350,17 +356,17 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
   }
 
   #pragma unroll
-  for (int mf = 0; mf < MFRAG; ++mf) {
+  for (int mf = 0; mf < MR; ++mf) {
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int m_loc = wave * (MFRAG * 16) + mf * 16 + (lane >> 4) * 4 + r;
+      int m_loc = wm * (MR * 16) + mf * 16 + (lane >> 4) * 4 + r;
       int ib = m_loc / (TH * TW);
       int pix = m_loc % (TH * TW);
       int py = pix / TW, px = pix % TW;
       int ho = ty * TH + py;
       int64_t base = ((((int64_t)(img0 + ib)) * g.Ho + ho) * g.Wo + px) * g.Cout;
       #pragma unroll
-      for (int nf = 0; nf < 2; ++nf) {
+      for (int nf = 0; nf < NR; ++nf) {
         int n = n0 + nf * 16 + fr;
         if (n < g.Cout) {
           float v = acc[mf][nf][r];
@@ -648,7 +654,8 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   }
 
   // direct tiled 3x3 path: CIFAR geometries, any channel count.
-  // FAA_CONV_DIRECT=0 falls back to the im2col kernel below.
+  // FAA_CONV_DIRECT=0 falls back to the im2col kernel below; =small/=big
+  // force the narrow (BN=32) / wide (BN=64) wave configs for sweeps.
   {
     const char* de = getenv("FAA_CONV_DIRECT");
     bool want = !(de && de[0] == '0');
@@ -659,23 +666,45 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                         || (g.Wd == 8 && g.H == 8 && g.B % 2 == 0)));
     if (want && geom_ok) {
       int tiles_h = g.H / 8;
-      int gn = (g.Cout + 31) / 32;
       int ib = (g.Wd == 8) ? 2 : 1;
-      dim3 dgrid((unsigned)((g.B / ib) * tiles_h * gn));
-      #define CT_LAUNCH(TW_, IB_, MF_, HB_)                                        \
-        hipLaunchKernelGGL((conv3x3_tile_kernel<8, TW_, IB_, MF_, HB_>), dgrid,    \
-                           dim3(256), 0, stream, (const short*)xc.data_ptr(),      \
+      // wide config (BN=64): fewer weight re-reads per FLOP; only when the
+      // grid still fills the chip and for W=32 the TH=4 variant exists
+      int gn64 = (g.Cout + 63) / 64;
+      int tiles_h4 = g.H / 4;
+      int64_t blocks64 = (g.Wd == 32)
+          ? (int64_t)g.B * tiles_h4 * gn64
+          : (int64_t)(g.B / ib) * tiles_h * gn64;
+      bool big = (g.Cin > 64) && (blocks64 >= 512);
+      if (de && de[0] == 'b') big = true;
+      if (de && de[0] == 's') big = false;
+      #define CT_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, HB_, GRID_, THTILES_, GN_)   \
+        hipLaunchKernelGGL((conv3x3_tile_kernel<TH_, TW_, IB_, WN_, MR_, NR_, HB_>),\
+                           dim3((unsigned)(GRID_)), dim3(256), 0, stream,          \
+                           (const short*)xc.data_ptr(),                            \
                            (const short*)wc.data_ptr(), bptr,                      \
-                           (short*)y.data_ptr(), g, tiles_h, gn)
-      if (has_bias) {
-        if (g.Wd == 32) CT_LAUNCH(32, 1, 4, true);
-        else if (g.Wd == 16) CT_LAUNCH(16, 1, 2, true);
-        else CT_LAUNCH(8, 2, 2, true);
+                           (short*)y.data_ptr(), g, THTILES_, GN_)
+      #define CT_BOTH(TH_, TW_, IB_, WN_, MR_, NR_, GRID_, THTILES_, GN_)          \
+        do {                                                                       \
+          if (has_bias) CT_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, true,              \
+                                  GRID_, THTILES_, GN_);                           \
+          else CT_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, false,                      \
+                         GRID_, THTILES_, GN_);                                    \
+        } while (0)
+      if (big) {
+        if (g.Wd == 32)
+          CT_BOTH(4, 32, 1, 2, 4, 2, (int64_t)g.B * tiles_h4 * gn64, tiles_h4, gn64);
+        else if (g.Wd == 16)
+          CT_BOTH(8, 16, 1, 2, 4, 2, (int64_t)g.B * tiles_h * gn64, tiles_h, gn64);
+        else
+          CT_BOTH(8, 8, 2, 2, 4, 2, (int64_t)(g.B / 2) * gn64, 1, gn64);
       } else {
-        if (g.Wd == 32) CT_LAUNCH(32, 1, 4, false);
-        else if (g.Wd == 16) CT_LAUNCH(16, 1, 2, false);
-        else CT_LAUNCH(8, 2, 2, false);
+        int gn = (g.Cout + 31) / 32;
+        int64_t grid32 = (int64_t)(g.B / ib) * tiles_h * gn;
+        if (g.Wd == 32) CT_BOTH(8, 32, 1, 1, 4, 2, grid32, tiles_h, gn);
+        else if (g.Wd == 16) CT_BOTH(8, 16, 1, 1, 2, 2, grid32, tiles_h, gn);
+        else CT_BOTH(8, 8, 2, 1, 2, 2, grid32, tiles_h, gn);
       }
+      #undef CT_BOTH
       #undef CT_LAUNCH
       return y;
     }

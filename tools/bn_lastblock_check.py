@@ -1,48 +1,70 @@
 #!/usr/bin/env python3
-"""Round-2 validation for FAA_BN_LASTBLOCK=1 (run BEFORE adding a GPU
-test for it — the kernel tail was written without GPU budget left in
-round 1)."""
+"""Round-2 validation for FAA_BN_LASTBLOCK=1 (last reduce block performs
+the finalize inline via persistent counter + threadfence).
+
+Note on tolerances: the inline finalize sums the per-block partials in a
+different ORDER (serial b-loop) than the bn_finalize_kernel (64-lane
+strided + warp reduce), so mean/invstd differ by fp32 rounding and a bf16
+output element near a rounding boundary can legitimately flip one ULP.
+Both paths are therefore compared against the fp32 torch reference, plus
+a bf16-ULP-aware cross-check.
+"""
+import os
 import sys
+
 sys.path.insert(0, ".")
 import torch
 
 
+def _torch_ref(x, bn):
+    ref = torch.nn.functional.batch_norm(
+        x.float(), None, None, bn.weight.float(), bn.bias.float(),
+        training=True, momentum=bn.momentum, eps=bn.eps)
+    return ref
+
+
 def main():
-    """FAA_BN_LASTBLOCK=1 (last reduce block performs the finalize inline,
-    staged for round 2) must match the default two-kernel path."""
-    import os
     torch.manual_seed(3)
-    for Ch in (32, 61):
-        x = torch.randn(8, Ch, 16, 16, device=torch.device("cuda:0"), dtype=torch.bfloat16).contiguous(
+    dev = torch.device("cuda:0")
+    from fast_autoaugment_amd.ops.bnrelu import fused_bn_relu
+    for Ch in (32, 61, 320):
+        x = torch.randn(8, Ch, 16, 16, device=dev, dtype=torch.bfloat16).contiguous(
             memory_format=torch.channels_last)
-        bn = torch.nn.BatchNorm2d(Ch, momentum=0.3).to(torch.device("cuda:0"))
-        bn2 = torch.nn.BatchNorm2d(Ch, momentum=0.3).to(torch.device("cuda:0"))
+        bn = torch.nn.BatchNorm2d(Ch, momentum=0.3).to(dev)
+        bn2 = torch.nn.BatchNorm2d(Ch, momentum=0.3).to(dev)
         bn2.load_state_dict(bn.state_dict())
-        from fast_autoaugment_amd.ops.bnrelu import fused_bn_relu
         bn.train(); bn2.train()
+        ref = torch.relu(_torch_ref(x, bn))
         out_ref = fused_bn_relu(x, bn)
         os.environ["FAA_BN_LASTBLOCK"] = "1"
         try:
             out = fused_bn_relu(x, bn2)
         finally:
             os.environ.pop("FAA_BN_LASTBLOCK", None)
-        assert (out.float() - out_ref.float()).abs().max().item() < 1e-5
-        assert (bn2.running_mean - bn.running_mean).abs().max().item() < 1e-5
-        assert (bn2.running_var - bn.running_var).abs().max().item() < 1e-5
+        scale = ref.abs().max().item() + 1e-3
+        e_def = (out_ref.float() - ref).abs().max().item() / scale
+        e_lb = (out.float() - ref).abs().max().item() / scale
+        e_x = (out.float() - out_ref.float()).abs().max().item() / scale
+        e_rm = (bn2.running_mean - bn.running_mean).abs().max().item()
+        e_rv = (bn2.running_var - bn.running_var).abs().max().item()
+        print(f"C={Ch}: vs-ref default={e_def:.2e} lastblock={e_lb:.2e} "
+              f"cross={e_x:.2e} rmean={e_rm:.2e} rvar={e_rv:.2e}")
+        assert e_def < 2e-2 and e_lb < 2e-2, "path diverges from torch ref"
+        assert e_x < 1e-2, "cross-path difference beyond bf16 ULP noise"
+        assert e_rm < 1e-4 and e_rv < 1e-4, "running stats diverge"
 
 
 def main_bwd():
-    import os
     torch.manual_seed(5)
     dev = torch.device("cuda:0")
-    for Ch in (32, 61):
+    from fast_autoaugment_amd.ops.bnrelu import fused_bn_relu
+    for Ch in (32, 61, 320):
         x = torch.randn(8, Ch, 16, 16, device=dev, dtype=torch.bfloat16).contiguous(
             memory_format=torch.channels_last).requires_grad_(True)
         x2 = x.detach().clone().requires_grad_(True)
         bn = torch.nn.BatchNorm2d(Ch).to(dev)
         bn2 = torch.nn.BatchNorm2d(Ch).to(dev)
         bn2.load_state_dict(bn.state_dict())
-        from fast_autoaugment_amd.ops.bnrelu import fused_bn_relu
         g = torch.randn(8, Ch, 16, 16, device=dev, dtype=torch.bfloat16)
         fused_bn_relu(x, bn).backward(g)
         os.environ["FAA_BN_LASTBLOCK"] = "1"
@@ -50,12 +72,35 @@ def main_bwd():
             fused_bn_relu(x2, bn2).backward(g)
         finally:
             os.environ.pop("FAA_BN_LASTBLOCK", None)
-        assert (x.grad.float() - x2.grad.float()).abs().max().item() < 1e-5
-        assert (bn.weight.grad.float() - bn2.weight.grad.float()).abs().max().item() < 1e-4
-        assert (bn.bias.grad.float() - bn2.bias.grad.float()).abs().max().item() < 1e-4
+        sx = x.grad.float().abs().max().item() + 1e-3
+        e_dx = (x.grad.float() - x2.grad.float()).abs().max().item() / sx
+        e_dg = (bn.weight.grad.float() - bn2.weight.grad.float()).abs().max().item()
+        e_db = (bn.bias.grad.float() - bn2.bias.grad.float()).abs().max().item()
+        print(f"C={Ch} bwd: dx={e_dx:.2e} dgamma={e_dg:.2e} dbeta={e_db:.2e}")
+        assert e_dx < 1e-2 and e_dg < 1e-2 and e_db < 1e-2
+
+
+def stress():
+    """many back-to-back lastblock calls (counter reset discipline)."""
+    os.environ["FAA_BN_LASTBLOCK"] = "1"
+    try:
+        from fast_autoaugment_amd.ops.bnrelu import fused_bn_relu
+        dev = torch.device("cuda:0")
+        bn = torch.nn.BatchNorm2d(64).to(dev).train()
+        x = torch.randn(32, 64, 16, 16, device=dev, dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        ref = torch.relu(_torch_ref(x, bn))
+        for _ in range(200):
+            out = fused_bn_relu(x, bn)
+        err = (out.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+        print(f"stress x200: err={err:.2e}")
+        assert err < 2e-2
+    finally:
+        os.environ.pop("FAA_BN_LASTBLOCK", None)
 
 
 if __name__ == "__main__":
     main()
     main_bwd()
+    stress()
     print("BN_LASTBLOCK_OK")
