@@ -38,7 +38,14 @@ def parse_args():
                    help="gather: backward assigns fresh grads (no per-param "
                         "accumulation kernels) packed by one multi-tensor "
                         "kernel; flat: grads accumulate into flat views")
-    return p.parse_args()
+    args = p.parse_args()
+    # env overrides so subprocess harnesses (tools/nan_flake.py) can toggle
+    # the capture mode without changing the CLI contract
+    if os.environ.get("FAA_BENCH_GRAPHS"):
+        args.graphs = int(os.environ["FAA_BENCH_GRAPHS"])
+    if os.environ.get("FAA_BENCH_GRAD_MODE"):
+        args.grad_mode = os.environ["FAA_BENCH_GRAD_MODE"]
+    return args
 
 
 def main():

@@ -244,11 +244,13 @@ void conv_fwd_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
 //  * K loop = cin slabs of CS=32 (zero-padded tail), 9 taps each
 //  * LDS pixel stride PS=40 shorts: 16B-aligned ds_read_b128 and 2-way-
 //    max bank aliasing (fr*20 mod 32 covers 8 banks x 2 lanes = free)
-template <int TH, int TW, int IB, int WN, int MR, int NR, bool HAS_BIAS>
+template <int TH, int TW, int IB, int WN, int MR, int NR, bool HAS_BIAS,
+          bool SPLIT = false>
 __global__ __launch_bounds__(256)
 void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
                          const short* __restrict__ bias, short* __restrict__ Y,
-                         ConvGeom g, int tiles_h, int grid_n) {
+                         ConvGeom g, int tiles_h, int grid_n,
+                         float* __restrict__ Yacc = nullptr) {
   constexpr int CS = 32;             // cin slab depth (one MFMA k)
   constexpr int PS = 40;             // per-pixel LDS stride in shorts
   constexpr int XR = TH + 2;
@@ -292,7 +294,15 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
   f32x4 acc[MR][NR] = {};
   const int row0 = ty * TH - 1;      // pad = 1
 
-  for (int cs = 0; cs < g.Cin; cs += CS) {
+  int cs_lo = 0, cs_hi = g.Cin;
+  if (SPLIT) {                       // blockIdx.y partitions the cin slabs
+    int slabs = (g.Cin + CS - 1) / CS;
+    int per = (slabs + gridDim.y - 1) / gridDim.y;
+    cs_lo = blockIdx.y * per * CS;
+    cs_hi = min(g.Cin, cs_lo + per * CS);
+    if (cs_lo >= cs_hi) return;
+  }
+  for (int cs = cs_lo; cs < cs_hi; cs += CS) {
     // stage the X tile slab (borders/tails zero-filled)
     constexpr int NV = IB * XR * XC * 4;     // bf16x8 stores
     for (int v = tid; v < NV; v += 256) {
@@ -370,8 +380,13 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
         int n = n0 + nf * 16 + fr;
         if (n < g.Cout) {
           float v = acc[mf][nf][r];
-          if (HAS_BIAS) v += b2f(bias[n]);
-          Y[base + n] = f2b(v);
+          if (SPLIT) {
+            int64_t m = ((int64_t)(img0 + ib) * g.Ho + ho) * g.Wo + px;
+            atomicAdd(&Yacc[m * g.Cout + n], v);
+          } else {
+            if (HAS_BIAS) v += b2f(bias[n]);
+            Y[base + n] = f2b(v);
+          }
         }
       }
     }
@@ -499,6 +514,178 @@ void conv_wrw_kernel(const short* __restrict__ X, const short* __restrict__ dY,
         atomicAdd(&dWacc[(int64_t)kp * g.Cout + n], acc[fk][r]);
     }
   }
+}
+
+// ------------------------------------------------- wrw v3 (direct tiled)
+// dW_tap[ci][co] = sum_p X[p + off_tap][ci] * dY[p][co] — contraction over
+// OUTPUT pixels p. The round-1 conv_wrw_kernel (im2col split-K, 32-deep m
+// chunks) loses 1.2-4x to MIOpen at C >= 160: per 32-m chunk it re-stages
+// both operands and issues only 3 MFMAs/wave. Here a block stages a
+// TRANSPOSED X tile (with halo) and dY tile once and contracts PIX_OUT
+// pixels for ALL taps from LDS: per 32-pixel chunk a wave issues
+// KK*KK*MR*NR MFMAs with 2 fresh LDS b-frags.
+//
+//  * block tile: 64 ci x 64 co, waves 2x2 (each 32x32), acc[KK*KK][2][2]
+//  * pixel tiles: full image rows (TW = W), TH rows; grid.y m-slices
+//  * LDS rows padded to wrw3_pitch: 16B-aligned and 2-way-max bank step
+//  * fp32 atomicAdd into dWacc[tap][Cin][Cout]; cast kernel reorders to
+//    the torch channels_last weight layout
+constexpr int wrw3_pitch(int n) {
+  for (int p = (n + 7) / 8 * 8;; p += 8) {
+    int step = (p / 2) % 32;
+    if (step == 4 || step == 12 || step == 20 || step == 28) return p;
+  }
+}
+
+template <int TH, int TW, int KK>
+__global__ __launch_bounds__(256)
+void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
+                      float* __restrict__ dWacc, ConvGeom g,
+                      int ci_tiles, int co_tiles, int tiles_per_slice) {
+  constexpr int XR = (KK == 3) ? TH + 2 : TH;
+  constexpr int XC = (KK == 3) ? TW + 2 : TW;
+  constexpr int PIX_IN = XR * XC;
+  constexpr int PIX_OUT = TH * TW;
+  constexpr int XTP = wrw3_pitch(PIX_IN);
+  constexpr int YTP = wrw3_pitch(PIX_OUT);
+  static_assert(PIX_OUT % 32 == 0, "pixel tile must be a multiple of 32");
+
+  __shared__ short ldsXT[64 * XTP];
+  __shared__ short ldsYT[64 * YTP];
+
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, idx = wg / 8;
+    if (q > 0) wg = (xcd < r) ? (xcd * (q + 1) + idx) : (r * (q + 1) + (xcd - r) * q + idx);
+  }
+  const int ci_t = wg % ci_tiles;
+  const int co_t = wg / ci_tiles;
+  const int ci0g = ci_t * 64;
+  const int co0g = co_t * 64;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 1;          // ci half
+  const int wn = wave & 1;           // co half
+  const int fr = lane & 15;
+  const int kq = (lane >> 4) * 8;
+
+  const int tiles_h = g.H / TH;
+  const int tiles_total = g.B * tiles_h;
+  int t0 = blockIdx.y * tiles_per_slice;
+  int t1 = min(tiles_total, t0 + tiles_per_slice);
+
+  f32x4 acc[KK * KK][2][2] = {};
+
+  for (int tile = t0; tile < t1; ++tile) {
+    const int b = tile / tiles_h;
+    const int ty = tile % tiles_h;
+    const int row0 = ty * TH - (KK == 3 ? 1 : 0);
+    // ---- stage X^T [ci][input pixel] (zero-filled borders/tails)
+    for (int v = tid; v < PIX_IN * 8; v += 256) {
+      int kc = v & 7;
+      int cell = v >> 3;
+      int row = cell / XC, col = cell % XC;
+      int hi = row0 + row;
+      int wi = col - (KK == 3 ? 1 : 0);
+      int ci0 = ci0g + kc * 8;
+      bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.Wd && ci0 < g.Cin) {
+        const short* p = X + (((int64_t)b * g.H + hi) * g.Wd + wi) * g.Cin + ci0;
+        if (ci0 + 8 <= g.Cin) {
+          val = *reinterpret_cast<const bf16x8*>(p);
+        } else {
+          for (int j = 0; j < g.Cin - ci0; ++j) val[j] = p[j];
+        }
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) ldsXT[(kc * 8 + j) * XTP + cell] = val[j];
+    }
+    // ---- stage dY^T [co][output pixel]
+    for (int v = tid; v < PIX_OUT * 8; v += 256) {
+      int kc = v & 7;
+      int cell = v >> 3;
+      int row = cell / TW, col = cell % TW;
+      int ho = ty * TH + row;
+      int co0 = co0g + kc * 8;
+      bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (ho < g.Ho && co0 < g.Cout) {
+        const short* p = dY + (((int64_t)b * g.Ho + ho) * g.Wo + col) * g.Cout + co0;
+        if (co0 + 8 <= g.Cout) {
+          val = *reinterpret_cast<const bf16x8*>(p);
+        } else {
+          for (int j = 0; j < g.Cout - co0; ++j) val[j] = p[j];
+        }
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) ldsYT[(kc * 8 + j) * YTP + cell] = val[j];
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int pc = 0; pc < PIX_OUT / 32; ++pc) {
+      const int p8 = pc * 32 + kq;       // this lane's first pixel
+      const int py = p8 / TW, px = p8 % TW;
+      bf16x8 bfrag[2];
+      #pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        bfrag[nf] = *reinterpret_cast<const bf16x8*>(
+            &ldsYT[(wn * 32 + nf * 16 + fr) * YTP + pc * 32 + kq]);
+      #pragma unroll
+      for (int kh = 0; kh < KK; ++kh) {
+        #pragma unroll
+        for (int kw = 0; kw < KK; ++kw) {
+          const int ip = (py + kh) * XC + (px + kw);
+          #pragma unroll
+          for (int mf = 0; mf < 2; ++mf) {
+            bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                &ldsXT[(wm * 32 + mf * 16 + fr) * XTP + ip]);
+            #pragma unroll
+            for (int nf = 0; nf < 2; ++nf)
+              acc[kh * KK + kw][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, bfrag[nf], acc[kh * KK + kw][mf][nf], 0, 0, 0);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: one atomicAdd per accumulator element
+  #pragma unroll
+  for (int tap = 0; tap < KK * KK; ++tap) {
+    #pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int ci = ci0g + wm * 32 + mf * 16 + (lane >> 4) * 4 + r;
+        if (ci >= g.Cin) continue;
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          int co = co0g + wn * 32 + nf * 16 + fr;
+          if (co < g.Cout)
+            atomicAdd(&dWacc[((int64_t)tap * g.Cin + ci) * g.Cout + co],
+                      acc[tap][mf][nf][r]);
+        }
+      }
+    }
+  }
+}
+
+// dWacc[tap][Cin][Cout] -> torch channels_last weight grad [Cout][KH][KW][Cin]
+__global__ void wrw3_cast_kernel(const float* __restrict__ dWacc, short* __restrict__ dW,
+                                 int Cout, int KK2, int Cin) {
+  int64_t total = (int64_t)Cout * KK2 * Cin;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int ci = i % Cin;
+  int64_t t = i / Cin;
+  int tap = t % KK2;
+  int co = t / KK2;
+  dW[i] = f2b(dWacc[((int64_t)tap * Cin + ci) * Cout + co]);
 }
 
 // cast dW accumulator [kpad][Cout] back to torch layout [Cout][KH][KW][Cin]
@@ -691,12 +878,55 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                          GRID_, THTILES_, GN_);                                    \
         } while (0)
       if (big) {
-        if (g.Wd == 32)
+        if (g.Wd == 32) {
           CT_BOTH(4, 32, 1, 2, 4, 2, (int64_t)g.B * tiles_h4 * gn64, tiles_h4, gn64);
-        else if (g.Wd == 16)
+        } else if (g.Wd == 16) {
           CT_BOTH(8, 16, 1, 2, 4, 2, (int64_t)g.B * tiles_h * gn64, tiles_h, gn64);
-        else
-          CT_BOTH(8, 8, 2, 2, 4, 2, (int64_t)(g.B / 2) * gn64, 1, gn64);
+        } else {
+          // W=8 deep stages: M is tiny (B*64 px) and K is huge, so weight
+          // re-reads dominate. IB=4 (M=256/block) halves per-FLOP weight
+          // traffic; split-K over cin slabs restores grid occupancy.
+          // FAA_CONV_D8 = ib2 | ib4 | ib2sk2 | ib4sk2 | ib4sk4 forces.
+          const char* d8 = getenv("FAA_CONV_D8");
+          int ibb = 2, sk = 1;
+          if (d8) {
+            ibb = (d8[2] == '4') ? 4 : 2;
+            sk = (d8[3] == 's') ? (d8[5] - '0') : 1;
+          } else if (g.B % 4 == 0) {
+            ibb = 4;
+            int64_t blk = (int64_t)(g.B / 4) * gn64;
+            sk = (blk < 512 && g.Cin >= 256) ? (blk < 256 ? 4 : 2) : 1;
+          }
+          if (ibb == 4 && g.B % 4 != 0) ibb = 2;
+          if (sk > 1) {
+            auto yacc = torch::zeros({(int64_t)M, (int64_t)g.Cout},
+                                     xc.options().dtype(torch::kFloat32));
+            dim3 sgrid((unsigned)((g.B / ibb) * gn64), sk);
+            #define CT_SK(IB_, MR_)                                               \
+              hipLaunchKernelGGL((conv3x3_tile_kernel<8, 8, IB_, 2, MR_, 2,       \
+                                  false, true>), sgrid, dim3(256), 0, stream,     \
+                                 (const short*)xc.data_ptr(),                     \
+                                 (const short*)wc.data_ptr(), nullptr,            \
+                                 (short*)y.data_ptr(), g, 1, gn64,                \
+                                 yacc.data_ptr<float>())
+            if (ibb == 4) CT_SK(4, 8); else CT_SK(2, 4);
+            #undef CT_SK
+            int64_t total = (int64_t)M * g.Cout;
+            dim3 cgrid((unsigned)((total + 255) / 256));
+            if (has_bias)
+              hipLaunchKernelGGL((conv_splitk_cast_kernel<true>), cgrid, dim3(256),
+                                 0, stream, yacc.data_ptr<float>(), bptr,
+                                 (short*)y.data_ptr(), total, g.Cout);
+            else
+              hipLaunchKernelGGL((conv_splitk_cast_kernel<false>), cgrid, dim3(256),
+                                 0, stream, yacc.data_ptr<float>(), nullptr,
+                                 (short*)y.data_ptr(), total, g.Cout);
+          } else if (ibb == 4) {
+            CT_BOTH(8, 8, 4, 2, 8, 2, (int64_t)(g.B / 4) * gn64, 1, gn64);
+          } else {
+            CT_BOTH(8, 8, 2, 2, 4, 2, (int64_t)(g.B / 2) * gn64, 1, gn64);
+          }
+        }
       } else {
         int gn = (g.Cout + 31) / 32;
         int64_t grid32 = (int64_t)(g.B / ib) * tiles_h * gn;
@@ -867,6 +1097,58 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
   auto stream = at::hip::getCurrentHIPStream().stream();
 
   auto f32 = xc.options().dtype(torch::kFloat32);
+
+  // wrw v3 direct tiled path (stride-1 3x3/1x1 CIFAR geometries)
+  {
+    const char* e = getenv("FAA_WRW_V3");
+    bool want = !(e && e[0] == '0');
+    bool k3 = (KH == 3 && KW == 3 && pad == 1);
+    bool k1 = (KH == 1 && KW == 1 && pad == 0);
+    bool geom_ok = stride == 1 && (k3 || k1)
+        && ((g.Wd == 32 && g.H % 4 == 0)
+            || ((g.Wd == 16 || g.Wd == 8) && g.H % 8 == 0));
+    if (want && geom_ok) {
+      int KK = k3 ? 3 : 1;
+      int ci_tiles = (g.Cin + 63) / 64;
+      int co_tiles = (Cout + 63) / 64;
+      int TH = (g.Wd == 32) ? 4 : 8;
+      int tiles_total = g.B * (g.H / TH);
+      int blocks_xy = ci_tiles * co_tiles;
+      int slices = std::max(1, std::min(768 / blocks_xy, tiles_total));
+      int tps = (tiles_total + slices - 1) / slices;
+      slices = (tiles_total + tps - 1) / tps;
+      auto dwacc = torch::zeros({(int64_t)KK * KK, g.Cin, (int64_t)Cout}, f32);
+      dim3 grid(blocks_xy, slices);
+      #define WRW3_LAUNCH(TH_, TW_, KK_)                                        \
+        hipLaunchKernelGGL((conv_wrw3_kernel<TH_, TW_, KK_>), grid, dim3(256),  \
+                           0, stream, (const short*)xc.data_ptr(),              \
+                           (const short*)dyc.data_ptr(),                        \
+                           dwacc.data_ptr<float>(), g, ci_tiles, co_tiles, tps)
+      if (k3) {
+        if (g.Wd == 32) WRW3_LAUNCH(4, 32, 3);
+        else if (g.Wd == 16) WRW3_LAUNCH(8, 16, 3);
+        else WRW3_LAUNCH(8, 8, 3);
+      } else {
+        if (g.Wd == 32) WRW3_LAUNCH(4, 32, 1);
+        else if (g.Wd == 16) WRW3_LAUNCH(8, 16, 1);
+        else WRW3_LAUNCH(8, 8, 1);
+      }
+      #undef WRW3_LAUNCH
+      auto dw = torch::empty({Cout, g.Cin, (int64_t)KH, (int64_t)KW},
+                             xc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+      int64_t total3 = (int64_t)Cout * KK * KK * g.Cin;
+      hipLaunchKernelGGL(wrw3_cast_kernel, dim3((total3 + 255) / 256), dim3(256),
+                         0, stream, dwacc.data_ptr<float>(), (short*)dw.data_ptr(),
+                         Cout, KK * KK, g.Cin);
+      torch::Tensor dbias;
+      if (want_bias) {
+        if (Cout % 8 == 0) dbias = colsum_bf16(dyc);
+        else dbias = dyc.sum(/*dim=*/{0, 2, 3}).to(xc.scalar_type());
+      }
+      return {dw, dbias};
+    }
+  }
+
   auto dwacc = torch::zeros({g.kpad, Cout}, f32);
   int grid_k = g.kpad / 32;
   int grid_n = (Cout + 63) / 64;

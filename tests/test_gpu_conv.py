@@ -191,3 +191,27 @@ def test_colsum_v2_matches_reference(M, Ch):
     got = C.colsum_bf16(dy4).float()
     err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
     assert err < 1e-2, f"colsum rel err {err}"
+
+
+@pytest.mark.parametrize("d8", ["ib2", "ib4", "ib2sk2", "ib4sk2", "ib4sk4"])
+def test_conv_fwd_direct_d8_variants(d8):
+    """W=8 big-C direct-kernel variants (IB4 / cin-split-K) vs fp32 torch."""
+    import os
+    torch.manual_seed(7)
+    B, Cin, H, Cout = 8, 256, 8, 256
+    x = torch.randn(B, Cin, H, H, device=dev()) * 0.5
+    w = torch.randn(Cout, Cin, 3, 3, device=dev()) * 0.02
+    b = torch.randn(Cout, device=dev()) * 0.1
+    ref = torch.nn.functional.conv2d(x, w, b, stride=1, padding=1)
+    os.environ["FAA_CONV_DIRECT"] = "big"
+    os.environ["FAA_CONV_D8"] = d8
+    try:
+        got = C.conv2d_fwd(
+            x.bfloat16().contiguous(memory_format=torch.channels_last),
+            w.bfloat16().contiguous(memory_format=torch.channels_last),
+            b.bfloat16(), 1, 1).float()
+    finally:
+        os.environ.pop("FAA_CONV_D8", None)
+        os.environ.pop("FAA_CONV_DIRECT", None)
+    err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert err < 2e-2, f"d8={d8} rel err {err}"

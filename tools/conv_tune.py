@@ -28,8 +28,35 @@ def bench(fn, iters=50):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--big", action="store_true")
+    ap.add_argument("--d8", action="store_true",
+                    help="sweep FAA_CONV_D8 variants on the deep 8x8 shapes")
     ap.add_argument("--iters", type=int, default=50)
     args = ap.parse_args()
+    if args.d8:
+        from fast_autoaugment_amd.ops import ext as _ext
+        C = _ext()
+        variants = ["ib2", "ib4", "ib2sk2", "ib4sk2", "ib4sk4"]
+        shapes8 = [(128, 128, 8, 128, 3, 1), (128, 256, 8, 256, 3, 1),
+                   (128, 384, 8, 384, 3, 1), (128, 640, 8, 640, 3, 1)]
+        print(f"{'shape':<26} {'miopen':>8} {'auto':>8} "
+              + " ".join(f"{v:>8}" for v in variants))
+        for B, Cin, H, Cout, k, s in shapes8:
+            x = (torch.randn(B, Cin, H, H, device="cuda") * 0.5).bfloat16() \
+                .contiguous(memory_format=torch.channels_last)
+            w = (torch.randn(Cout, Cin, k, k, device="cuda") * 0.05).bfloat16() \
+                .contiguous(memory_format=torch.channels_last)
+            row = [f"{B}x{Cin}x{H}^2->{Cout}"]
+            row.append(f"{bench(lambda: torch.nn.functional.conv2d(x, w, stride=s, padding=k//2), args.iters):8.1f}")
+            os.environ.pop("FAA_CONV_D8", None)
+            os.environ["FAA_CONV_DIRECT"] = "big"
+            row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
+            for v in variants:
+                os.environ["FAA_CONV_D8"] = v
+                row.append(f"{bench(lambda: C.conv2d_fwd(x, w, torch.Tensor(), s, k//2), args.iters):8.1f}")
+            os.environ.pop("FAA_CONV_D8", None)
+            os.environ.pop("FAA_CONV_DIRECT", None)
+            print(" ".join(row))
+        return
     from fast_autoaugment_amd.ops import ext
     C = ext()
     shapes = ([(128, 256, 16, 256, 3, 1), (128, 384, 8, 384, 3, 1),
