@@ -154,13 +154,26 @@ def hflip(img: np.ndarray) -> np.ndarray:
 
 # ------------------------------------------------------------ program executor
 
-def apply_program_image(img: np.ndarray, prog: np.ndarray) -> np.ndarray:
+def sample_pairing(img: np.ndarray, partner_raw: np.ndarray, alpha: float) -> np.ndarray:
+    """PIL Image.blend(img, partner, alpha): float lerp, clip, TRUNCATING
+    uint8 cast (Blend.c CLIP8) — reference augmentations.py:147-152."""
+    f = img.astype(np.float32) + alpha * (partner_raw.astype(np.float32)
+                                          - img.astype(np.float32))
+    return np.clip(f, 0, 255).astype(np.uint8)
+
+
+def apply_program_image(img: np.ndarray, prog: np.ndarray,
+                        raw_batch: np.ndarray = None) -> np.ndarray:
     """Run one image's program (PROG_SLOTS x PROG_WIDTH float32)."""
     for s in range(prog.shape[0]):
         code = int(prog[s, 0])
         p = prog[s, 1:]
         if code == OpCode.NOP:
             continue
+        elif code == OpCode.PAIRING:
+            if raw_batch is None:
+                continue
+            img = sample_pairing(img, raw_batch[int(p[1])], float(p[0]))
         elif code == OpCode.AFFINE:
             img = affine_nearest(img, tuple(float(x) for x in p[:6]))
         elif code == OpCode.AUTOCONTRAST:
@@ -192,7 +205,8 @@ def apply_program_image(img: np.ndarray, prog: np.ndarray) -> np.ndarray:
 
 def apply_program_batch(batch: np.ndarray, prog: np.ndarray) -> np.ndarray:
     """batch uint8 [B,H,W,3], prog float32 [B,PROG_SLOTS,PROG_WIDTH]."""
-    return np.stack([apply_program_image(batch[b], prog[b]) for b in range(batch.shape[0])])
+    return np.stack([apply_program_image(batch[b], prog[b], raw_batch=batch)
+                     for b in range(batch.shape[0])])
 
 
 def apply_post_batch(batch: np.ndarray, post: np.ndarray,

@@ -37,6 +37,7 @@ class OpCode(enum.IntEnum):
     BRIGHTNESS = 10
     SHARPNESS = 11
     CUTOUT = 12       # p0..p3 = x0,y0,x1,y1 (filled with 125,123,114)
+    PAIRING = 13      # p0 = alpha, p1 = partner batch slot (raw image blend)
 
 
 # name -> (low, high) level ranges (reference augmentations.py:156-182)
@@ -128,6 +129,12 @@ def _op_to_slot(name: str, value: float, w: int, h: int, rng: np.random.Generato
         return int(OpCode.BRIGHTNESS), [value, 0, 0, 0, 0, 0]
     if name == "Sharpness":
         return int(OpCode.SHARPNESS), [value, 0, 0, 0, 0, 0]
+    if name == "SamplePairing":
+        # reference augmentations.py:147-152 blends with a random raw train
+        # image; unused by every shipped policy (commented out of
+        # augment_list at :173). Here the partner is a random batch slot's
+        # raw image, drawn by host RNG like all other randomness.
+        return int(OpCode.PAIRING), [value, -1.0, 0, 0, 0, 0]
     if name in ("Cutout", "CutoutAbs"):
         # reference augmentations.py:125-150: center uniform over the image,
         # clipped at the edges; v<=0 is a no-op for Cutout.
@@ -169,6 +176,8 @@ def compile_program(policy: Sequence, batch: int, w: int, h: int,
             code, params = _op_to_slot(name, level_to_value(name, level), w, h, rng)
             if code == OpCode.NOP:
                 continue
+            if code == OpCode.PAIRING:
+                params[1] = float(rng.integers(0, batch))  # partner batch slot
             prog[b, slot, 0] = code
             prog[b, slot, 1:1 + len(params)] = params
             slot += 1
@@ -324,6 +333,11 @@ def compile_program_fast(policy: Sequence, batch: int, w: int, h: int,
                             "Posterize2": OpCode.POSTERIZE, "Contrast": OpCode.CONTRAST,
                             "Color": OpCode.COLOR, "Brightness": OpCode.BRIGHTNESS,
                             "Sharpness": OpCode.SHARPNESS}
+                if name == "SamplePairing":
+                    prog[idx, pos, 0] = OpCode.PAIRING
+                    prog[idx, pos, 1] = val
+                    prog[idx, pos, 2] = rng.integers(0, batch, size=len(idx))
+                    continue
                 code = code_map.get(name)
                 if code is None:
                     continue

@@ -31,6 +31,7 @@ enum OpCode {
   OP_NOP = 0, OP_AFFINE = 1, OP_AUTOCONTRAST = 2, OP_INVERT = 3, OP_EQUALIZE = 4,
   OP_FLIP = 5, OP_SOLARIZE = 6, OP_POSTERIZE = 7, OP_CONTRAST = 8, OP_COLOR = 9,
   OP_BRIGHTNESS = 10, OP_SHARPNESS = 11, OP_CUTOUT = 12,
+  OP_PAIRING = 13,
 };
 
 __device__ __forceinline__ uint32_t pack_rgb(int r, int g, int b) {
@@ -195,10 +196,29 @@ __device__ void op_cutout(const uint32_t* src, uint32_t* dst, int W, int H,
   }
 }
 
+// SamplePairing (reference augmentations.py:147-152): blend with another
+// batch slot's RAW image. PIL Image.blend = float lerp + clip + truncating
+// uint8 cast.
+__device__ void op_pairing(const uint32_t* src, uint32_t* dst, int n,
+                           const uint8_t* raw2, float alpha) {
+  #pragma clang fp contract(off)
+  for (int i = threadIdx.x; i < n; i += blockDim.x) {
+    uint32_t px = src[i];
+    float r = (float)ch_r(px) + alpha * ((float)raw2[i * 3 + 0] - (float)ch_r(px));
+    float g = (float)ch_g(px) + alpha * ((float)raw2[i * 3 + 1] - (float)ch_g(px));
+    float b = (float)ch_b(px) + alpha * ((float)raw2[i * 3 + 2] - (float)ch_b(px));
+    dst[i] = pack_rgb((int)fminf(fmaxf(r, 0.0f), 255.0f),
+                      (int)fminf(fmaxf(g, 0.0f), 255.0f),
+                      (int)fminf(fmaxf(b, 0.0f), 255.0f));
+  }
+}
+
 // run one image's op program over ping-pong buffers; returns final src
 __device__ uint32_t* run_program(uint32_t* src, uint32_t* dst, int W, int H,
                                  const float* bp, uint32_t* hist, uint8_t* lut,
-                                 uint32_t* scalar_acc) {
+                                 uint32_t* scalar_acc,
+                                 const uint8_t* images, int64_t img_stride,
+                                 const int64_t* sel) {
   const int n = W * H;
   for (int s = 0; s < PROG_SLOTS; ++s) {
     const float* slot = bp + s * PROG_WIDTH;
@@ -277,6 +297,9 @@ __device__ uint32_t* run_program(uint32_t* src, uint32_t* dst, int W, int H,
       case OP_CUTOUT:
         op_cutout(src, dst, W, H, p);
         break;
+      case OP_PAIRING:
+        op_pairing(src, dst, n, images + sel[(int)p[1]] * img_stride, p[0]);
+        break;
       default:
         for (int i = threadIdx.x; i < n; i += blockDim.x) dst[i] = src[i];
     }
@@ -325,7 +348,8 @@ __global__ void aug_pipeline_kernel(
     uint32_t* src = bufA;
     uint32_t* dst = bufB;
     const float* bp = prog + (int64_t)b * PROG_SLOTS * PROG_WIDTH;
-    src = run_program(src, dst, W, H, bp, hist, lut, scalar_acc);
+    src = run_program(src, dst, W, H, bp, hist, lut, scalar_acc,
+                      images, img_stride, sel);
     dst = (src == bufA) ? bufB : bufA;
 
     // ---- post stage: crop-shift, hflip, normalize, cutout-to-zero ---------
@@ -443,7 +467,8 @@ __global__ void aug_pipeline_in_kernel(
     __syncthreads();
 
     const float* bp = prog + (int64_t)b * PROG_SLOTS * PROG_WIDTH;
-    uint32_t* src = run_program(bufA, bufB, W, H, bp, hist, lut, scalar_acc);
+    uint32_t* src = run_program(bufA, bufB, W, H, bp, hist, lut, scalar_acc,
+                                images, img_stride, sel);
     uint32_t* dst = (src == bufA) ? bufB : bufA;
 
     const float* pp = post + (int64_t)b * 18;

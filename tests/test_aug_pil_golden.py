@@ -164,3 +164,12 @@ def test_cutout_fill():
 def test_hflip():
     a = rand_img()
     np.testing.assert_array_equal(cpu_exec.hflip(a), from_pil(PIL.ImageOps.mirror(to_pil(a))))
+
+
+@pytest.mark.parametrize("alpha", [0.0, 0.13, 0.4])
+def test_sample_pairing(alpha):
+    """SamplePairing = PIL Image.blend (reference augmentations.py:147-152)."""
+    a, b = rand_img(), rand_img()
+    ref = from_pil(PIL.Image.blend(to_pil(a), to_pil(b), alpha))
+    got = cpu_exec.sample_pairing(a, b, alpha)
+    assert np.abs(got.astype(int) - ref.astype(int)).max() <= 1

@@ -62,6 +62,7 @@ def test_aug_pipeline_post_only():
     (10, [1.9, 0, 0, 0, 0, 0]),                    # brightness
     (11, [0.3, 0, 0, 0, 0, 0]),                    # sharpness
     (12, [5, 7, 20, 22, 0, 0]),                    # cutout fill
+    (13, [0.35, 3, 0, 0, 0, 0]),                   # sample pairing (batch slot 3)
 ])
 def test_aug_pipeline_single_op(code, params):
     def setp(prog, rng):
