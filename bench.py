@@ -25,8 +25,8 @@ import torch
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=50)
-    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--batch", type=int, default=128, help="per-GPU batch (reference conf)")
     p.add_argument("--model", type=str, default="wresnet40_2")
     p.add_argument("--dataset", type=str, default="cifar10")
@@ -247,15 +247,28 @@ def main():
         data = torch.from_numpy(out).permute(0, 3, 1, 2).contiguous()
         return data, store.labels[torch.from_numpy(np.ascontiguousarray(sel))]
 
+    # FAA_BENCH_FIXED_DATA=1: replace the in-graph aug pipeline +
+    # index_select with one pre-generated static batch (nan_flake bisect:
+    # is aug-in-graph a required ingredient of the corruption?)
+    fixed_data = os.environ.get("FAA_BENCH_FIXED_DATA") == "1"
+    fixed = {}
+
     def gpu_fwd_bwd():
         """aug + forward + loss + backward on static inputs (capturable)."""
-        if imagenet:
-            data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s, post_s,
-                                            mean_t, std_t, out_size, out_size, bf16)
+        if fixed_data:
+            if "data" not in fixed:
+                fixed["data"] = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
+                                                mean_t, std_t, bf16).clone()
+                fixed["label"] = store.labels.index_select(0, sel_s).clone()
+            data, label = fixed["data"], fixed["label"]
         else:
-            data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
-                                   mean_t, std_t, bf16)
-        label = store.labels.index_select(0, sel_s)
+            if imagenet:
+                data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s, post_s,
+                                                mean_t, std_t, out_size, out_size, bf16)
+            else:
+                data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
+                                       mean_t, std_t, bf16)
+            label = store.labels.index_select(0, sel_s)
         preds = model(data)
         loss = crit(preds, label)
         loss.backward()
@@ -435,6 +448,46 @@ def main():
                 print(f"# debug eval loss: {l.item():.4f}", file=__import__('sys').stderr)
                 model.train()
 
+    if (not cpu_mode) and rank == 0 and os.environ.get("FAA_BENCH_PHASES", "1") == "1":
+        # per-phase breakdown (EAGER re-measure, so graphed totals are lower;
+        # use this to LOCALIZE regressions, not as the headline)
+        ev = [torch.cuda.Event(enable_timing=True) for _ in range(5)]
+        acc = [0.0] * 4
+        iters = 10
+        for _ in range(iters):
+            upload_next()
+            torch.cuda.synchronize()
+            ev[0].record()
+            if imagenet:
+                data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s, post_s,
+                                                mean_t, std_t, out_size, out_size, bf16)
+            else:
+                data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
+                                       mean_t, std_t, bf16)
+            label = store.labels.index_select(0, sel_s)
+            ev[1].record()
+            preds = model(data)
+            loss = crit(preds, label)
+            ev[2].record()
+            for p_ in flat.params:
+                p_.grad = None
+            loss.backward()
+            ev[3].record()
+            opt.step()
+            ev[4].record()
+            torch.cuda.synchronize()
+            for i in range(4):
+                acc[i] += ev[i].elapsed_time(ev[i + 1])
+        print(f"# phases eager ms/step (n={iters}): aug={acc[0]/iters:.3f} "
+              f"fwd={acc[1]/iters:.3f} bwd={acc[2]/iters:.3f} opt={acc[3]/iters:.3f}",
+              flush=True)
+
+    if os.environ.get("FAA_DBIAS_DEBUG") == "1" and not cpu_mode:
+        from fast_autoaugment_amd.ops.conv import dbias_debug_max
+        torch.cuda.synchronize()
+        print(f"# dbias colsum-vs-sum max abs diff: {dbias_debug_max(dev):.6f}",
+              flush=True)
+
     save_p = os.environ.get("FAA_BENCH_SAVE")
     if save_p and rank == 0:
         if os.environ.get("FAA_BENCH_SAVE_GRAD") == "1":
@@ -445,6 +498,22 @@ def main():
 
     total_images = args.batch * world_size * args.steps
     ips = total_images / elapsed
+    # vs_baseline: BASELINE.md publishes no images/sec (only GPU-hours), so
+    # the ratio is against the round-1 driver-measured value on this same
+    # config (BENCH_r01.json: 20817.3 img/s, wresnet40_2 b128 1xMI355X),
+    # overridable via FAA_BASELINE_IPS. Other configs stay null.
+    _R01 = {("wresnet40_2", 128, 1): 20817.3}
+    vs_baseline = None
+    baseline_src = None
+    env_b = os.environ.get("FAA_BASELINE_IPS")
+    if env_b:
+        vs_baseline = round(ips / float(env_b), 4)
+        baseline_src = "FAA_BASELINE_IPS"
+    else:
+        b = _R01.get((args.model, args.batch, world_size))
+        if b:
+            vs_baseline = round(ips / b, 4)
+            baseline_src = "round1 driver bench (BENCH_r01.json), same config"
     if rank == 0:
         print(json.dumps({
             "metric": "images/sec",
@@ -456,7 +525,8 @@ def main():
             "ms_per_step": round(elapsed / args.steps * 1000, 3),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": None,
+            "vs_baseline": vs_baseline,
+            "baseline_src": baseline_src,
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {"model": args.model, "global_batch": args.batch * world_size,

@@ -45,6 +45,17 @@ def _faa_wrw_wins(cin: int) -> bool:
 
 _dbias_static = {}
 
+# FAA_DBIAS_DEBUG=1: per-call, accumulate max |colsum - at::sum| into a
+# persistent device scalar (graph-safe: pure device ops). bench.py prints it
+# at exit — a direct in-graph check of whether colsum's VALUES deviate on
+# replay (tools/nan_flake.py round-4 bisect).
+_dbias_dbg = {}
+
+
+def dbias_debug_max(device) -> float:
+    t = _dbias_dbg.get(torch.device(device).index)
+    return float(t.item()) if t is not None else -1.0
+
 
 def _dbias(dy: torch.Tensor) -> torch.Tensor:
     """Bias gradient = column sum of dy over (B,H,W).
@@ -62,6 +73,19 @@ def _dbias(dy: torch.Tensor) -> torch.Tensor:
     C = dy.size(1)
     if C % 8 != 0:
         return dy.sum(dim=(0, 2, 3))
+    only_c = _os.environ.get("FAA_DBIAS_ONLY_C")
+    if only_c is not None and C != int(only_c):
+        return dy.sum(dim=(0, 2, 3))
+    if _os.environ.get("FAA_DBIAS_DEBUG") == "1":
+        di = dy.device.index
+        if di not in _dbias_dbg:
+            if torch.cuda.is_current_stream_capturing():
+                raise RuntimeError("FAA_DBIAS_DEBUG needs an eager warmup pass")
+            _dbias_dbg[di] = torch.zeros((), device=dy.device)
+        cs = ext().colsum_bf16(dy).float()
+        ts = dy.sum(dim=(0, 2, 3)).float()
+        d = _dbias_dbg[di]
+        torch.maximum(d, (cs - ts).abs().max(), out=d)
     if mode == "colsum":
         return ext().colsum_bf16(dy)
     if mode == "legacy":

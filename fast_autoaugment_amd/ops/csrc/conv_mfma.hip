@@ -861,7 +861,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
       int64_t blocks64 = (g.Wd == 32)
           ? (int64_t)g.B * tiles_h4 * gn64
           : (int64_t)(g.B / ib) * tiles_h * gn64;
-      bool big = (g.Cin > 64) && (blocks64 >= 512);
+      bool big = (g.Cin > 64) && (blocks64 >= (g.Wd == 8 ? 192 : 512));
       if (de && de[0] == 'b') big = true;
       if (de && de[0] == 's') big = false;
       #define CT_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, HB_, GRID_, THTILES_, GN_)   \
@@ -892,10 +892,11 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
           if (d8) {
             ibb = (d8[2] == '4') ? 4 : 2;
             sk = (d8[3] == 's') ? (d8[5] - '0') : 1;
-          } else if (g.B % 4 == 0) {
-            ibb = 4;
-            int64_t blk = (int64_t)(g.B / 4) * gn64;
-            sk = (blk < 512 && g.Cin >= 256) ? (blk < 256 ? 4 : 2) : 1;
+          } else {
+            // measured (gpurun_out/call4.log D8 sweep): plain ib2 wins to
+            // C=384 (35us vs 75 MIOpen at 256), ib4 wins at C>=512
+            // (150 vs 159 at 640); every split-K variant loses.
+            ibb = (g.Cin >= 512) ? 4 : 2;
           }
           if (ibb == 4 && g.B % 4 != 0) ibb = 2;
           if (sk > 1) {
@@ -1098,10 +1099,13 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
 
   auto f32 = xc.options().dtype(torch::kFloat32);
 
-  // wrw v3 direct tiled path (stride-1 3x3/1x1 CIFAR geometries)
+  // wrw v3 direct tiled path (stride-1 3x3/1x1 CIFAR geometries).
+  // Default OFF pending tuning: call4 measured the v2 split-K kernel (and
+  // MIOpen) ahead at small C — v3's epilogue atomics dominate when the
+  // (ci,co) grid is tiny and every block covers one pixel tile.
   {
     const char* e = getenv("FAA_WRW_V3");
-    bool want = !(e && e[0] == '0');
+    bool want = (e && e[0] == '1');
     bool k3 = (KH == 3 && KW == 3 && pad == 1);
     bool k1 = (KH == 1 && KW == 1 && pad == 0);
     bool geom_ok = stride == 1 && (k3 || k1)
@@ -1114,7 +1118,12 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
       int TH = (g.Wd == 32) ? 4 : 8;
       int tiles_total = g.B * (g.H / TH);
       int blocks_xy = ci_tiles * co_tiles;
-      int slices = std::max(1, std::min(768 / blocks_xy, tiles_total));
+      // amortize the per-block epilogue (64x64xKK^2 atomics) over >=2 tiles
+      int slices = std::max(1, std::min(512 / blocks_xy, tiles_total / 2));
+      if (const char* se = getenv("FAA_WRW3_SLICES")) {
+        int f = atoi(se);
+        if (f > 0) slices = std::min(f, tiles_total);
+      }
       int tps = (tiles_total + slices - 1) / slices;
       slices = (tiles_total + tps - 1) / tps;
       auto dwacc = torch::zeros({(int64_t)KK * KK, g.Cin, (int64_t)Cout}, f32);
