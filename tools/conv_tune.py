@@ -30,8 +30,40 @@ def main():
     ap.add_argument("--big", action="store_true")
     ap.add_argument("--d8", action="store_true",
                     help="sweep FAA_CONV_D8 variants on the deep 8x8 shapes")
+    ap.add_argument("--wrw", action="store_true",
+                    help="sweep wrw v3 slice counts vs v2 vs MIOpen")
     ap.add_argument("--iters", type=int, default=50)
     args = ap.parse_args()
+    if args.wrw:
+        from fast_autoaugment_amd.ops import ext as _ext
+        C = _ext()
+        shapes_w = [(128, 16, 32, 32, 3, 1), (128, 32, 32, 32, 3, 1),
+                    (128, 64, 16, 64, 3, 1), (128, 128, 8, 128, 3, 1),
+                    (128, 160, 32, 160, 3, 1), (128, 320, 16, 320, 3, 1),
+                    (128, 640, 8, 640, 3, 1)]
+        slc = ["8", "16", "32", "64", "128", "256"]
+        print(f"{'shape':<24} {'miopen':>8} {'v2':>8} {'v3auto':>8} "
+              + " ".join(f"s{v:>7}" for v in slc))
+        for B, Cin, H, Cout, k, s in shapes_w:
+            x = (torch.randn(B, Cin, H, H, device="cuda") * 0.5).bfloat16() \
+                .contiguous(memory_format=torch.channels_last)
+            dy = (torch.randn(B, Cout, H, H, device="cuda") * 0.1).bfloat16() \
+                .contiguous(memory_format=torch.channels_last)
+            w_shape = [Cout, Cin, k, k]
+            row = [f"{B}x{Cin}x{H}^2->{Cout}"]
+            row.append(f"{bench(lambda: torch.nn.grad.conv2d_weight(x, w_shape, dy, stride=s, padding=k//2), args.iters):8.1f}")
+            os.environ.pop("FAA_WRW_V3", None)
+            row.append(f"{bench(lambda: C.conv2d_bwd_weight(dy, x, s, k//2, k, k, False), args.iters):8.1f}")
+            os.environ["FAA_WRW_V3"] = "1"
+            os.environ.pop("FAA_WRW3_SLICES", None)
+            row.append(f"{bench(lambda: C.conv2d_bwd_weight(dy, x, s, k//2, k, k, False), args.iters):8.1f}")
+            for v in slc:
+                os.environ["FAA_WRW3_SLICES"] = v
+                row.append(f"{bench(lambda: C.conv2d_bwd_weight(dy, x, s, k//2, k, k, False), args.iters):8.1f}")
+            os.environ.pop("FAA_WRW3_SLICES", None)
+            os.environ.pop("FAA_WRW_V3", None)
+            print(" ".join(row))
+        return
     if args.d8:
         from fast_autoaugment_amd.ops import ext as _ext
         C = _ext()
