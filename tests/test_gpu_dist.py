@@ -1,0 +1,50 @@
+"""Real-RCCL distributed-path tests on one GPU (VERDICT r1 item 4 / gap 9).
+
+world_size=1 over the actual nccl(=RCCL) backend: init_process_group,
+flat-param broadcast, and the captured/eager all-reduce paths execute for
+real (collective semantics at ws=1 are identity, but the full RCCL
+enqueue/graph-capture machinery runs). The 8-GPU scaling run itself is the
+driver's; these prove the code path on the hardware we can reach.
+"""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _run_bench(extra_env, steps=5):
+    env = dict(os.environ,
+               FAA_BENCH_FORCE_DIST="1",
+               FAA_BENCH_IMGS="2048",
+               FAA_BENCH_SAVE="/tmp/dist_p.pt",
+               MASTER_ADDR="127.0.0.1",
+               MASTER_PORT="29617",
+               RANK="0", WORLD_SIZE="1", LOCAL_RANK="0",
+               **extra_env)
+    r = subprocess.run([sys.executable, "bench.py", "--steps", str(steps),
+                        "--warmup", "3"],
+                       capture_output=True, text=True, timeout=280, env=env)
+    assert r.returncode == 0, r.stderr[-1500:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    params = torch.load("/tmp/dist_p.pt").float()
+    assert torch.isfinite(params).all(), "non-finite params after RCCL steps"
+    return out
+
+
+def test_rccl_overlap_graph_step():
+    """FlatDDP bucketed all-reduce CAPTURED inside the step graph (comm
+    stream hooks), replayed over real RCCL at ws=1."""
+    out = _run_bench({"FAA_BENCH_DIST_MODE": "overlap"})
+    assert out["n_gpus"] == 1 and out["ms_per_step"] > 0
+
+
+def test_rccl_eager_allreduce_step():
+    """round-1 style: graph holds fwd+bwd, eager flat-grad all-reduce."""
+    out = _run_bench({"FAA_BENCH_DIST_MODE": "eager"})
+    assert out["n_gpus"] == 1 and out["ms_per_step"] > 0
