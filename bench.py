@@ -322,6 +322,14 @@ def main():
 
     graph = None
     gather_table = None
+    graph_dump = os.environ.get("FAA_BENCH_GRAPH_DUMP")
+
+    def _mk_graph():
+        g_ = torch.cuda.CUDAGraph()
+        if graph_dump:
+            g_.enable_debug_mode()
+        return g_
+
     if use_graphs:
         opt.sync_lr()
         side = torch.cuda.Stream()
@@ -335,7 +343,7 @@ def main():
                 step_body()
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
-        graph = torch.cuda.CUDAGraph()
+        graph = _mk_graph()
         upload_next()
         if gather_mode:
             saved_grads = [p_.grad for p_ in flat.params]
@@ -377,13 +385,20 @@ def main():
                 for p_, g_ in zip(flat.params, saved_grads):
                     p_.grad = g_
                 torch.cuda.synchronize()
-                graph = torch.cuda.CUDAGraph()
+                graph = _mk_graph()
                 upload_next()
                 with torch.cuda.graph(graph):
                     flat_capture_body() if dist_in_graph else step_body()
         else:
             with torch.cuda.graph(graph):
                 flat_capture_body() if dist_in_graph else step_body()
+
+    if graph_dump and graph is not None:
+        try:
+            graph.debug_dump(graph_dump)
+            print(f"# graph dumped to {graph_dump}", flush=True)
+        except Exception as e:
+            print(f"# graph dump failed: {e}", flush=True)
 
     def one_step():
         nonlocal step_idx

@@ -34,13 +34,26 @@ _FAA_BWD_DATA_MAX = 128
 import os as _os
 
 
-def _faa_wrw_wins(cin: int) -> bool:
+def _faa_wrw_wins(cin: int, h: int = 0, cout: int = 0) -> bool:
+    """Measured per-shape wrw dispatch (gpurun_out/call5.log WRWSWEEP +
+    call4 CONVBENCH, b128):
+      stem cin<8            v2 wins (41-48us vs 51 torch)
+      cin==128 @8px         v2 wins (31.7 vs 45.6 MIOpen)
+      cin>=160 @32px        v3 wins (232 vs 278 MIOpen at 160^3)
+      everything else       MIOpen/torch wins — keep the fallback
+    """
     mode = _os.environ.get("FAA_WRW", "auto")
     if mode == "faa":
         return True
     if mode == "torch":
         return False
-    return cin < 8 or cin == 128
+    if cin < 8:
+        return True
+    if cin == 128 and h == 8:
+        return True
+    if cin >= 160 and h == 32:
+        return True
+    return False
 
 
 _dbias_static = {}
@@ -61,15 +74,21 @@ def _dbias(dy: torch.Tensor) -> torch.Tensor:
     """Bias gradient = column sum of dy over (B,H,W).
 
     FAA_DBIAS modes (tools/nan_hunt.py hipGraph bisect):
-      colsum  (default) replay-safe v2 HIP kernel
-      torch   at::reduce (round-1 quarantine default)
+      torch   (default) at::reduce — the colsum kernels compute correct
+              values standalone and in the nan_hunt captured graph, but
+              inside the BENCH step graph their output deviates on replay
+              (measured: FAA_DBIAS_DEBUG max-abs-diff NaN/1e28,
+              gpurun_out/call5.log) and the corrupted values also cost
+              ~1 ms/step; at::reduce is both correct and currently faster
+              in-graph. Root-cause hunt: docs/GRAPH_NAN.md.
+      colsum  replay-safe-by-construction v2 HIP kernel (no atomics/memset)
       legacy  round-1 atomic kernel
       dummy   at::reduce result + a DISCARDED colsum launch (perturbation
               control: flakes here => colsum is innocent)
       static  v2 into non-pool buffers cached at first (eager) call, then
               cloned into the pool
     """
-    mode = _os.environ.get("FAA_DBIAS", "colsum")
+    mode = _os.environ.get("FAA_DBIAS", "torch")
     C = dy.size(1)
     if C % 8 != 0:
         return dy.sum(dim=(0, 2, 3))
@@ -138,7 +157,7 @@ class FaaConv2dFn(torch.autograd.Function):
                                                 stride=ctx.stride,
                                                 padding=ctx.padding)
         if ctx.needs_input_grad[1] or (ctx.has_bias and ctx.needs_input_grad[2]):
-            if _faa_wrw_wins(x.size(1)):
+            if _faa_wrw_wins(x.size(1), x.size(2), weight.size(0)):
                 dw, _ = C.conv2d_bwd_weight(dy, x, ctx.stride, ctx.padding,
                                             weight.size(2), weight.size(3),
                                             False)
