@@ -332,16 +332,23 @@ def main():
 
     if use_graphs:
         opt.sync_lr()
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
+        # FAA_BENCH_WARMUP_STREAM=default: warm up on the default stream
+        # instead of a side stream (c128 colsum corruption bisect axis)
+        if os.environ.get("FAA_BENCH_WARMUP_STREAM") == "default":
             for _ in range(3):
                 upload_next()
-                # warmup always runs the FULL step (flat grad views are still
-                # attached in gather mode) so both grad modes reach capture
-                # with identical param/momentum state
                 step_body()
-        torch.cuda.current_stream().wait_stream(side)
+        else:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    upload_next()
+                    # warmup always runs the FULL step (flat grad views are
+                    # still attached in gather mode) so both grad modes reach
+                    # capture with identical param/momentum state
+                    step_body()
+            torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         graph = _mk_graph()
         upload_next()

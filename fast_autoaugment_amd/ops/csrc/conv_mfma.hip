@@ -537,7 +537,11 @@ constexpr int wrw3_pitch(int n) {
   }
 }
 
-template <int TH, int TW, int KK>
+// WSPLIT=4: small-C variant (Cin,Cout <= 32). The 2x2 wave quadrants of
+// the 64x64 tile would leave 3 of 4 waves entirely out of range, so all
+// waves map to quadrant (0,0) and SPLIT the pixel chunks instead; their
+// duplicate-position partial sums merge through the same epilogue atomics.
+template <int TH, int TW, int KK, int WSPLIT = 1>
 __global__ __launch_bounds__(256)
 void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
                       float* __restrict__ dWacc, ConvGeom g,
@@ -568,8 +572,8 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wm = wave >> 1;          // ci half
-  const int wn = wave & 1;           // co half
+  const int wm = (WSPLIT == 1) ? (wave >> 1) : 0;   // ci half
+  const int wn = (WSPLIT == 1) ? (wave & 1) : 0;    // co half
   const int fr = lane & 15;
   const int kq = (lane >> 4) * 8;
 
@@ -626,7 +630,8 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
     __syncthreads();
 
     #pragma unroll
-    for (int pc = 0; pc < PIX_OUT / 32; ++pc) {
+    for (int pc = (WSPLIT == 1 ? 0 : wave); pc < PIX_OUT / 32;
+         pc += (WSPLIT == 1 ? 1 : 4)) {
       const int p8 = pc * 32 + kq;       // this lane's first pixel
       const int py = p8 / TW, px = p8 % TW;
       bf16x8 bfrag[2];
@@ -1130,20 +1135,27 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
       slices = (tiles_total + tps - 1) / tps;
       auto dwacc = torch::zeros({(int64_t)KK * KK, g.Cin, (int64_t)Cout}, f32);
       dim3 grid(blocks_xy, slices);
-      #define WRW3_LAUNCH(TH_, TW_, KK_)                                        \
-        hipLaunchKernelGGL((conv_wrw3_kernel<TH_, TW_, KK_>), grid, dim3(256),  \
-                           0, stream, (const short*)xc.data_ptr(),              \
+      bool small = (g.Cin <= 32 && Cout <= 32);
+      #define WRW3_LAUNCH(TH_, TW_, KK_, WS_)                                   \
+        hipLaunchKernelGGL((conv_wrw3_kernel<TH_, TW_, KK_, WS_>), grid,        \
+                           dim3(256), 0, stream, (const short*)xc.data_ptr(),   \
                            (const short*)dyc.data_ptr(),                        \
                            dwacc.data_ptr<float>(), g, ci_tiles, co_tiles, tps)
+      #define WRW3_PICK(TH_, TW_, KK_)                                          \
+        do {                                                                    \
+          if (small) WRW3_LAUNCH(TH_, TW_, KK_, 4);                             \
+          else WRW3_LAUNCH(TH_, TW_, KK_, 1);                                   \
+        } while (0)
       if (k3) {
-        if (g.Wd == 32) WRW3_LAUNCH(4, 32, 3);
-        else if (g.Wd == 16) WRW3_LAUNCH(8, 16, 3);
-        else WRW3_LAUNCH(8, 8, 3);
+        if (g.Wd == 32) WRW3_PICK(4, 32, 3);
+        else if (g.Wd == 16) WRW3_PICK(8, 16, 3);
+        else WRW3_PICK(8, 8, 3);
       } else {
-        if (g.Wd == 32) WRW3_LAUNCH(4, 32, 1);
-        else if (g.Wd == 16) WRW3_LAUNCH(8, 16, 1);
-        else WRW3_LAUNCH(8, 8, 1);
+        if (g.Wd == 32) WRW3_PICK(4, 32, 1);
+        else if (g.Wd == 16) WRW3_PICK(8, 16, 1);
+        else WRW3_PICK(8, 8, 1);
       }
+      #undef WRW3_PICK
       #undef WRW3_LAUNCH
       auto dw = torch::empty({Cout, g.Cin, (int64_t)KH, (int64_t)KW},
                              xc.options().memory_format(torch::MemoryFormat::ChannelsLast));

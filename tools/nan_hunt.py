@@ -42,7 +42,8 @@ def build(model_name="wresnet40_2"):
     return model, flat, opt, crit
 
 
-def run_variant(mode, replays, eager_tail, batch=128, seed=0):
+def run_variant(mode, replays, eager_tail, batch=128, seed=0,
+                side_warmup=False):
     os.environ["FAA_DBIAS"] = mode
     model, flat, opt, crit = build()
     torch.manual_seed(1234 + seed)
@@ -55,11 +56,22 @@ def run_variant(mode, replays, eager_tail, batch=128, seed=0):
         loss.backward()
         return loss
 
-    # eager warmup (3 full steps, also instantiates any static buffers)
-    for _ in range(3):
-        opt.zero_grad()
-        fwd_bwd()
-        opt.step()
+    # eager warmup (3 full steps, also instantiates any static buffers);
+    # --side-warmup mirrors bench.py's side-stream warmup (the last
+    # structural difference vs the bench graph in the c128 bisect)
+    def warm():
+        for _ in range(3):
+            opt.zero_grad()
+            fwd_bwd()
+            opt.step()
+    if side_warmup:
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            warm()
+        torch.cuda.current_stream().wait_stream(side)
+    else:
+        warm()
     torch.cuda.synchronize()
 
     # gather-mode capture: assignment-mode backward
@@ -111,14 +123,16 @@ def main():
     ap.add_argument("--variants", type=str,
                     default="colsum,legacy,torch,dummy,static")
     ap.add_argument("--no-eager-tail", action="store_true")
+    ap.add_argument("--side-warmup", action="store_true")
     args = ap.parse_args()
     for mode in args.variants.split(","):
         for rep in range(args.repeats):
             bad_at, bad_param = run_variant(mode, args.replays,
-                                            not args.no_eager_tail, seed=rep)
+                                            not args.no_eager_tail, seed=rep,
+                                            side_warmup=args.side_warmup)
             status = "CLEAN" if bad_at < 0 else f"NAN@replay{bad_at} param={bad_param}"
-            print(f"{mode:>8} rep{rep} tail={not args.no_eager_tail}: {status}",
-                  flush=True)
+            print(f"{mode:>8} rep{rep} tail={not args.no_eager_tail} "
+                  f"sidewarm={args.side_warmup}: {status}", flush=True)
 
 
 if __name__ == "__main__":
