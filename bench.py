@@ -152,10 +152,13 @@ def main():
             import torch.distributed as dist
             dist.broadcast(flat.flat_param, 0)   # rank-0 weight sync
     if work_dtype == torch.bfloat16:
-        from fast_autoaugment_amd.ops.conv import patch_convs
+        from fast_autoaugment_amd.ops.conv import conv_flip_all, patch_convs
         n_patched = patch_convs(model)
+        os.environ.setdefault("FAA_FLIP_BATCH", "1")
         if rank == 0:
             print(f"# {n_patched} convs on MFMA kernels", flush=True)
+    else:
+        conv_flip_all = None
     lr0 = conf["lr"] * world_size
     opt = FusedSGD(flat, lr=lr0, momentum=0.9, nesterov=True,
                    weight_decay=conf["optimizer"]["decay"], grad_clip=5.0)
@@ -267,6 +270,8 @@ def main():
 
     def gpu_fwd_bwd():
         """aug + forward + loss + backward on static inputs (capturable)."""
+        if conv_flip_all is not None and os.environ.get("FAA_FLIP_BATCH") == "1":
+            conv_flip_all()
         if fixed_data:
             if "data" not in fixed:
                 fixed["data"] = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,

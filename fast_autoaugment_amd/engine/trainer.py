@@ -63,8 +63,13 @@ def run_epoch(model, loader, loss_fn, optimizer, desc_default="", epoch=0,
     # per-step scalars accumulate on-device; ONE host sync per epoch
     acc = torch.zeros(3, device=device)
 
+    flip_batch = (optimizer is not None and device != "cpu"
+                  and os.environ.get("FAA_FLIP_BATCH") == "1")
     for data, label in loader:
         steps += 1
+        if flip_batch:
+            from ..ops.conv import conv_flip_all
+            conv_flip_all()   # one launch refreshes all bwd-data repacks
         data = data.to(device, non_blocking=True)
         label = label.to(device, non_blocking=True)
         if device != "cpu":

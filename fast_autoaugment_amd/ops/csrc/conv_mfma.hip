@@ -420,6 +420,33 @@ __global__ void weight_flip_kernel(const short* __restrict__ W, short* __restric
   W2[(((int64_t)ci * KH + (KH - 1 - kh)) * KW + (KW - 1 - kw)) * Cout + co] = W[i];
 }
 
+// batched weight flip: one launch refreshes EVERY conv's bwd-data repack
+// (vs 31 x 4.8us weight_flip launches per step in the round-1 profile).
+// table rows: [src_ptr, dst_ptr, Cout, KH, KW, Cin, elem_offset]
+__global__ void flip_weights_batched_kernel(const int64_t* __restrict__ table,
+                                            int nrows, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    // binary search the row whose [offset, offset+numel) contains i
+    int lo = 0, hi = nrows - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (table[mid * 7 + 6] <= i) lo = mid; else hi = mid - 1;
+    }
+    const int64_t* r = table + lo * 7;
+    const short* W = (const short*)r[0];
+    short* W2 = (short*)r[1];
+    int Cout = (int)r[2], KH = (int)r[3], KW = (int)r[4], Cin = (int)r[5];
+    int64_t j = i - r[6];
+    int ci = j % Cin;
+    int64_t t = j / Cin;
+    int kw = t % KW; t /= KW;
+    int kh = t % KH; t /= KH;
+    int co = (int)t;
+    W2[(((int64_t)ci * KH + (KH - 1 - kh)) * KW + (KW - 1 - kw)) * Cout + co] = W[j];
+  }
+}
+
 // ------------------------------------------------------------- bwd-weight
 // dW_acc[k=(kh,kw,ci)][n=co] += sum_m A[m][k] * dY[m][n]  (fp32 atomics)
 // block tile: 32 k-rows x 64 co, m-chunks of 32, split over blockIdx.z
@@ -1064,6 +1091,15 @@ void colsum_bf16_ws(torch::Tensor dy, torch::Tensor part, torch::Tensor out) {
                      (const short*)dyc.data_ptr(), part.data_ptr<float>(), M, C);
   hipLaunchKernelGGL(colsum_finish_kernel, dim3((C + 255) / 256), dim3(256), 0,
                      stream, part.data_ptr<float>(), (short*)out.data_ptr(), nb, C);
+}
+
+void flip_weights_batched(torch::Tensor table, int64_t total) {
+  TORCH_CHECK(table.scalar_type() == torch::kInt64 && table.size(1) == 7,
+              "flip table must be int64 [n,7]");
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(flip_weights_batched_kernel, dim3(grid), dim3(256), 0,
+                     stream, table.data_ptr<int64_t>(), (int)table.size(0), total);
 }
 
 torch::Tensor colsum_bf16_legacy(torch::Tensor dy) {

@@ -48,6 +48,7 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
 torch::Tensor colsum_bf16(torch::Tensor dy);
 torch::Tensor colsum_bf16_legacy(torch::Tensor dy);
 void colsum_bf16_ws(torch::Tensor dy, torch::Tensor part, torch::Tensor out);
+void flip_weights_batched(torch::Tensor table, int64_t total);
 torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                          int64_t stride, int64_t pt, int64_t pb, int64_t pl, int64_t pr);
 torch::Tensor dwconv_bwd_data(torch::Tensor dy, torch::Tensor w, int64_t stride,
@@ -86,6 +87,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "round-1 atomic colsum, kept for the hipGraph corruption bisect only");
   m.def("colsum_bf16_ws", &colsum_bf16_ws,
         "colsum v2 into caller-provided workspace/out (graph bisect)");
+  m.def("flip_weights_batched", &flip_weights_batched,
+        "one-launch bwd-data weight repack for every registered conv");
   m.def("dwconv_fwd", &dwconv_fwd, "depthwise conv forward (NHWC bf16)");
   m.def("dwconv_bwd_data", &dwconv_bwd_data);
   m.def("dwconv_bwd_weight", &dwconv_bwd_weight);
