@@ -115,7 +115,8 @@ def main():
     store = TensorStore(imgs, labels, device=str(dev))
     mean, std = dataset_stats(args.dataset)
     out_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
-    loader = AugLoader(store, args.batch, resolve_aug(aug_name),
+    no_loader = os.environ.get("FAA_BENCH_NO_LOADER") == "1"
+    loader = None if no_loader else AugLoader(store, args.batch, resolve_aug(aug_name),
                        train=True, mean=mean, std=std, cutout=conf["cutout"],
                        seed=rank, out_dtype=out_dtype, prefetch=4,
                        imagenet_size=out_size if imagenet else 0)
@@ -167,7 +168,7 @@ def main():
     amp_dtype = None   # bf16 runs natively through the flat bf16 weights
 
     model.train()
-    steps_per_epoch = max(len(loader), 1)
+    steps_per_epoch = max(len(loader), 1) if loader is not None else 390
     step_idx = 0
 
     from fast_autoaugment_amd.aug import ops as aug_ops
@@ -524,8 +525,9 @@ def main():
     if os.environ.get("FAA_DBIAS_DEBUG") == "1" and not cpu_mode:
         from fast_autoaugment_amd.ops.conv import dbias_debug_max
         torch.cuda.synchronize()
-        print(f"# dbias colsum-vs-sum max abs diff: {dbias_debug_max(dev):.6f}",
-              flush=True)
+        dvi, dpo = dbias_debug_max(dev)
+        print(f"# dbias colsum-vs-sum max abs diff: {dvi:.6f} "
+              f"sum-vs-sum(positional): {dpo:.6f}", flush=True)
 
     save_p = os.environ.get("FAA_BENCH_SAVE")
     if save_p and rank == 0:

@@ -65,9 +65,12 @@ _dbias_static = {}
 _dbias_dbg = {}
 
 
-def dbias_debug_max(device) -> float:
-    t = _dbias_dbg.get(torch.device(device).index)
-    return float(t.item()) if t is not None else -1.0
+def dbias_debug_max(device):
+    di = torch.device(device).index
+    t = _dbias_dbg.get(di)
+    p = _dbias_dbg.get((di, "pos"))
+    return (float(t.item()) if t is not None else -1.0,
+            float(p.item()) if p is not None else -1.0)
 
 
 def _dbias(dy: torch.Tensor) -> torch.Tensor:
@@ -103,8 +106,17 @@ def _dbias(dy: torch.Tensor) -> torch.Tensor:
             _dbias_dbg[di] = torch.zeros((), device=dy.device)
         cs = ext().colsum_bf16(dy).float()
         ts = dy.sum(dim=(0, 2, 3)).float()
+        ts2 = dy.sum(dim=(0, 2, 3)).float()
         d = _dbias_dbg[di]
+        # d[()] tracks colsum-vs-sum; the positional probe (sum-vs-sum of
+        # the SAME dy) goes to a second scalar: if THAT explodes, dy itself
+        # is being trashed between adjacent reads at replay
         torch.maximum(d, (cs - ts).abs().max(), out=d)
+        d2 = _dbias_dbg.get((di, "pos"))
+        if d2 is None:
+            d2 = torch.zeros((), device=dy.device)
+            _dbias_dbg[(di, "pos")] = d2
+        torch.maximum(d2, (ts - ts2).abs().max(), out=d2)
     if mode == "colsum":
         return ext().colsum_bf16(dy)
     if mode == "legacy":
@@ -112,6 +124,18 @@ def _dbias(dy: torch.Tensor) -> torch.Tensor:
     if mode == "dummy":
         ext().colsum_bf16(dy)
         return dy.sum(dim=(0, 2, 3))
+    if mode == "colsum_pad":
+        # layout probe: same kernels, pool workspace padded by FAA_DBIAS_PAD
+        # elements — if the NaN rate moves with pure padding, a
+        # layout-dependent stale-pointer scribble is confirmed
+        import math
+        q = C // math.gcd(C, 2048)
+        nb = -(-64 // q) * q
+        pad = int(_os.environ.get("FAA_DBIAS_PAD", "4096"))
+        part = torch.empty(nb * C + pad, dtype=torch.float32, device=dy.device)
+        out = torch.empty(C, dtype=torch.bfloat16, device=dy.device)
+        ext().colsum_bf16_ws(dy, part, out)
+        return out
     if mode == "static":
         import math
         key = (C, dy.device.index)
@@ -154,6 +178,10 @@ class FaaConv2dFn(torch.autograd.Function):
             faa_ok = (ctx.stride == 1 and ch <= 640
                       and not (ch >= 512 and x.size(2) <= 8)
                       and _os.environ.get("FAA_BWD_DATA", "faa") != "torch")
+            s2_ok = (ctx.stride == 2 and weight.size(2) == 3 and ctx.padding == 1
+                     and x.size(3) in (16, 32) and x.size(2) % 8 == 0
+                     and x.size(2) == 2 * dy.size(2) and x.size(3) == 2 * dy.size(3)
+                     and _os.environ.get("FAA_BWD_DATA", "faa") != "torch")
             if faa_ok:
                 w2 = (_flip_cache.get(weight.data_ptr())
                       if _os.environ.get("FAA_FLIP_BATCH") == "1" else None)
@@ -164,6 +192,14 @@ class FaaConv2dFn(torch.autograd.Function):
                 else:
                     dx = C.conv2d_bwd_data(dy, weight, 1, ctx.padding,
                                            x.size(2), x.size(3))
+            elif s2_ok:
+                # stride-2 3x3: phase-decomposition kernel (4 dense sub-convs)
+                w2 = (_flip_cache.get(weight.data_ptr())
+                      if _os.environ.get("FAA_FLIP_BATCH") == "1" else None)
+                if w2 is not None:
+                    dx = C.conv2d_bwd_data_s2(dy, w2, x.size(2), x.size(3), True)
+                else:
+                    dx = C.conv2d_bwd_data_s2(dy, weight, x.size(2), x.size(3), False)
             else:
                 dx = torch.nn.grad.conv2d_input(list(x.shape), weight, dy,
                                                 stride=ctx.stride,
@@ -232,7 +268,11 @@ _flip_table = [None, 0]
 
 
 def _register_flip(m: torch.nn.Conv2d) -> None:
-    if m.stride[0] != 1 or m.kernel_size[0] not in (1, 3):
+    # s1 convs use the flipped weights for bwd-data-as-fwd; s2 3x3 convs for
+    # the phase-decomposition kernel. 1x1 s2 has no in-house bwd-data.
+    if m.kernel_size[0] not in (1, 3):
+        return
+    if m.stride[0] == 2 and m.kernel_size[0] != 3:
         return
     w = m.weight
     if not w.is_cuda or w.dtype != torch.bfloat16:

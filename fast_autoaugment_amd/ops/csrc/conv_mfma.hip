@@ -420,6 +420,155 @@ __global__ void weight_flip_kernel(const short* __restrict__ W, short* __restric
   W2[(((int64_t)ci * KH + (KH - 1 - kh)) * KW + (KW - 1 - kw)) * Cout + co] = W[i];
 }
 
+// ------------------------------------------ stride-2 bwd-data (k3 s2 p1)
+// Phase decomposition (no dilated-zero waste): dx[y,x] with output parity
+// (py,px) = (y&1, x&1) is a DENSE stride-1 correlation of dy with the
+// sub-filter taps {kh : kh === (y+1) mod 2} x {kw : ...}: even coords use
+// the single kh=1 tap, odd coords the {0,2} pair. The 4 phases map 1:1
+// onto the 4 waves of a block (each wave's m-fragments stay phase-uniform
+// so MFMA K is uniform per fragment). dy tile staged once per cin-slab in
+// LDS; flipped weights W2[ci][fh][fw][co] (fh = 2-kh, the conv_flip_all
+// layout) read to registers.
+//   geom g here: Cin = dy channels (orig Cout), Cout = dx channels,
+//   H/Wd = dx dims, Ho/Wo = dy dims.
+template <int TH, int TW>
+__global__ __launch_bounds__(256)
+void conv3x3s2_dx_kernel(const short* __restrict__ dY, const short* __restrict__ W2,
+                         short* __restrict__ dX, ConvGeom g,
+                         int tiles_h, int grid_n) {
+  constexpr int CS = 32;
+  constexpr int PS = 40;
+  constexpr int XR = TH / 2 + 1;       // dy rows per tile (+1 forward halo)
+  constexpr int XC = TW / 2 + 1;
+  constexpr int MT = TH * TW;          // dx pixels per block
+  constexpr int MR = MT / 64;          // fragments per wave (wave == phase)
+  static_assert(MT % 64 == 0, "tile must give whole fragments per phase");
+
+  __shared__ short ldsY[XR * XC * PS];
+
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, idx = wg / 8;
+    if (q > 0) wg = (xcd < r) ? (xcd * (q + 1) + idx) : (r * (q + 1) + (xcd - r) * q + idx);
+  }
+  const int nt = wg % grid_n;
+  int t = wg / grid_n;
+  const int ty = t % tiles_h;
+  const int b = t / tiles_h;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;           // == phase index (py*2+px)
+  const int py = wave >> 1;
+  const int px = wave & 1;
+  const int fr = lane & 15;
+  const int kq = (lane >> 4) * 8;
+  const int n0 = nt * 32;
+
+  // per-fragment LDS base: lane's dx pixel inside the phase quarter
+  int aoff[MR];
+  #pragma unroll
+  for (int mf = 0; mf < MR; ++mf) {
+    int rest = mf * 16 + fr;
+    int ry = rest / (TW / 2);
+    int rx = rest % (TW / 2);
+    aoff[mf] = (ry * XC + rx) * PS + kq;
+  }
+
+  // taps for this wave's phase: doy/dox are the +0/+1 dy offsets and
+  // (fh,fw) index the FLIPPED weight layout (fh = 2-kh)
+  int tap_doy[2], tap_fh[2], nty;
+  if (py == 0) { nty = 1; tap_doy[0] = 0; tap_fh[0] = 1; }
+  else { nty = 2; tap_doy[0] = 1; tap_fh[0] = 2; tap_doy[1] = 0; tap_fh[1] = 0; }
+  int tap_dox[2], tap_fw[2], ntx;
+  if (px == 0) { ntx = 1; tap_dox[0] = 0; tap_fw[0] = 1; }
+  else { ntx = 2; tap_dox[0] = 1; tap_fw[0] = 2; tap_dox[1] = 0; tap_fw[1] = 0; }
+
+  f32x4 acc[MR][2] = {};
+  const int oy0 = (ty * TH) >> 1;      // first dy row of this tile
+
+  for (int cs = 0; cs < g.Cin; cs += CS) {
+    // stage the dy tile slab (zero-filled beyond Ho/Wo and channel tail)
+    constexpr int NV = XR * XC * 4;
+    for (int v = tid; v < NV; v += 256) {
+      int kc = v & 3;
+      int cell = v >> 2;
+      int col = cell % XC;
+      int row = cell / XC;
+      int oy = oy0 + row;
+      int ci0 = cs + kc * 8;
+      bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (oy < g.Ho && col < g.Wo && ci0 < g.Cin) {
+        const short* p = dY + (((int64_t)b * g.Ho + oy) * g.Wo + col) * g.Cin + ci0;
+        if (ci0 + 8 <= g.Cin) {
+          val = *reinterpret_cast<const bf16x8*>(p);
+        } else {
+          for (int j = 0; j < g.Cin - ci0; ++j) val[j] = p[j];
+        }
+      }
+      *reinterpret_cast<bf16x8*>(&ldsY[cell * PS + kc * 8]) = val;
+    }
+    // flipped-weight fragments for this phase's taps
+    bf16x8 wreg[4][2];
+    for (int tyi = 0; tyi < nty; ++tyi) {
+      for (int txi = 0; txi < ntx; ++txi) {
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          int n = n0 + nf * 16 + fr;   // dx channel
+          int ci0 = cs + kq;
+          bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+          if (n < g.Cout && ci0 < g.Cin) {
+            const short* p = W2 + (((int64_t)n * 3 + tap_fh[tyi]) * 3
+                                   + tap_fw[txi]) * g.Cin + ci0;
+            if (ci0 + 8 <= g.Cin) {
+              v = *reinterpret_cast<const bf16x8*>(p);
+            } else {
+              for (int j = 0; j < g.Cin - ci0; ++j) v[j] = p[j];
+            }
+          }
+          wreg[tyi * 2 + txi][nf] = v;
+        }
+      }
+    }
+    __syncthreads();
+    for (int tyi = 0; tyi < nty; ++tyi) {
+      for (int txi = 0; txi < ntx; ++txi) {
+        const int toff = (tap_doy[tyi] * XC + tap_dox[txi]) * PS;
+        #pragma unroll
+        for (int mf = 0; mf < MR; ++mf) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(&ldsY[aoff[mf] + toff]);
+          #pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, wreg[tyi * 2 + txi][nf], acc[mf][nf], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  #pragma unroll
+  for (int mf = 0; mf < MR; ++mf) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int rest = mf * 16 + (lane >> 4) * 4 + r;
+      int ry = rest / (TW / 2);
+      int rx = rest % (TW / 2);
+      int y = ty * TH + 2 * ry + py;
+      int x = 2 * rx + px;
+      int64_t base = (((int64_t)b * g.H + y) * g.Wd + x) * g.Cout;
+      #pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int n = n0 + nf * 16 + fr;
+        if (n < g.Cout)
+          dX[base + n] = f2b(acc[mf][nf][r]);
+      }
+    }
+  }
+}
+
 // batched weight flip: one launch refreshes EVERY conv's bwd-data repack
 // (vs 31 x 4.8us weight_flip launches per step in the round-1 profile).
 // table rows: [src_ptr, dst_ptr, Cout, KH, KW, Cin, elem_offset]
@@ -1050,6 +1199,54 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w,
   return conv2d_fwd(dyc, w2, torch::Tensor(), 1, KH - 1 - pad);
 }
 
+torch::Tensor conv2d_bwd_data_s2(torch::Tensor dy, torch::Tensor w,
+                                  int64_t H, int64_t W, bool pre_flipped) {
+  // stride-2 3x3 pad-1 bwd-data by phase decomposition (conv3x3s2_dx_kernel).
+  // w: either the conv weight [Cout][Cin][3][3] (channels_last) or, when
+  // pre_flipped, the conv_flip_all repack [Cin][Cout][3][3] (channels_last,
+  // i.e. [ci][fh][fw][co] in memory).
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  int B = dyc.size(0), Cdy = dyc.size(1), Ho = dyc.size(2), Wo = dyc.size(3);
+  TORCH_CHECK((W == 16 || W == 32) && H % 8 == 0 && H == 2 * Ho && W == 2 * Wo,
+              "conv2d_bwd_data_s2: unsupported geometry");
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  torch::Tensor w2;
+  int Cdx;
+  if (pre_flipped) {
+    w2 = w;
+    Cdx = w.size(0);
+    TORCH_CHECK(w.size(1) == Cdy && w.is_contiguous(torch::MemoryFormat::ChannelsLast));
+  } else {
+    auto wc = w.contiguous(torch::MemoryFormat::ChannelsLast);
+    Cdx = wc.size(1);
+    TORCH_CHECK(wc.size(0) == Cdy && wc.size(2) == 3 && wc.size(3) == 3);
+    w2 = torch::empty({Cdx, Cdy, 3, 3},
+                      wc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+    int64_t total = (int64_t)Cdy * 9 * Cdx;
+    hipLaunchKernelGGL(weight_flip_kernel, dim3((total + 255) / 256), dim3(256),
+                       0, stream, (const short*)wc.data_ptr(), (short*)w2.data_ptr(),
+                       Cdy, 3, 3, Cdx);
+  }
+  ConvGeom g;
+  g.B = B; g.Cin = Cdy; g.Cout = Cdx; g.H = H; g.Wd = W;
+  g.Ho = Ho; g.Wo = Wo; g.KH = 3; g.KW = 3; g.stride = 2; g.pad = 1;
+  g.cin_chunks = (Cdy + 7) / 8; g.kpad = 0;
+  auto dx = torch::empty({B, Cdx, H, W},
+                         dyc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  int grid_n = (Cdx + 31) / 32;
+  int tiles_h = H / 8;
+  dim3 grid((unsigned)((int64_t)B * tiles_h * grid_n));
+  if (W == 32)
+    hipLaunchKernelGGL((conv3x3s2_dx_kernel<8, 32>), grid, dim3(256), 0, stream,
+                       (const short*)dyc.data_ptr(), (const short*)w2.data_ptr(),
+                       (short*)dx.data_ptr(), g, tiles_h, grid_n);
+  else
+    hipLaunchKernelGGL((conv3x3s2_dx_kernel<8, 16>), grid, dim3(256), 0, stream,
+                       (const short*)dyc.data_ptr(), (const short*)w2.data_ptr(),
+                       (short*)dx.data_ptr(), g, tiles_h, grid_n);
+  return dx;
+}
+
 torch::Tensor colsum_bf16(torch::Tensor dy) {
   // dbias[n] = sum over rows of a [*, C] bf16 channels-last tensor.
   // v2 two-kernel scheme: no atomics, no memset — every byte read was
@@ -1171,7 +1368,10 @@ std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
       slices = (tiles_total + tps - 1) / tps;
       auto dwacc = torch::zeros({(int64_t)KK * KK, g.Cin, (int64_t)Cout}, f32);
       dim3 grid(blocks_xy, slices);
-      bool small = (g.Cin <= 32 && Cout <= 32);
+      // WSPLIT=4 measured SLOWER at small C (call7: 4x epilogue atomics
+      // dominate); kept behind an env for future cross-wave-reduce work
+      const char* wse = getenv("FAA_WRW3_WSPLIT");
+      bool small = (wse && wse[0] == '1') && (g.Cin <= 32 && Cout <= 32);
       #define WRW3_LAUNCH(TH_, TW_, KK_, WS_)                                   \
         hipLaunchKernelGGL((conv_wrw3_kernel<TH_, TW_, KK_, WS_>), grid,        \
                            dim3(256), 0, stream, (const short*)xc.data_ptr(),   \

@@ -215,3 +215,39 @@ def test_conv_fwd_direct_d8_variants(d8):
         os.environ.pop("FAA_CONV_DIRECT", None)
     err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
     assert err < 2e-2, f"d8={d8} rel err {err}"
+
+
+@pytest.mark.parametrize("B,Cin,H,Cout", [(8, 32, 32, 64), (8, 64, 16, 128),
+                                          (4, 160, 32, 320)])
+def test_conv_bwd_data_stride2_phase(B, Cin, H, Cout):
+    """Phase-decomposition stride-2 3x3 bwd-data vs fp32 torch."""
+    torch.manual_seed(11)
+    x = torch.randn(B, Cin, H, H, device=dev()) * 0.5
+    w = torch.randn(Cout, Cin, 3, 3, device=dev()) * 0.05
+    dy = torch.randn(B, Cout, H // 2, H // 2, device=dev()) * 0.5
+    ref = torch.nn.grad.conv2d_input(list(x.shape), w, dy, stride=2, padding=1)
+    dyb = dy.bfloat16().contiguous(memory_format=torch.channels_last)
+    wb = w.bfloat16().contiguous(memory_format=torch.channels_last)
+    got = C.conv2d_bwd_data_s2(dyb, wb, H, H, False).float()
+    err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert err < 2e-2, f"s2 bwd-data rel err {err}"
+
+
+def test_batched_flip_matches_per_conv():
+    """conv_flip_all (one launch) == per-conv weight_flip semantics, via the
+    cached bwd-data path producing identical dx."""
+    import os
+    from fast_autoaugment_amd.ops.conv import (conv_flip_all, patch_convs,
+                                               _flip_cache)
+    torch.manual_seed(3)
+    m = torch.nn.Conv2d(32, 64, 3, padding=1).to(dev()).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    assert patch_convs(m) == 1
+    conv_flip_all()
+    dy = torch.randn(8, 64, 16, 16, device=dev()).bfloat16().contiguous(
+        memory_format=torch.channels_last)
+    ref = C.conv2d_bwd_data(dy, m.weight, 1, 1, 16, 16).float()
+    w2 = _flip_cache[m.weight.data_ptr()]
+    got = C.conv2d_fwd(dy, w2, torch.Tensor(), 1, 1).float()
+    err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert err < 1e-3, f"batched flip mismatch {err}"
