@@ -25,6 +25,7 @@ SHAPES = [
     (16, 32, 160, 3, 1), (160, 32, 160, 3, 1),
     (160, 32, 320, 3, 2), (320, 16, 320, 3, 1),
     (320, 16, 640, 3, 2), (640, 8, 640, 3, 1),
+    (3, 224, 64, 7, 2),    # ResNet ImageNet 7x7 stem
 ]
 
 
