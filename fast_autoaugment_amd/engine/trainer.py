@@ -48,6 +48,67 @@ def _apply_manual_wd(params, wd: float):
         torch._foreach_add_([p.grad for p in ps], ps, alpha=wd)
 
 
+class _GraphedTrainStep:
+    """hipGraph-captured train step for the trainer hot loop.
+
+    The eager loop issues ~1500 kernel launches per step and measured 61%
+    GPU-busy (profiles/train_profile_r02.txt); capturing
+    flip+zero+fwd+loss+bwd+opt.step as ONE replayable graph removes the
+    launch tail. Batches are copied into static input buffers each step;
+    loss/preds are read back from the capture pool after replay (stream-
+    ordered). First 3 steps run eagerly (warmup), then capture.
+    Falls back (returns None) on shape changes.
+    """
+
+    WARMUP = 3
+
+    def __init__(self, model, loss_fn, optimizer, flip_batch: bool):
+        self.model = model
+        self.loss_fn = loss_fn
+        self.opt = optimizer
+        self.flip = flip_batch
+        self.graph = None
+        self.shape = None
+        self.warm = 0
+
+    def _body(self):
+        if self.flip:
+            from ..ops.conv import conv_flip_all
+            conv_flip_all()
+        self.opt.zero_grad(set_to_none=False)
+        preds = self.model(self.data_s)
+        loss = self.loss_fn(preds, self.label_s)
+        loss.backward()
+        if hasattr(self.model, "finish_gradient_sync"):
+            self.model.finish_gradient_sync()
+        self.opt.step()
+        return loss, preds
+
+    def step(self, data, label):
+        if self.shape is None:
+            self.shape = tuple(data.shape)
+        if tuple(data.shape) != self.shape:
+            return None
+        if self.graph is None:
+            self.warm += 1
+            if self.warm <= self.WARMUP:
+                return None                      # caller runs the eager step
+            torch.cuda.synchronize()
+            self.data_s = data.clone()
+            self.label_s = label.clone()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self.loss_s, self.preds_s = self._body()
+            self.graph = g
+            # capture records but does not execute: replay for this batch
+        else:
+            self.data_s.copy_(data, non_blocking=True)
+            self.label_s.copy_(label, non_blocking=True)
+        self.opt.sync_lr()
+        self.graph.replay()
+        return self.loss_s, self.preds_s
+
+
 def run_epoch(model, loader, loss_fn, optimizer, desc_default="", epoch=0,
               writer=None, verbose=False, scheduler=None, is_master=True,
               ema: Optional[EMA] = None, wd: float = 0.0, device="cpu",
@@ -65,41 +126,57 @@ def run_epoch(model, loader, loss_fn, optimizer, desc_default="", epoch=0,
 
     flip_batch = (optimizer is not None and device != "cpu"
                   and os.environ.get("FAA_FLIP_BATCH") == "1")
+    # graph-captured train step (pure-bf16 fused-optimizer path only)
+    graphed = None
+    if (optimizer is not None and device != "cpu" and not use_mixup and not amp
+            and os.environ.get("FAA_TRAIN_GRAPHS", "1") == "1"):
+        from ..optim import FusedRMSpropTF, FusedSGD
+        if isinstance(optimizer, (FusedSGD, FusedRMSpropTF)):
+            graphed = getattr(optimizer, "_faa_graph_step", None)
+            if graphed is None:
+                graphed = _GraphedTrainStep(model, loss_fn, optimizer, flip_batch)
+                optimizer._faa_graph_step = graphed
     for data, label in loader:
         steps += 1
-        if flip_batch:
-            from ..ops.conv import conv_flip_all
-            conv_flip_all()   # one launch refreshes all bwd-data repacks
         data = data.to(device, non_blocking=True)
         label = label.to(device, non_blocking=True)
         if device != "cpu":
             data = data.contiguous(memory_format=torch.channels_last)
 
-        with torch.autocast("cuda", dtype=autocast_dtype, enabled=amp):
-            if use_mixup:
-                data, targets, shuffled_targets, lam = mixup(data, label, conf["mixup"])
-                preds = model(data)
-                loss = loss_fn(preds, targets, shuffled_targets, lam)
-            else:
-                preds = model(data)
-                loss = loss_fn(preds, label)
-
-        if optimizer:
-            loss.backward()
-            if hasattr(model, "finish_gradient_sync"):
-                model.finish_gradient_sync()
-            from ..optim import FusedSGD, FusedRMSpropTF
-            if not isinstance(optimizer, (FusedSGD, FusedRMSpropTF)):
-                # FusedSGD folds manual WD + global clip into its kernels
-                if wd > 0.0:
-                    _apply_manual_wd(decay_params, wd)
-                grad_clip = conf["optimizer"].get("clip", 5.0)
-                if grad_clip > 0:
-                    nn.utils.clip_grad_norm_(model.parameters(), grad_clip)
-            optimizer.step()
-            optimizer.zero_grad(set_to_none=False)
+        res = graphed.step(data, label) if graphed is not None else None
+        if res is not None:
+            loss, preds = res
             if ema is not None:
                 ema(model, (epoch - 1) * total_steps + steps)
+        else:
+            if flip_batch:
+                from ..ops.conv import conv_flip_all
+                conv_flip_all()   # one launch refreshes all bwd-data repacks
+            with torch.autocast("cuda", dtype=autocast_dtype, enabled=amp):
+                if use_mixup:
+                    data, targets, shuffled_targets, lam = mixup(data, label, conf["mixup"])
+                    preds = model(data)
+                    loss = loss_fn(preds, targets, shuffled_targets, lam)
+                else:
+                    preds = model(data)
+                    loss = loss_fn(preds, label)
+
+            if optimizer:
+                loss.backward()
+                if hasattr(model, "finish_gradient_sync"):
+                    model.finish_gradient_sync()
+                from ..optim import FusedSGD, FusedRMSpropTF
+                if not isinstance(optimizer, (FusedSGD, FusedRMSpropTF)):
+                    # FusedSGD folds manual WD + global clip into its kernels
+                    if wd > 0.0:
+                        _apply_manual_wd(decay_params, wd)
+                    grad_clip = conf["optimizer"].get("clip", 5.0)
+                    if grad_clip > 0:
+                        nn.utils.clip_grad_norm_(model.parameters(), grad_clip)
+                optimizer.step()
+                optimizer.zero_grad(set_to_none=False)
+                if ema is not None:
+                    ema(model, (epoch - 1) * total_steps + steps)
 
         with torch.no_grad():
             top1, top5 = accuracy(preds, label, (1, 5))
