@@ -403,8 +403,26 @@ def main():
                 with torch.cuda.graph(graph):
                     flat_capture_body() if dist_in_graph else step_body()
         else:
-            with torch.cuda.graph(graph):
-                flat_capture_body() if dist_in_graph else step_body()
+            try:
+                with torch.cuda.graph(graph):
+                    flat_capture_body() if dist_in_graph else step_body()
+            except Exception as e:
+                if not distributed:
+                    raise
+                # distributed overlap capture failed (e.g. RCCL refuses
+                # graph capture on this stack): fall back to the eager
+                # all-reduce mode rather than dying under the driver's
+                # 8-GPU scaling run
+                print(f"# overlap capture failed ({type(e).__name__}: {e}); "
+                      f"falling back to eager dist mode", flush=True)
+                torch.cuda.synchronize()
+                dist_in_graph = True
+                if hasattr(model, "_sync_enabled"):
+                    model._sync_enabled = False   # hooks stay out of capture
+                graph = _mk_graph()
+                upload_next()
+                with torch.cuda.graph(graph):
+                    flat_capture_body()
 
     if graph_dump and graph is not None:
         try:
