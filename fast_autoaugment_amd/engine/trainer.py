@@ -58,6 +58,12 @@ class _GraphedTrainStep:
     loss/preds are read back from the capture pool after replay (stream-
     ordered). First 3 steps run eagerly (warmup), then capture.
     Falls back (returns None) on shape changes.
+
+    DEFAULT OFF (FAA_TRAIN_GRAPHS=1 to enable): the first GPU validation
+    (gpurun_out/call11.log search dry-run) showed training FREEZES after
+    capture (loss pinned at ln(10)); tools/graph_train_check.py isolates
+    whether the captured-optimizer step replays correctly before this can
+    be re-enabled.
     """
 
     WARMUP = 3
@@ -129,7 +135,7 @@ def run_epoch(model, loader, loss_fn, optimizer, desc_default="", epoch=0,
     # graph-captured train step (pure-bf16 fused-optimizer path only)
     graphed = None
     if (optimizer is not None and device != "cpu" and not use_mixup and not amp
-            and os.environ.get("FAA_TRAIN_GRAPHS", "1") == "1"):
+            and os.environ.get("FAA_TRAIN_GRAPHS", "0") == "1"):
         from ..optim import FusedRMSpropTF, FusedSGD
         if isinstance(optimizer, (FusedSGD, FusedRMSpropTF)):
             graphed = getattr(optimizer, "_faa_graph_step", None)

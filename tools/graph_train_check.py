@@ -1,0 +1,83 @@
+#!/usr/bin/env python3
+"""Does a hipGraph-captured train step actually TRAIN? (call11 freeze bisect)
+
+Three variants on one fixed batch, 60 steps each, losses every 10:
+  eager      plain loop (reference)
+  graph_fb   capture fwd+bwd only; opt.step eager between replays
+  graph_all  capture fwd+bwd+opt.step (the trainer _GraphedTrainStep shape)
+A frozen loss in graph_* that decreases in eager localizes the breakage.
+"""
+import sys
+
+sys.path.insert(0, ".")
+import torch
+
+
+def build(lr=0.05):
+    from fast_autoaugment_amd.metrics import CrossEntropyLabelSmooth
+    from fast_autoaugment_amd.models import build_model
+    from fast_autoaugment_amd.optim import FusedSGD
+    from fast_autoaugment_amd.ops.conv import patch_convs
+    from fast_autoaugment_amd.parallel.flat import flatten_module
+    torch.manual_seed(0)
+    m = build_model({"type": "wresnet40_2"}, 10).cuda().to(
+        memory_format=torch.channels_last)
+    flat = flatten_module(m, work_dtype=torch.bfloat16)
+    patch_convs(m)
+    opt = FusedSGD(flat, lr=lr, momentum=0.9, nesterov=True,
+                   weight_decay=2e-4, grad_clip=5.0)
+    crit = CrossEntropyLabelSmooth(10, 0.0)
+    m.train()
+    return m, flat, opt, crit
+
+
+def data_batch():
+    torch.manual_seed(7)
+    d = (torch.randn(128, 3, 32, 32, device="cuda") * 0.5).bfloat16() \
+        .contiguous(memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (128,), device="cuda")
+    return d, y
+
+
+def run(mode, steps=60):
+    m, flat, opt, crit = build()
+    d, y = data_batch()
+
+    def body():
+        opt.zero_grad(set_to_none=False)
+        loss = crit(m(d), y)
+        loss.backward()
+        return loss
+
+    losses = []
+    if mode == "eager":
+        for i in range(steps):
+            loss = body()
+            opt.step()
+            if i % 10 == 0 or i == steps - 1:
+                losses.append(round(loss.item(), 4))
+    else:
+        for _ in range(3):
+            loss = body()
+            opt.step()
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        if mode == "graph_all":
+            with torch.cuda.graph(g):
+                loss_s = body()
+                opt.step()
+        else:
+            with torch.cuda.graph(g):
+                loss_s = body()
+        for i in range(steps - 3):
+            g.replay()
+            if mode == "graph_fb":
+                opt.step()
+            if i % 10 == 0 or i == steps - 4:
+                losses.append(round(loss_s.item(), 4))
+    print(f"{mode:>10}: {losses}", flush=True)
+
+
+if __name__ == "__main__":
+    for mode in (sys.argv[1:] or ["eager", "graph_fb", "graph_all"]):
+        run(mode)
