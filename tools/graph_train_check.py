@@ -78,6 +78,76 @@ def run(mode, steps=60):
     print(f"{mode:>10}: {losses}", flush=True)
 
 
+
+
+def run_gather(mode, steps=60):
+    """bench.py-shaped capture variants.
+
+    gather_bwd: side-stream warmup, p.grad=None, capture fwd+BACKWARD
+                (round-1/2 bench gather mode — AccumulateGrad inside capture)
+    gather_ag : capture fwd + torch.autograd.grad (NO AccumulateGrad)
+    all_ag    : gather_ag + gather_grads + opt.step all inside the graph
+    """
+    from fast_autoaugment_amd.ops import ext
+    CX = ext()
+    m, flat, opt, crit = build()
+    d, y = data_batch()
+
+    def fwd_loss():
+        return crit(m(d), y)
+
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for _ in range(3):
+            opt.zero_grad(set_to_none=False)
+            loss = fwd_loss()
+            loss.backward()
+            opt.step()
+    torch.cuda.current_stream().wait_stream(side)
+    del loss
+    torch.cuda.synchronize()
+
+    g = torch.cuda.CUDAGraph()
+    base = flat.flat_param.data_ptr()
+    if mode == "gather_bwd":
+        for p in flat.params:
+            p.grad = None
+        with torch.cuda.graph(g):
+            loss_s = fwd_loss()
+            loss_s.backward()
+        gs = [p.grad for p in flat.params]
+    else:
+        with torch.cuda.graph(g):
+            loss_s = fwd_loss()
+            gs = torch.autograd.grad(loss_s, flat.params)
+            if mode == "all_ag":
+                rows, tb = [], None
+                for p, gr in zip(flat.params, gs):
+                    rows.append([gr.data_ptr(), (p.data.data_ptr() - base) // 2,
+                                 gr.numel()])
+                tb = torch.tensor(rows, dtype=torch.int64, device="cuda")
+                CX.gather_grads(tb, flat.flat_grad)
+                opt.step()
+    if mode != "all_ag":
+        rows = [[gr.data_ptr(), (p.data.data_ptr() - base) // 2, gr.numel()]
+                for p, gr in zip(flat.params, gs)]
+        tb = torch.tensor(rows, dtype=torch.int64, device="cuda")
+        flat.flat_grad.zero_()
+    losses = []
+    for i in range(steps - 3):
+        g.replay()
+        if mode != "all_ag":
+            CX.gather_grads(tb, flat.flat_grad)
+            opt.step()
+        if i % 10 == 0 or i == steps - 4:
+            losses.append(round(loss_s.item(), 4))
+    print(f"{mode:>10}: {losses}", flush=True)
+
+
 if __name__ == "__main__":
     for mode in (sys.argv[1:] or ["eager", "graph_fb", "graph_all"]):
-        run(mode)
+        if mode.startswith(("gather", "all_")):
+            run_gather(mode)
+        else:
+            run(mode)
