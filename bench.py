@@ -362,15 +362,45 @@ def main():
             saved_grads = [p_.grad for p_ in flat.params]
             for p_ in flat.params:
                 p_.grad = None          # capture assignment-mode backward
-            with torch.cuda.graph(graph):
-                gpu_fwd_bwd()
+            # FAA_BENCH_AG=1: compute grads with torch.autograd.grad inside
+            # the capture instead of loss.backward() — bypasses the
+            # AccumulateGrad nodes whose warmup-stream pinning injects
+            # cross-stream syncs into the capture (the round-1 colsum
+            # corruption mechanism, tools/graph_train_check.py)
+            ag_mode = os.environ.get("FAA_BENCH_AG", "1") == "1"
+            ag_grads = None
+            if ag_mode:
+                def fwd_loss_only():
+                    if conv_flip_all is not None and os.environ.get("FAA_FLIP_BATCH") == "1":
+                        conv_flip_all()
+                    if fixed_data:
+                        data, label = fixed["data"], fixed["label"]
+                    else:
+                        if imagenet:
+                            data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s,
+                                                            post_s, mean_t, std_t,
+                                                            out_size, out_size, bf16)
+                        else:
+                            data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
+                                                   mean_t, std_t, bf16)
+                        label = store.labels.index_select(0, sel_s)
+                    return crit(model(data), label)
+                with torch.cuda.graph(graph):
+                    loss_ag = fwd_loss_only()
+                    ag_grads = torch.autograd.grad(loss_ag, flat.params,
+                                                   allow_unused=True)
+            else:
+                with torch.cuda.graph(graph):
+                    gpu_fwd_bwd()
             # grads now live in the graph pool at replay-stable addresses
             base = flat.flat_param.data_ptr()
             rows = []
             ok = True
-            for p_ in flat.params:
-                g_ = p_.grad
-                if g_ is None or g_.dtype != torch.bfloat16:
+            for pi_, p_ in enumerate(flat.params):
+                g_ = ag_grads[pi_] if ag_mode else p_.grad
+                if g_ is None:
+                    continue            # unused param: flat region stays zero
+                if g_.dtype != torch.bfloat16:
                     ok = False
                     break
                 # element order must match the flat region's (channels_last

@@ -51,19 +51,19 @@ def _apply_manual_wd(params, wd: float):
 class _GraphedTrainStep:
     """hipGraph-captured train step for the trainer hot loop.
 
-    The eager loop issues ~1500 kernel launches per step and measured 61%
-    GPU-busy (profiles/train_profile_r02.txt); capturing
-    flip+zero+fwd+loss+bwd+opt.step as ONE replayable graph removes the
-    launch tail. Batches are copied into static input buffers each step;
-    loss/preds are read back from the capture pool after replay (stream-
-    ordered). First 3 steps run eagerly (warmup), then capture.
-    Falls back (returns None) on shape changes.
-
-    DEFAULT OFF (FAA_TRAIN_GRAPHS=1 to enable): the first GPU validation
-    (gpurun_out/call11.log search dry-run) showed training FREEZES after
-    capture (loss pinned at ln(10)); tools/graph_train_check.py isolates
-    whether the captured-optimizer step replays correctly before this can
-    be re-enabled.
+    The eager loop issues ~1500 kernel launches per step at 61% GPU-busy
+    (profiles/train_profile_r02.txt). The capture body computes grads with
+    ``torch.autograd.grad`` — NOT ``loss.backward()`` — because backward
+    routes through per-param AccumulateGrad nodes whose recorded stream
+    (from the first eager backward) injects cross-stream syncs into the
+    capture: segfault when that stream is the default stream, silently
+    mis-captured dependency edges otherwise (tools/graph_train_check.py,
+    gpurun_out/call12/13.log — the round-1 "colsum corruption" mechanism).
+    Grads land at stable capture-pool addresses; one gather kernel packs
+    them into the flat buffer and the fused optimizer steps eagerly after
+    each replay (4 launches/step). Batches are copied into static input
+    buffers; loss/preds read back after replay. Falls back (returns None)
+    on shape changes or layout-incompatible grads.
     """
 
     WARMUP = 3
@@ -76,21 +76,47 @@ class _GraphedTrainStep:
         self.graph = None
         self.shape = None
         self.warm = 0
+        self.dead = False
 
-    def _body(self):
-        if self.flip:
-            from ..ops.conv import conv_flip_all
-            conv_flip_all()
-        self.opt.zero_grad(set_to_none=False)
-        preds = self.model(self.data_s)
-        loss = self.loss_fn(preds, self.label_s)
-        loss.backward()
-        if hasattr(self.model, "finish_gradient_sync"):
-            self.model.finish_gradient_sync()
-        self.opt.step()
-        return loss, preds
+    def _capture(self, data, label):
+        import torch as _t
+        flat = self.opt.flat
+        self.data_s = data.clone()
+        self.label_s = label.clone()
+        _t.cuda.synchronize()
+        g = _t.cuda.CUDAGraph()
+        with _t.cuda.graph(g):
+            if self.flip:
+                from ..ops.conv import conv_flip_all
+                conv_flip_all()
+            preds = self.model(self.data_s)
+            loss = self.loss_fn(preds, self.label_s)
+            gs = _t.autograd.grad(loss, flat.params, allow_unused=True)
+        base = flat.flat_param.data_ptr()
+        rows = []
+        for p, gr in zip(flat.params, gs):
+            if gr is None:
+                continue
+            if gr.dtype != _t.bfloat16:
+                return False
+            if gr.dim() == 4:
+                ok = (gr.is_contiguous(memory_format=_t.channels_last)
+                      or (gr.is_contiguous()
+                          and (gr.size(1) == 1 or gr.size(2) * gr.size(3) == 1)))
+            else:
+                ok = gr.is_contiguous()
+            if not ok:
+                return False
+            rows.append([gr.data_ptr(), (p.data.data_ptr() - base) // 2, gr.numel()])
+        self.table = _t.tensor(rows, dtype=_t.int64, device=data.device)
+        flat.flat_grad.zero_()          # pad gaps / unused params stay zero
+        self.graph = g
+        self.loss_s, self.preds_s = loss, preds
+        return True
 
     def step(self, data, label):
+        if self.dead:
+            return None
         if self.shape is None:
             self.shape = tuple(data.shape)
         if tuple(data.shape) != self.shape:
@@ -99,19 +125,17 @@ class _GraphedTrainStep:
             self.warm += 1
             if self.warm <= self.WARMUP:
                 return None                      # caller runs the eager step
-            torch.cuda.synchronize()
-            self.data_s = data.clone()
-            self.label_s = label.clone()
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self.loss_s, self.preds_s = self._body()
-            self.graph = g
+            if not self._capture(data, label):
+                self.dead = True
+                return None
             # capture records but does not execute: replay for this batch
         else:
             self.data_s.copy_(data, non_blocking=True)
             self.label_s.copy_(label, non_blocking=True)
-        self.opt.sync_lr()
         self.graph.replay()
+        from ..ops import ext
+        ext().gather_grads(self.table, self.opt.flat.flat_grad)
+        self.opt.step()
         return self.loss_s, self.preds_s
 
 
