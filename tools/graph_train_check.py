@@ -149,5 +149,42 @@ if __name__ == "__main__":
     for mode in (sys.argv[1:] or ["eager", "graph_fb", "graph_all"]):
         if mode.startswith(("gather", "all_")):
             run_gather(mode)
+        elif mode == "class_fixed":
+            run_class(vary_data=False)
+        elif mode == "class_vary":
+            run_class(vary_data=True)
         else:
             run(mode)
+
+
+def run_class(steps=60, vary_data=False):
+    """Drive the actual engine._GraphedTrainStep class (freeze bisect)."""
+    from fast_autoaugment_amd.engine.trainer import _GraphedTrainStep
+    m, flat, opt, crit = build()
+    d, y = data_batch()
+    gstep = _GraphedTrainStep(m, crit, opt, False)
+    losses = []
+    torch.manual_seed(77)
+    for i in range(steps):
+        if vary_data:
+            di = (d + 0.01 * torch.randn_like(d.float()).bfloat16()).contiguous(
+                memory_format=torch.channels_last)
+        else:
+            di = d
+        res = gstep.step(di, y)
+        if res is None:
+            opt.zero_grad(set_to_none=False)
+            loss = crit(m(di), y)
+            loss.backward()
+            opt.step()
+            opt.zero_grad(set_to_none=False)
+            del loss
+        else:
+            loss = res[0]
+        if i % 10 == 0 or i == steps - 1:
+            losses.append(round(float(loss.item()), 4))
+        if i == 10:
+            gnorm = float(flat.flat_grad.float().norm().item())
+            print(f"  [class vary={vary_data}] grad norm after gather @10: {gnorm:.4f}",
+                  flush=True)
+    print(f"class(vary={vary_data}): {losses}", flush=True)
