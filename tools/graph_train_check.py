@@ -145,18 +145,6 @@ def run_gather(mode, steps=60):
     print(f"{mode:>10}: {losses}", flush=True)
 
 
-if __name__ == "__main__":
-    for mode in (sys.argv[1:] or ["eager", "graph_fb", "graph_all"]):
-        if mode.startswith(("gather", "all_")):
-            run_gather(mode)
-        elif mode == "class_fixed":
-            run_class(vary_data=False)
-        elif mode == "class_vary":
-            run_class(vary_data=True)
-        else:
-            run(mode)
-
-
 def run_class(steps=60, vary_data=False):
     """Drive the actual engine._GraphedTrainStep class (freeze bisect)."""
     from fast_autoaugment_amd.engine.trainer import _GraphedTrainStep
@@ -188,3 +176,15 @@ def run_class(steps=60, vary_data=False):
             print(f"  [class vary={vary_data}] grad norm after gather @10: {gnorm:.4f}",
                   flush=True)
     print(f"class(vary={vary_data}): {losses}", flush=True)
+
+
+if __name__ == "__main__":
+    for mode in (sys.argv[1:] or ["eager", "graph_fb", "graph_all"]):
+        if mode.startswith(("gather", "all_")):
+            run_gather(mode)
+        elif mode == "class_fixed":
+            run_class(vary_data=False)
+        elif mode == "class_vary":
+            run_class(vary_data=True)
+        else:
+            run(mode)
