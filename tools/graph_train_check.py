@@ -166,11 +166,12 @@ def run_class(steps=60, vary_data=False):
             loss.backward()
             opt.step()
             opt.zero_grad(set_to_none=False)
+            lv = float(loss.item())
             del loss
         else:
-            loss = res[0]
+            lv = float(res[0].item())
         if i % 10 == 0 or i == steps - 1:
-            losses.append(round(float(loss.item()), 4))
+            losses.append(round(lv, 4))
         if i == 10:
             gnorm = float(flat.flat_grad.float().norm().item())
             print(f"  [class vary={vary_data}] grad norm after gather @10: {gnorm:.4f}",
