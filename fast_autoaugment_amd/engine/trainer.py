@@ -136,6 +136,19 @@ class _GraphedTrainStep:
         from ..ops import ext
         ext().gather_grads(self.table, self.opt.flat.flat_grad)
         self.opt.step()
+        if os.environ.get("FAA_TRAIN_GRAPH_DEBUG") == "1":
+            self._dbg = getattr(self, "_dbg", 0) + 1
+            if self._dbg in (1, 10, 40):
+                import torch as _t
+                _t.cuda.synchronize()
+                fl = self.opt.flat
+                print(f"# graphstep dbg n={self._dbg} loss={float(self.loss_s):.4f} "
+                      f"|g|={float(fl.flat_grad.float().norm()):.4f} "
+                      f"|p|={float(fl.flat_param.float().norm()):.4f} "
+                      f"lr={self.opt.param_groups[0]['lr']:.5f} "
+                      f"rows={self.table.size(0)} "
+                      f"data|m|={float(self.data_s.float().abs().mean()):.4f}",
+                      flush=True)
         return self.loss_s, self.preds_s
 
 
