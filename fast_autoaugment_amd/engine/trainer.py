@@ -496,6 +496,14 @@ def train_and_eval(tag, dataroot, test_ratio=0.0, cv_fold=0, reporter=None,
                         "ema": ema.state_dict() if ema is not None else None,
                     }, save_path)
 
+    # drop the cached step graph + its capture pool before the next fold
+    # trains in this worker process (search runs folds sequentially)
+    if hasattr(optimizer, "_faa_graph_step"):
+        optimizer._faa_graph_step = None
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        import gc
+        gc.collect()
     del model
     if metric != "last":
         # reference train.py:321 assigns best_top1 unconditionally, which
