@@ -835,6 +835,38 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
     __syncthreads();
   }
 
+  if (WSPLIT == 4) {
+    // the 4 waves hold duplicate-position partials: fold them through LDS
+    // (reusing ldsXT, 256 lanes x 8 floats = 8 KB per round, within the
+    // smallest instantiation's 9.2 KB) so only wave 0 issues atomics
+    float* red = reinterpret_cast<float*>(ldsXT);
+    for (int tap = 0; tap < KK * KK; ++tap) {
+      for (int mf = 0; mf < 2; ++mf) {
+        __syncthreads();
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          #pragma unroll
+          for (int r = 0; r < 4; ++r)
+            red[(wave * 64 + lane) * 8 + nf * 4 + r] = acc[tap][mf][nf][r];
+        __syncthreads();
+        if (wave == 0) {
+          #pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              int j = nf * 4 + r;
+              acc[tap][mf][nf][r] = red[lane * 8 + j]
+                  + red[(64 + lane) * 8 + j]
+                  + red[(128 + lane) * 8 + j]
+                  + red[(192 + lane) * 8 + j];
+            }
+        }
+      }
+    }
+    __syncthreads();
+    if (wave != 0) return;
+  }
+
   // epilogue: one atomicAdd per accumulator element
   #pragma unroll
   for (int tap = 0; tap < KK * KK; ++tap) {
