@@ -251,3 +251,29 @@ def test_batched_flip_matches_per_conv():
     got = C.conv2d_fwd(dy, w2, torch.Tensor(), 1, 1).float()
     err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
     assert err < 1e-3, f"batched flip mismatch {err}"
+
+
+@pytest.mark.parametrize("wsplit", ["0", "1"])
+def test_wrw3_small_c_variants(wsplit):
+    """wrw v3 at small C: quadrant config vs wave-split + LDS cross-wave
+    reduce, both vs fp32 torch."""
+    import os
+    torch.manual_seed(13)
+    for (B, Cin, H, Cout, k) in [(8, 16, 32, 32, 3), (8, 32, 16, 32, 3),
+                                 (8, 32, 32, 32, 1)]:
+        x = torch.randn(B, Cin, H, H, device=dev()) * 0.5
+        w_shape = [Cout, Cin, k, k]
+        dy = torch.randn(B, Cout, H, H, device=dev()) * 0.1
+        ref = torch.nn.grad.conv2d_weight(x, w_shape, dy, stride=1, padding=k // 2)
+        os.environ["FAA_WRW_V3"] = "1"
+        os.environ["FAA_WRW3_WSPLIT"] = wsplit
+        try:
+            dw, _ = C.conv2d_bwd_weight(
+                dy.bfloat16().contiguous(memory_format=torch.channels_last),
+                x.bfloat16().contiguous(memory_format=torch.channels_last),
+                1, k // 2, k, k, False)
+        finally:
+            os.environ.pop("FAA_WRW_V3", None)
+            os.environ.pop("FAA_WRW3_WSPLIT", None)
+        err = (dw.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+        assert err < 2e-2, f"wsplit={wsplit} {Cin}x{H}->{Cout} k{k} rel err {err}"

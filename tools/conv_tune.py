@@ -42,7 +42,7 @@ def main():
                     (128, 160, 32, 160, 3, 1), (128, 320, 16, 320, 3, 1),
                     (128, 640, 8, 640, 3, 1)]
         slc = ["8", "16", "32", "64", "128", "256"]
-        print(f"{'shape':<24} {'miopen':>8} {'v2':>8} {'v3auto':>8} "
+        print(f"{'shape':<24} {'miopen':>8} {'v2':>8} {'v3auto':>8} {'v3ws':>8} "
               + " ".join(f"s{v:>7}" for v in slc))
         for B, Cin, H, Cout, k, s in shapes_w:
             x = (torch.randn(B, Cin, H, H, device="cuda") * 0.5).bfloat16() \
@@ -57,6 +57,9 @@ def main():
             os.environ["FAA_WRW_V3"] = "1"
             os.environ.pop("FAA_WRW3_SLICES", None)
             row.append(f"{bench(lambda: C.conv2d_bwd_weight(dy, x, s, k//2, k, k, False), args.iters):8.1f}")
+            os.environ["FAA_WRW3_WSPLIT"] = "1"
+            row.append(f"{bench(lambda: C.conv2d_bwd_weight(dy, x, s, k//2, k, k, False), args.iters):8.1f}")
+            os.environ.pop("FAA_WRW3_WSPLIT", None)
             for v in slc:
                 os.environ["FAA_WRW3_SLICES"] = v
                 row.append(f"{bench(lambda: C.conv2d_bwd_weight(dy, x, s, k//2, k, k, False), args.iters):8.1f}")
