@@ -238,6 +238,59 @@ def faa_conv2d(x, weight, bias, stride: int, padding: int):
     return FaaConv2dFn.apply(x, weight, bias, stride, padding)
 
 
+class FaaGroupedConvFn(torch.autograd.Function):
+    """Grouped 3x3 s1 conv (ShakeResNeXt cardinality branches,
+    reference shake_resnext.py:34): in-house direct-kernel forward,
+    torch/MIOpen backward (grouped bwd shapes unmeasured in-house)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding, groups):
+        C = ext()
+        ctx.save_for_backward(x, weight)
+        ctx.stride, ctx.padding, ctx.groups = stride, padding, groups
+        ctx.has_bias = bias is not None
+        b = bias if bias is not None else torch.Tensor()
+        return C.conv2d_fwd_grouped(x, weight, b, stride, padding, groups)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = dw = dbias = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.nn.grad.conv2d_input(list(x.shape), weight, dy,
+                                            stride=ctx.stride, padding=ctx.padding,
+                                            groups=ctx.groups)
+        if ctx.needs_input_grad[1]:
+            dw = torch.nn.grad.conv2d_weight(
+                x, list(weight.shape), dy, stride=ctx.stride,
+                padding=ctx.padding, groups=ctx.groups).contiguous(
+                    memory_format=torch.channels_last)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            dbias = _dbias(dy)
+        return dx, dw, dbias, None, None, None
+
+
+def _grouped_eligible(m: torch.nn.Conv2d) -> bool:
+    return (m.groups > 1 and m.groups != m.in_channels
+            and m.kernel_size == (3, 3) and m.stride == (1, 1)
+            and m.padding == (1, 1) and m.dilation == (1, 1)
+            and (m.in_channels // m.groups) >= 16
+            and (m.in_channels // m.groups) % 8 == 0
+            and (m.out_channels // m.groups) % 32 == 0)
+
+
+def _faa_grouped_forward(self, x):
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and self.weight.dtype == torch.bfloat16
+            and x.size(3) in (8, 16, 32) and x.size(2) % 8 == 0
+            and not (x.size(3) == 8 and (x.size(2) != 8 or x.size(0) % 2))):
+        return FaaGroupedConvFn.apply(x, self.weight, self.bias,
+                                      self.stride[0], self.padding[0], self.groups)
+    return F.conv2d(x, self.weight, self.bias, self.stride, self.padding,
+                    self.dilation, self.groups)
+
+
 def _eligible(m: torch.nn.Conv2d) -> bool:
     k = m.kernel_size
     base = (k[0] == k[1] and k[0] in (1, 3)
@@ -331,6 +384,9 @@ def patch_convs(model: torch.nn.Module) -> int:
             continue
         if _dw_eligible(m):
             m.forward = types.MethodType(_faa_dw_forward, m)
+            n += 1
+        elif _grouped_eligible(m):
+            m.forward = types.MethodType(_faa_grouped_forward, m)
             n += 1
         elif _eligible(m):
             m.forward = types.MethodType(_faa_forward, m)

@@ -47,6 +47,7 @@ struct ConvGeom {
   int KH, KW, stride, pad;
   int cin_chunks;      // ceil(Cin/8)
   int kpad;            // KH*KW*cin_chunks*8
+  int groups = 1;      // grouped conv (direct tiled kernel only)
 };
 
 // load 8 input channels (zero-filled out of bounds / tail) as bf16x8
@@ -294,12 +295,17 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
   f32x4 acc[MR][NR] = {};
   const int row0 = ty * TH - 1;      // pad = 1
 
-  int cs_lo = 0, cs_hi = g.Cin;
+  // grouped conv: this n-tile's group selects a cin slice; the weight's
+  // inner dim is cin_per (host guarantees the n-tile never spans groups)
+  const int cin_per = (g.groups > 1) ? (g.Cin / g.groups) : g.Cin;
+  const int cbase = (g.groups > 1) ? (n0 / (g.Cout / g.groups)) * cin_per : 0;
+
+  int cs_lo = 0, cs_hi = cin_per;
   if (SPLIT) {                       // blockIdx.y partitions the cin slabs
-    int slabs = (g.Cin + CS - 1) / CS;
+    int slabs = (cin_per + CS - 1) / CS;
     int per = (slabs + gridDim.y - 1) / gridDim.y;
     cs_lo = blockIdx.y * per * CS;
-    cs_hi = min(g.Cin, cs_lo + per * CS);
+    cs_hi = min(cin_per, cs_lo + per * CS);
     if (cs_lo >= cs_hi) return;
   }
   for (int cs = cs_lo; cs < cs_hi; cs += CS) {
@@ -316,12 +322,13 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
       int wi = col - 1;
       int ci0 = cs + kc * 8;
       bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.Wd && ci0 < g.Cin) {
-        const short* p = X + ((((int64_t)(img0 + ib)) * g.H + hi) * g.Wd + wi) * g.Cin + ci0;
-        if (ci0 + 8 <= g.Cin) {
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.Wd && ci0 < cin_per) {
+        const short* p = X + ((((int64_t)(img0 + ib)) * g.H + hi) * g.Wd + wi) * g.Cin
+                         + cbase + ci0;
+        if (ci0 + 8 <= cin_per) {
           val = *reinterpret_cast<const bf16x8*>(p);
         } else {
-          for (int j = 0; j < g.Cin - ci0; ++j) val[j] = p[j];
+          for (int j = 0; j < cin_per - ci0; ++j) val[j] = p[j];
         }
       }
       *reinterpret_cast<bf16x8*>(&ldsX[cell * PS + kc * 8]) = val;
@@ -335,12 +342,12 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
         int n = n0 + nf * 16 + fr;
         int ci0 = cs + kq;
         bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (n < g.Cout && ci0 < g.Cin) {
-          const short* p = Wt + ((int64_t)n * 9 + tap) * g.Cin + ci0;
-          if (ci0 + 8 <= g.Cin) {
+        if (n < g.Cout && ci0 < cin_per) {
+          const short* p = Wt + ((int64_t)n * 9 + tap) * cin_per + ci0;
+          if (ci0 + 8 <= cin_per) {
             v = *reinterpret_cast<const bf16x8*>(p);
           } else {
-            for (int j = 0; j < g.Cin - ci0; ++j) v[j] = p[j];
+            for (int j = 0; j < cin_per - ci0; ++j) v[j] = p[j];
           }
         }
         wreg[tap][nf] = v;
@@ -1229,6 +1236,73 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w,
                      Cout, KH, KW, Cin);
   // full-correlation: dX = conv(dY, W2, pad = KH-1-pad)
   return conv2d_fwd(dyc, w2, torch::Tensor(), 1, KH - 1 - pad);
+}
+
+torch::Tensor conv2d_fwd_grouped(torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor bias, int64_t stride,
+                                  int64_t pad, int64_t groups) {
+  // grouped 3x3 s1 conv through the direct tiled kernel (ShakeResNeXt
+  // cardinality-4 branches, reference shake_resnext.py:34). Python gates
+  // eligibility; hard requirements checked here.
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && w.scalar_type() == torch::kBFloat16);
+  auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto wc = w.contiguous(torch::MemoryFormat::ChannelsLast);
+  ConvGeom g;
+  g.B = xc.size(0); g.Cin = xc.size(1); g.H = xc.size(2); g.Wd = xc.size(3);
+  g.Cout = wc.size(0); g.KH = wc.size(2); g.KW = wc.size(3);
+  g.stride = stride; g.pad = pad; g.groups = (int)groups;
+  g.Ho = g.H; g.Wo = g.Wd;
+  g.cin_chunks = 0; g.kpad = 0;
+  int cin_per = wc.size(1), cout_per = g.Cout / (int)groups;
+  TORCH_CHECK(g.Cin == cin_per * groups && g.Cout % groups == 0,
+              "grouped geometry mismatch");
+  TORCH_CHECK(g.KH == 3 && g.KW == 3 && stride == 1 && pad == 1
+              && cin_per >= 16 && cout_per % 32 == 0
+              && ((g.Wd == 32 && g.H % 8 == 0) || (g.Wd == 16 && g.H % 8 == 0)
+                  || (g.Wd == 8 && g.H == 8 && g.B % 2 == 0)),
+              "conv2d_fwd_grouped: unsupported geometry");
+  auto y = torch::empty({g.B, g.Cout, g.Ho, g.Wo},
+                        xc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  bool has_bias = bias.defined() && bias.numel() > 0;
+  torch::Tensor bc;
+  const short* bptr = nullptr;
+  if (has_bias) {
+    bc = bias.contiguous();
+    bptr = (const short*)bc.data_ptr();
+  }
+  int tiles_h = g.H / 8;
+  int ib = (g.Wd == 8) ? 2 : 1;
+  bool big = (cin_per > 64) && (cout_per % 64 == 0);
+  #define CTG_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, HB_, GRID_, THTILES_, GN_)   \
+    hipLaunchKernelGGL((conv3x3_tile_kernel<TH_, TW_, IB_, WN_, MR_, NR_, HB_>),\
+                       dim3((unsigned)(GRID_)), dim3(256), 0, stream,           \
+                       (const short*)xc.data_ptr(),                             \
+                       (const short*)wc.data_ptr(), bptr,                       \
+                       (short*)y.data_ptr(), g, THTILES_, GN_)
+  #define CTG_BOTH(TH_, TW_, IB_, WN_, MR_, NR_, GRID_, THTILES_, GN_)          \
+    do {                                                                        \
+      if (has_bias) CTG_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, true,              \
+                               GRID_, THTILES_, GN_);                           \
+      else CTG_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, false,                      \
+                      GRID_, THTILES_, GN_);                                    \
+    } while (0)
+  if (big) {
+    int gn64 = g.Cout / 64;
+    int tiles_h4 = g.H / 4;
+    if (g.Wd == 32) CTG_BOTH(4, 32, 1, 2, 4, 2, (int64_t)g.B * tiles_h4 * gn64, tiles_h4, gn64);
+    else if (g.Wd == 16) CTG_BOTH(8, 16, 1, 2, 4, 2, (int64_t)g.B * tiles_h * gn64, tiles_h, gn64);
+    else CTG_BOTH(8, 8, 2, 2, 4, 2, (int64_t)(g.B / 2) * gn64, 1, gn64);
+  } else {
+    int gn = g.Cout / 32;
+    int64_t grid32 = (int64_t)(g.B / ib) * tiles_h * gn;
+    if (g.Wd == 32) CTG_BOTH(8, 32, 1, 1, 4, 2, grid32, tiles_h, gn);
+    else if (g.Wd == 16) CTG_BOTH(8, 16, 1, 1, 2, 2, grid32, tiles_h, gn);
+    else CTG_BOTH(8, 8, 2, 1, 2, 2, grid32, tiles_h, gn);
+  }
+  #undef CTG_BOTH
+  #undef CTG_LAUNCH
+  return y;
 }
 
 torch::Tensor conv2d_bwd_data_s2(torch::Tensor dy, torch::Tensor w,

@@ -44,6 +44,9 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w, int64_t stride,
                               int64_t pad, int64_t H, int64_t W_in);
 torch::Tensor conv2d_bwd_data_s2(torch::Tensor dy, torch::Tensor w,
                                  int64_t H, int64_t W, bool pre_flipped);
+torch::Tensor conv2d_fwd_grouped(torch::Tensor x, torch::Tensor w,
+                                 torch::Tensor bias, int64_t stride,
+                                 int64_t pad, int64_t groups);
 std::vector<torch::Tensor> conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                              int64_t stride, int64_t pad,
                                              int64_t KH, int64_t KW, bool want_bias);
@@ -85,6 +88,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_data_s2", &conv2d_bwd_data_s2,
         "stride-2 3x3 bwd-data via phase decomposition");
+  m.def("conv2d_fwd_grouped", &conv2d_fwd_grouped,
+        "grouped 3x3 s1 conv (cardinality branches) on the direct kernel");
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("colsum_bf16", &colsum_bf16, "channel column-sum (bias grad), replay-safe v2");
   m.def("colsum_bf16_legacy", &colsum_bf16_legacy,
