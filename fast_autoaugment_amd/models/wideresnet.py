@@ -12,7 +12,7 @@ from __future__ import annotations
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.modules import bn_relu
+from ..ops.modules import bn_relu, res_add
 
 
 class WideBasic(nn.Module):
@@ -34,7 +34,7 @@ class WideBasic(nn.Module):
         if self.dropout.p > 0:
             out = self.dropout(out)
         out = self.conv2(bn_relu(out, self.bn2))
-        return out + self.shortcut(x)
+        return res_add(out, self.shortcut(x))
 
 
 class WideResNet(nn.Module):

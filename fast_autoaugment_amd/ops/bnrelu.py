@@ -10,13 +10,54 @@ from . import ext
 ACT_NONE, ACT_RELU, ACT_SWISH = 0, 1, 2
 
 
+# handoff from the fused residual-add (add + bn fwd-reduce in one pass):
+# keyed by the add-output's data_ptr, popped by the consuming BN call
+_pending_bn_stats = {}
+
+
+class ResidualAddBnStatsFn(torch.autograd.Function):
+    """out = a + b, also producing the FOLLOWING BatchNorm's fwd-reduce
+    partials (pre-act join points: reference wideresnet.py:41). The add's
+    backward is identity fan-out."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        C = ext()
+        out, scratch = C.residual_add_bn_stats(a, b)
+        _pending_bn_stats.clear()        # at most one join in flight
+        _pending_bn_stats[out.data_ptr()] = scratch
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        return dy, dy
+
+
+def residual_add(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Residual join that pre-computes the next BN's reduce when the fused
+    path applies; plain a + b otherwise."""
+    import os
+    if (a.is_cuda and a.dtype == torch.bfloat16 and b.dtype == a.dtype
+            and a.shape == b.shape and a.size(1) % 8 == 0
+            and a.numel() % 8 == 0 and a.size(1) <= 2048
+            and os.environ.get("FAA_ADD_BN_FUSE", "1") == "1"):
+        return ResidualAddBnStatsFn.apply(a, b)
+    return a + b
+
+
+_EMPTY = torch.Tensor()
+
+
 class FusedBNReLUFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
                 momentum, eps, act):
         C = ext()
+        pre = _pending_bn_stats.pop(x.data_ptr(), None)
+        if pre is None or not training:
+            pre = _EMPTY
         out, mean, invstd = C.bn_relu_fwd(x, gamma, beta, running_mean, running_var,
-                                          training, momentum, eps, act)
+                                          training, momentum, eps, act, pre)
         ctx.save_for_backward(x, out, mean, invstd, gamma, beta)
         ctx.training = training
         ctx.act = act

@@ -28,6 +28,11 @@ __device__ __forceinline__ float bf16_bits_to_f(short u) {
   return c.f;
 }
 
+__device__ __forceinline__ short f_to_bf16_bits(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
+}
+
 // ------------------------------------------------- last-block finalize
 // FAA_BN_LASTBLOCK=1 (staged for round 2, default off): the LAST reduce
 // block to finish performs the finalize inline instead of a separate
@@ -214,6 +219,86 @@ __global__ void bn_reduce_anyc_kernel(const T* __restrict__ x, float* __restrict
   bn_fin_tail_run(tail, scratch, C);
 }
 
+// residual add fused with the FOLLOWING BatchNorm's fwd reduce: the join
+// points of the pre-act nets (WideBasic's `out + shortcut`, reference
+// wideresnet.py:41) feed straight into the next block's bn1, so summing
+// sum/sumsq while writing the add result removes bn_reduce's separate
+// full-tensor read pass (profiles/step_profile_wrn40_2_r02.txt: at::add
+// 3.1% + bn_reduce_vec 5.1%). Same channel-invariance contract as
+// bn_reduce_vec_kernel (host picks nb via bn_nblocks).
+template <typename T>
+__global__ void add_bn_reduce_vec_kernel(const T* __restrict__ a,
+                                         const T* __restrict__ b,
+                                         T* __restrict__ out,
+                                         float* __restrict__ scratch,
+                                         int64_t total, int C) {
+  float s[8] = {0}, ss[8] = {0};
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  if constexpr (sizeof(T) == 2) {
+    const short* as = reinterpret_cast<const short*>(a);
+    const short* bs = reinterpret_cast<const short*>(b);
+    short* os = reinterpret_cast<short*>(out);
+    for (int64_t i = i0; i < total; i += stride) {
+      short4v a0 = *reinterpret_cast<const short4v*>(as + i);
+      short4v a1 = *reinterpret_cast<const short4v*>(as + i + 4);
+      short4v b0 = *reinterpret_cast<const short4v*>(bs + i);
+      short4v b1 = *reinterpret_cast<const short4v*>(bs + i + 4);
+      short4v o0, o1;
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float f0 = bf16_bits_to_f(a0[k]) + bf16_bits_to_f(b0[k]);
+        float f1 = bf16_bits_to_f(a1[k]) + bf16_bits_to_f(b1[k]);
+        o0[k] = f_to_bf16_bits(f0);
+        o1[k] = f_to_bf16_bits(f1);
+        // accumulate the ROUNDED value: the next BN reads the bf16 result
+        f0 = bf16_bits_to_f(o0[k]);
+        f1 = bf16_bits_to_f(o1[k]);
+        s[k] += f0; ss[k] += f0 * f0;
+        s[k + 4] += f1; ss[k + 4] += f1 * f1;
+      }
+      *reinterpret_cast<short4v*>(os + i) = o0;
+      *reinterpret_cast<short4v*>(os + i + 4) = o1;
+    }
+  } else {
+    for (int64_t i = i0; i < total; i += stride) {
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float f = faa_to_float(a[i + k]) + faa_to_float(b[i + k]);
+        faa_from_float(f, &out[i + k]);
+        s[k] += f; ss[k] += f * f;
+      }
+    }
+  }
+  __shared__ float lds[256 * 8];
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = s[k];
+  __syncthreads();
+  float* outp = scratch + (int64_t)blockIdx.x * 2 * C;
+  const int groups = C / 8;
+  const int shift = (int)(((int64_t)blockIdx.x * blockDim.x) % groups);
+  for (int ch = threadIdx.x; ch < C; ch += (int)blockDim.x) {
+    int oct = ch / 8, lane = ch % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups)
+      acc += lds[t * 8 + lane];
+    outp[ch] = acc;
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = ss[k];
+  __syncthreads();
+  for (int ch = threadIdx.x; ch < C; ch += (int)blockDim.x) {
+    int oct = ch / 8, lane = ch % 8;
+    int t0 = (oct - shift + groups) % groups;
+    float acc = 0;
+    for (int t = t0; t < (int)blockDim.x; t += groups)
+      acc += lds[t * 8 + lane];
+    outp[C + ch] = acc;
+  }
+}
+
 // one 64-lane wave per channel: lanes stride the partial blocks in parallel
 // (a serial per-thread loop over ~64 partials costs ~16us in pure latency)
 __global__ void bn_finalize_kernel(const float* __restrict__ scratch, int nblocks,
@@ -243,10 +328,6 @@ __global__ void bn_finalize_kernel(const float* __restrict__ scratch, int nblock
 }
 
 // ---------------------------------------------------------------- fwd apply
-__device__ __forceinline__ short f_to_bf16_bits(float f) {
-  __hip_bfloat16 h = __float2bfloat16(f);
-  return *reinterpret_cast<short*>(&h);
-}
 
 // fused activation after the affine transform: 0 = none, 1 = relu, 2 = swish
 template <int ACT>
@@ -601,10 +682,32 @@ static int bn_nblocks_anyc(int C, int64_t total) {
   return ((base + q - 1) / q) * q;
 }
 
+std::vector<torch::Tensor> residual_add_bn_stats(torch::Tensor a, torch::Tensor b) {
+  // out = a + b (NHWC) + per-block BN partials for the FOLLOWING bn_relu_fwd
+  auto ac = a.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto bc = b.contiguous(torch::MemoryFormat::ChannelsLast);
+  int C = ac.size(1);
+  int64_t total = ac.numel();
+  TORCH_CHECK(C % 8 == 0 && total % 8 == 0 && C <= 2048,
+              "residual_add_bn_stats: vec-path geometry required");
+  auto out = torch::empty_like(ac);
+  int nb = bn_nblocks(C, total);
+  auto scratch = torch::empty({nb, 2 * C}, ac.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  DISPATCH_FB(ac.scalar_type(), "add_bn_reduce", [&] {
+    hipLaunchKernelGGL((add_bn_reduce_vec_kernel<scalar_t>), dim3(nb), dim3(256),
+                       0, stream, (const scalar_t*)ac.data_ptr(),
+                       (const scalar_t*)bc.data_ptr(), (scalar_t*)out.data_ptr(),
+                       scratch.data_ptr<float>(), total, C);
+  });
+  return {out, scratch};
+}
+
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
-                                       double momentum, double eps, int64_t act) {
+                                       double momentum, double eps, int64_t act,
+                                       torch::Tensor pre_scratch) {
   TORCH_CHECK(x.dim() == 4, "bn_relu: 4-D input expected");
   auto xc = x.contiguous(torch::MemoryFormat::ChannelsLast);
   int C = xc.size(1);
@@ -633,7 +736,18 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
       tail.eps = (float)eps;
       tail.momentum = (float)momentum;
     }
-    if (vec) {
+    bool have_pre = pre_scratch.defined() && pre_scratch.numel() > 0
+        && pre_scratch.dim() == 2 && pre_scratch.size(1) == 2 * C;
+    if (have_pre && !lastblock) {
+      // partials precomputed by residual_add_bn_stats: skip the reduce pass
+      int nb = pre_scratch.size(0);
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0, stream,
+                         pre_scratch.data_ptr<float>(), nb, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                         C, rows, (float)eps, (float)momentum);
+    } else if (vec) {
       int nb = bn_nblocks(C, total);
       auto scratch = torch::empty({nb, 2 * C}, f32);
       DISPATCH_FB(xc.scalar_type(), "bn_reduce", [&] {

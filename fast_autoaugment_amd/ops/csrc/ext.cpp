@@ -33,7 +33,9 @@ torch::Tensor aug_pipeline_imagenet(torch::Tensor images, torch::Tensor sel,
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta, torch::Tensor running_mean,
                                        torch::Tensor running_var, bool training,
-                                       double momentum, double eps, int64_t act);
+                                       double momentum, double eps, int64_t act,
+                                       torch::Tensor pre_scratch);
+std::vector<torch::Tensor> residual_add_bn_stats(torch::Tensor a, torch::Tensor b);
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor out, torch::Tensor mean,
                                        torch::Tensor invstd, torch::Tensor gamma,
@@ -84,6 +86,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("aug_pipeline", &aug_pipeline,
         "batched augmentation program executor (uint8 NHWC -> normalized bf16/f32)");
   m.def("bn_relu_fwd", &bn_relu_fwd);
+  m.def("residual_add_bn_stats", &residual_add_bn_stats,
+        "out = a + b, plus BN fwd-reduce partials for the following bn");
   m.def("conv2d_fwd", &conv2d_fwd, "MFMA implicit-GEMM NHWC bf16 conv forward");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_data_s2", &conv2d_bwd_data_s2,

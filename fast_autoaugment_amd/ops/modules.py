@@ -43,3 +43,13 @@ def bn_swish(x: torch.Tensor, bn: torch.nn.BatchNorm2d) -> torch.Tensor:
         return fused_bn_swish(x, bn)
     from .functional import swish
     return swish(bn(x))
+
+
+def res_add(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Residual join; on GPU it also pre-computes the next BN's fwd-reduce
+    partials in the same pass (consumed by the following bn_relu call)."""
+    if (a.is_cuda and has_ext()
+            and os.environ.get("FAA_NO_FUSED_BN") != "1"):
+        from .bnrelu import residual_add
+        return residual_add(a, b)
+    return a + b

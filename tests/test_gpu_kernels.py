@@ -478,3 +478,32 @@ def test_bn_swish_fwd_bwd_vs_torch(Ch):
     oute = fused_bn_swish(xe, bn)
     assert (oute.float() - ze * torch.sigmoid(ze)).abs().max().item() < tol
 
+
+
+def test_residual_add_bn_stats_fusion():
+    """Fused residual-add + BN fwd-reduce handoff == separate add + BN."""
+    import os
+    torch.manual_seed(21)
+    dev_ = dev()
+    a = (torch.randn(16, 64, 16, 16, device=dev_) * 0.5).bfloat16().contiguous(
+        memory_format=torch.channels_last)
+    b = (torch.randn(16, 64, 16, 16, device=dev_) * 0.5).bfloat16().contiguous(
+        memory_format=torch.channels_last)
+    out, scratch = C.residual_add_bn_stats(a, b)
+    ref_out = a + b
+    assert torch.equal(out, ref_out), "fused add output differs from a+b"
+
+    from fast_autoaugment_amd.ops.bnrelu import fused_bn_relu, residual_add
+    bn1 = torch.nn.BatchNorm2d(64, momentum=0.3).to(dev_).train()
+    bn2 = torch.nn.BatchNorm2d(64, momentum=0.3).to(dev_).train()
+    bn2.load_state_dict(bn1.state_dict())
+    os.environ["FAA_ADD_BN_FUSE"] = "1"
+    try:
+        y1 = fused_bn_relu(residual_add(a, b), bn1)
+    finally:
+        os.environ.pop("FAA_ADD_BN_FUSE", None)
+    y2 = fused_bn_relu(ref_out, bn2)
+    s = y2.float().abs().max().item() + 1e-3
+    assert (y1.float() - y2.float()).abs().max().item() / s < 1e-2
+    assert (bn1.running_mean - bn2.running_mean).abs().max().item() < 1e-4
+    assert (bn1.running_var - bn2.running_var).abs().max().item() < 1e-4
