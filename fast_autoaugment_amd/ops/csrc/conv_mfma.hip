@@ -733,8 +733,13 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
   constexpr int XC = (KK == 3) ? TW + 2 : TW;
   constexpr int PIX_IN = XR * XC;
   constexpr int PIX_OUT = TH * TW;
-  constexpr int XTP = wrw3_pitch(PIX_IN);
-  constexpr int YTP = wrw3_pitch(PIX_OUT);
+  // row-dependent swizzle (multiples of 8 shorts = 16B): the staging
+  // scatter writes 8 rows (kc*8+j, j fixed) per instruction, whose bank
+  // step without a swizzle is a multiple of 32 -> up to 16-way conflicts.
+  // shift(row) spreads those rows over 8 bank groups while keeping the
+  // contiguous 8-pixel fragment reads aligned.
+  constexpr int XTP = wrw3_pitch(PIX_IN + 56);
+  constexpr int YTP = wrw3_pitch(PIX_OUT + 56);
   static_assert(PIX_OUT % 32 == 0, "pixel tile must be a multiple of 32");
 
   __shared__ short ldsXT[64 * XTP];
@@ -789,7 +794,11 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
         }
       }
       #pragma unroll
-      for (int j = 0; j < 8; ++j) ldsXT[(kc * 8 + j) * XTP + cell] = val[j];
+      for (int j = 0; j < 8; ++j) {
+        int row = kc * 8 + j;
+        int sh = (((row >> 3) + (row & 7)) & 7) * 8;
+        ldsXT[row * XTP + sh + cell] = val[j];
+      }
     }
     // ---- stage dY^T [co][output pixel]
     for (int v = tid; v < PIX_OUT * 8; v += 256) {
@@ -808,7 +817,11 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
         }
       }
       #pragma unroll
-      for (int j = 0; j < 8; ++j) ldsYT[(kc * 8 + j) * YTP + cell] = val[j];
+      for (int j = 0; j < 8; ++j) {
+        int row = kc * 8 + j;
+        int sh = (((row >> 3) + (row & 7)) & 7) * 8;
+        ldsYT[row * YTP + sh + cell] = val[j];
+      }
     }
     __syncthreads();
 
@@ -819,9 +832,12 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
       const int py = p8 / TW, px = p8 % TW;
       bf16x8 bfrag[2];
       #pragma unroll
-      for (int nf = 0; nf < 2; ++nf)
+      for (int nf = 0; nf < 2; ++nf) {
+        int row = wn * 32 + nf * 16 + fr;
+        int sh = (((row >> 3) + (row & 7)) & 7) * 8;
         bfrag[nf] = *reinterpret_cast<const bf16x8*>(
-            &ldsYT[(wn * 32 + nf * 16 + fr) * YTP + pc * 32 + kq]);
+            &ldsYT[row * YTP + sh + pc * 32 + kq]);
+      }
       #pragma unroll
       for (int kh = 0; kh < KK; ++kh) {
         #pragma unroll
@@ -829,8 +845,10 @@ void conv_wrw3_kernel(const short* __restrict__ X, const short* __restrict__ dY,
           const int ip = (py + kh) * XC + (px + kw);
           #pragma unroll
           for (int mf = 0; mf < 2; ++mf) {
+            int arow = wm * 32 + mf * 16 + fr;
+            int ash = (((arow >> 3) + (arow & 7)) & 7) * 8;
             bf16x8 a = *reinterpret_cast<const bf16x8*>(
-                &ldsXT[(wm * 32 + mf * 16 + fr) * XTP + ip]);
+                &ldsXT[arow * XTP + ash + ip]);
             #pragma unroll
             for (int nf = 0; nf < 2; ++nf)
               acc[kh * KK + kw][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
