@@ -131,14 +131,20 @@ def main():
     # CPU/gloo (and --graphs 0) use FlatDDP's bucketed overlapped all-reduce.
     use_graphs = bool(args.graphs) and not cpu_mode
     # Distributed capture modes (FAA_BENCH_DIST_MODE):
-    #   overlap (default): FlatDDP's bucketed all-reduce on the comm stream
-    #     is CAPTURED inside the step graph via the post-accumulate hooks,
-    #     so each replay replays the backward-overlapped comm schedule
-    #     (VERDICT r1 item 4). Requires flat grad mode.
-    #   eager: round-1 behavior — graph holds zero+fwd+bwd, one eager
-    #     all-reduce of the whole flat grad between replay and the step.
-    dist_mode = os.environ.get("FAA_BENCH_DIST_MODE", "overlap")
-    dist_in_graph = distributed and use_graphs and dist_mode == "eager"
+    #   gather (default): the single-GPU-proven capture (fwd + autograd.grad,
+    #     no backward/hooks in graph); per step: replay -> gather kernel ->
+    #     eager RCCL all-reduce of the flat grad -> fused step. Survives
+    #     200+ replays (call27/28 validation).
+    #   overlap: FlatDDP's bucketed all-reduce CAPTURED via backward hooks —
+    #     overlaps comm with backward inside the replay, but hits the
+    #     platform graph-pool corruption after ~150 replays on this stack
+    #     (NaN params, gpurun_out/call28.log; docs/GRAPH_NAN.md). Kept for
+    #     future stacks / short-horizon measurement.
+    #   eager: flat-accumulation capture + whole-buffer eager all-reduce.
+    dist_mode = os.environ.get("FAA_BENCH_DIST_MODE", "gather")
+    dist_in_graph = distributed and use_graphs and dist_mode != "overlap"
+    if distributed and use_graphs and dist_mode == "eager":
+        args.grad_mode = "flat"
     if distributed and not dist_in_graph:
         from fast_autoaugment_amd.parallel.ddp import FlatDDP
         if use_graphs:

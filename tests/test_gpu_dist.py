@@ -37,9 +37,19 @@ def _run_bench(extra_env, steps=5):
     return out
 
 
+def test_rccl_gather_graph_step():
+    """Default distributed mode: AG-gather capture + eager gather ->
+    all-reduce -> fused step over real RCCL at ws=1, 40 steps (the
+    replay-endurance horizon that exposed the overlap mode's corruption)."""
+    out = _run_bench({"FAA_BENCH_DIST_MODE": "gather"}, steps=40)
+    assert out["n_gpus"] == 1 and out["ms_per_step"] > 0
+
+
 def test_rccl_overlap_graph_step():
     """FlatDDP bucketed all-reduce CAPTURED inside the step graph (comm
-    stream hooks), replayed over real RCCL at ws=1."""
+    stream hooks), replayed over real RCCL at ws=1. Short horizon only:
+    this mode corrupts after ~150 replays on the current stack
+    (docs/GRAPH_NAN.md) and is not the default."""
     out = _run_bench({"FAA_BENCH_DIST_MODE": "overlap"})
     assert out["n_gpus"] == 1 and out["ms_per_step"] > 0
 
