@@ -77,6 +77,13 @@ class _GraphedTrainStep:
         self.shape = None
         self.warm = 0
         self.dead = False
+        self.replays = 0
+        # platform-bug workaround (docs/GRAPH_NAN.md): the graph pool
+        # corrupts after ~100-150 replays when eager allocations churn
+        # between replays; re-capturing with a FRESH graph+pool every
+        # RECAP_EVERY steps stays under the onset (~1 extra eager-step
+        # cost amortized over the window)
+        self.recap_every = int(os.environ.get("FAA_TRAIN_GRAPH_RECAP", "64"))
 
     def _capture(self, data, label):
         import torch as _t
@@ -128,10 +135,21 @@ class _GraphedTrainStep:
             if not self._capture(data, label):
                 self.dead = True
                 return None
+            self.replays = 0
             # capture records but does not execute: replay for this batch
+        elif self.replays >= self.recap_every > 0:
+            import gc
+            self.graph = None
+            torch.cuda.synchronize()
+            gc.collect()                         # free the old graph + pool
+            if not self._capture(data, label):
+                self.dead = True
+                return None
+            self.replays = 0
         else:
             self.data_s.copy_(data, non_blocking=True)
             self.label_s.copy_(label, non_blocking=True)
+        self.replays += 1
         self.graph.replay()
         from ..ops import ext
         ext().gather_grads(self.table, self.opt.flat.flat_grad)
