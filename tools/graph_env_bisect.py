@@ -97,7 +97,7 @@ def main(mode):
                     correct += (preds.argmax(1) == y).sum()
                     cnt += y.numel()
                 sched.step(ep - 1 + step / 18.0)
-            if ep % 5 == 0 and cnt:
+            if ep % 2 == 0 and cnt:
                 print(mode, "ep", ep, "top1",
                       round(float(correct.item()) / cnt, 4), flush=True)
     print(mode, "wall", round(time.time() - t0, 1), flush=True)
