@@ -177,6 +177,76 @@ __global__ void dw_fwd_tpl_kernel(const short* __restrict__ X, const short* __re
   }
 }
 
+// v3: LDS-staged input tile. A block owns ONE channel octet and a TxT
+// output tile of one image: the (T*S + K-1)^2 x 8ch input window is staged
+// in LDS once (every input byte read ~once from HBM vs ~K^2/pw times in
+// the register-sliding kernels — round-1 measured those ~9x off the HBM
+// roofline), the 8ch x K^2 weights live in LDS (broadcast reads), and
+// each thread does K^2 LDS vec-reads per output.
+template <bool HAS_BIAS, int K, int S, int T>
+__global__ __launch_bounds__(256)
+void dw_fwd_v3_kernel(const short* __restrict__ X, const short* __restrict__ Wt,
+                      const short* __restrict__ bias, short* __restrict__ Y,
+                      DwGeom g, int tiles_h, int tiles_w) {
+  constexpr int IT = T * S + K - 1;           // staged input tile edge
+  __shared__ short ldsX[IT * IT * 8];
+  __shared__ short ldsW[K * K * 8];
+  __shared__ float ldsB[8];
+
+  int bid = blockIdx.x;
+  const int tx = bid % tiles_w; bid /= tiles_w;
+  const int ty = bid % tiles_h; bid /= tiles_h;
+  const int oct = bid % (g.C >> 3);
+  const int b = bid / (g.C >> 3);
+  const int c0 = oct << 3;
+  const int tid = threadIdx.x;
+
+  const int hi0 = ty * T * S - g.pt;
+  const int wi0 = tx * T * S - g.pl;
+  // stage input window (zero-filled out of bounds)
+  for (int v = tid; v < IT * IT; v += 256) {
+    int r = v / IT, c = v % IT;
+    int hi = hi0 + r, wi = wi0 + c;
+    bf16v8 xv = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+      xv = *reinterpret_cast<const bf16v8*>(
+          X + (((int64_t)b * g.H + hi) * g.W + wi) * g.C + c0);
+    *reinterpret_cast<bf16v8*>(&ldsX[v * 8]) = xv;
+  }
+  if (tid < K * K) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      ldsW[tid * 8 + j] = Wt[(int64_t)(c0 + j) * K * K + tid];
+  }
+  if (HAS_BIAS && tid < 8) ldsB[tid] = dw_b2f(bias[c0 + tid]);
+  __syncthreads();
+
+  for (int p = tid; p < T * T; p += 256) {
+    int hl = p / T, wl = p % T;
+    int ho = ty * T + hl, wo = tx * T + wl;
+    if (ho >= g.Ho || wo >= g.Wo) continue;
+    float acc[8] = {};
+    #pragma unroll
+    for (int kh = 0; kh < K; ++kh) {
+      #pragma unroll
+      for (int kw = 0; kw < K; ++kw) {
+        bf16v8 xv = *reinterpret_cast<const bf16v8*>(
+            &ldsX[((hl * S + kh) * IT + wl * S + kw) * 8]);
+        const short* wv = &ldsW[(kh * K + kw) * 8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[j] += dw_b2f(xv[j]) * dw_b2f(wv[j]);
+      }
+    }
+    bf16v8 ov;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      ov[j] = dw_f2b(acc[j] + (HAS_BIAS ? ldsB[j] : 0.0f));
+    *reinterpret_cast<bf16v8*>(
+        Y + (((int64_t)b * g.Ho + ho) * g.Wo + wo) * g.C + c0) = ov;
+  }
+}
+
 __global__ void dw_bwd_data_kernel(const short* __restrict__ dY, const short* __restrict__ Wt,
                                    short* __restrict__ dX, DwGeom g) {
   int64_t total = (int64_t)g.B * g.H * g.W * g.C;
@@ -317,6 +387,40 @@ torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   torch::Tensor bc;
   const short* bp = nullptr;
   if (hb) { bc = bias.contiguous(); bp = (const short*)bc.data_ptr(); }
+
+  // v3 LDS-staged tile path (default; FAA_DW_V3=0 falls back)
+  static const bool v3_off = []() {
+    const char* e = getenv("FAA_DW_V3");
+    return e && e[0] == '0';
+  }();
+  if (!v3_off && (KH == KW) && (KH == 3 || KH == 5)
+      && (stride == 1 || stride == 2)) {
+    int T = (Ho >= 12 && Wo >= 12) ? 16 : 8;
+    int tiles_h = (Ho + T - 1) / T;
+    int tiles_w = (Wo + T - 1) / T;
+    int64_t grid = (int64_t)g.B * (g.C / 8) * tiles_h * tiles_w;
+    #define DWV3(K_, S_, T_)                                                    \
+      do {                                                                      \
+        if (hb)                                                                 \
+          hipLaunchKernelGGL((dw_fwd_v3_kernel<true, K_, S_, T_>),              \
+                             dim3((unsigned)grid), dim3(256), 0, stream,        \
+                             (const short*)xc.data_ptr(),                       \
+                             (const short*)wc.data_ptr(), bp,                   \
+                             (short*)y.data_ptr(), g, tiles_h, tiles_w);        \
+        else                                                                    \
+          hipLaunchKernelGGL((dw_fwd_v3_kernel<false, K_, S_, T_>),             \
+                             dim3((unsigned)grid), dim3(256), 0, stream,        \
+                             (const short*)xc.data_ptr(),                       \
+                             (const short*)wc.data_ptr(), nullptr,              \
+                             (short*)y.data_ptr(), g, tiles_h, tiles_w);        \
+      } while (0)
+    if (KH == 3 && stride == 1) { if (T == 16) DWV3(3, 1, 16); else DWV3(3, 1, 8); }
+    else if (KH == 3)           { if (T == 16) DWV3(3, 2, 16); else DWV3(3, 2, 8); }
+    else if (stride == 1)       { if (T == 16) DWV3(5, 1, 16); else DWV3(5, 1, 8); }
+    else                        { if (T == 16) DWV3(5, 2, 16); else DWV3(5, 2, 8); }
+    #undef DWV3
+    return y;
+  }
 
   // specialized path: K in {3,5}, stride in {1,2}; octet-invariant grid.
   // k=5 keeps 25 bf16x8 weight vectors in registers (~100 VGPRs) which cuts

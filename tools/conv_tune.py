@@ -32,8 +32,36 @@ def main():
                     help="sweep FAA_CONV_D8 variants on the deep 8x8 shapes")
     ap.add_argument("--wrw", action="store_true",
                     help="sweep wrw v3 slice counts vs v2 vs MIOpen")
+    ap.add_argument("--dw", action="store_true",
+                    help="depthwise fwd: v3 LDS-tile vs register-sliding vs torch")
     ap.add_argument("--iters", type=int, default=50)
     args = ap.parse_args()
+    if args.dw:
+        import math
+        from fast_autoaugment_amd.ops import ext as _ext
+        C = _ext()
+        shapes_d = [(64, 32, 112, 3, 1), (64, 96, 112, 3, 2), (64, 144, 56, 5, 2),
+                    (64, 240, 28, 5, 1), (64, 576, 14, 5, 1), (64, 1152, 7, 5, 1)]
+        print(f"{'shape':<24} {'torch':>8} {'v3':>8} {'old':>8}")
+        for B, Ch, H, k, s in shapes_d:
+            x = (torch.randn(B, Ch, H, H, device="cuda") * 0.5).bfloat16() \
+                .contiguous(memory_format=torch.channels_last)
+            w = (torch.randn(Ch, 1, k, k, device="cuda") * 0.2).bfloat16() \
+                .contiguous(memory_format=torch.channels_last)
+            ph = max((math.ceil(H / s) - 1) * s + k - H, 0)
+            pl, pt = ph // 2, ph // 2
+            pr, pb = ph - pl, ph - pt
+            xp = torch.nn.functional.pad(x.float(), (pl, pr, pt, pb)).bfloat16() \
+                .contiguous(memory_format=torch.channels_last)
+            row = [f"{B}x{Ch}x{H}^2 k{k}s{s}"]
+            row.append(f"{bench(lambda: torch.nn.functional.conv2d(xp, w, stride=s, groups=Ch), args.iters):8.1f}")
+            os.environ.pop("FAA_DW_V3", None)
+            row.append(f"{bench(lambda: C.dwconv_fwd(x, w, torch.Tensor(), s, pt, pb, pl, pr), args.iters):8.1f}")
+            os.environ["FAA_DW_V3"] = "0"
+            row.append(f"{bench(lambda: C.dwconv_fwd(x, w, torch.Tensor(), s, pt, pb, pl, pr), args.iters):8.1f}")
+            os.environ.pop("FAA_DW_V3", None)
+            print(" ".join(row))
+        return
     if args.wrw:
         from fast_autoaugment_amd.ops import ext as _ext
         C = _ext()
