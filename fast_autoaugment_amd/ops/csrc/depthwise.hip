@@ -389,10 +389,8 @@ torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   if (hb) { bc = bias.contiguous(); bp = (const short*)bc.data_ptr(); }
 
   // v3 LDS-staged tile path (default; FAA_DW_V3=0 falls back)
-  static const bool v3_off = []() {
-    const char* e = getenv("FAA_DW_V3");
-    return e && e[0] == '0';
-  }();
+  const char* v3e = getenv("FAA_DW_V3");
+  const bool v3_off = v3e && v3e[0] == '0';
   if (!v3_off && (KH == KW) && (KH == 3 || KH == 5)
       && (stride == 1 || stride == 2)) {
     int T = (Ho >= 12 && Wo >= 12) ? 16 : 8;
@@ -426,10 +424,8 @@ torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   // k=5 keeps 25 bf16x8 weight vectors in registers (~100 VGPRs) which cuts
   // occupancy below what the latency-bound loop needs — measured slower on
   // EfficientNet-B0 — so k=5 stays on the generic kernel unless FAA_DW_TPL5=1.
-  static const bool tpl5 = []() {
-    const char* e = getenv("FAA_DW_TPL5");
-    return e && e[0] == '1';
-  }();
+  const char* t5e = getenv("FAA_DW_TPL5");
+  const bool tpl5 = t5e && t5e[0] == '1';
   bool tpl = (KH == KW) && (KH == 3 || (KH == 5 && tpl5)) && (stride == 1 || stride == 2);
   if (tpl) {
     auto gcd = [](int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; };
