@@ -26,7 +26,7 @@ import torch
 
 def build(model_name="wresnet40_2"):
     from fast_autoaugment_amd.metrics import CrossEntropyLabelSmooth
-    from fast_autoaugment_amd.models import build_model, num_class
+    from fast_autoaugment_amd.models import build_model
     from fast_autoaugment_amd.optim import FusedSGD
     from fast_autoaugment_amd.ops.conv import patch_convs
     from fast_autoaugment_amd.parallel.flat import flatten_module
