@@ -381,6 +381,7 @@ void conv3x3_tile_kernel(const short* __restrict__ X, const short* __restrict__ 
       int pix = m_loc % (TH * TW);
       int py = pix / TW, px = pix % TW;
       int ho = ty * TH + py;
+      if (ho >= g.Ho || px >= g.Wo) continue;   // masked tail tiles
       int64_t base = ((((int64_t)(img0 + ib)) * g.Ho + ho) * g.Wo + px) * g.Cout;
       #pragma unroll
       for (int nf = 0; nf < NR; ++nf) {
@@ -1084,22 +1085,26 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   {
     const char* de = getenv("FAA_CONV_DIRECT");
     bool want = !(de && de[0] == '0');
+    // tile width covering Wd (exact for the CIFAR sizes; masked tails for
+    // the ResNet-50 ImageNet spatials 56/28/14/7 — epilogue guards skip
+    // out-of-range rows/cols, staging zero-fills them)
+    int twsel = (g.Wd <= 8) ? 8 : (g.Wd <= 16) ? 16 : (g.Wd <= 32) ? 32
+                : (g.Wd <= 56) ? 56 : 0;
     bool geom_ok = (g.KH == 3 && g.KW == 3 && g.stride == 1 && g.pad == 1
-                    && g.Cin >= 16
-                    && ((g.Wd == 32 && g.H % 8 == 0)
-                        || (g.Wd == 16 && g.H % 8 == 0)
-                        || (g.Wd == 8 && g.H == 8 && g.B % 2 == 0)));
+                    && g.Cin >= 16 && twsel > 0
+                    && (twsel != 8 || g.B % 2 == 0));
     if (want && geom_ok) {
-      int tiles_h = g.H / 8;
-      int ib = (g.Wd == 8) ? 2 : 1;
+      int tiles_h = (g.H + 7) / 8;
+      int ib = (twsel == 8) ? 2 : 1;
       // wide config (BN=64): fewer weight re-reads per FLOP; only when the
       // grid still fills the chip and for W=32 the TH=4 variant exists
       int gn64 = (g.Cout + 63) / 64;
-      int tiles_h4 = g.H / 4;
-      int64_t blocks64 = (g.Wd == 32)
+      int tiles_h4 = (g.H + 3) / 4;
+      int64_t blocks64 = (twsel == 32)
           ? (int64_t)g.B * tiles_h4 * gn64
           : (int64_t)(g.B / ib) * tiles_h * gn64;
-      bool big = (g.Cin > 64) && (blocks64 >= (g.Wd == 8 ? 192 : 512));
+      bool big = (g.Cin > 64) && (twsel != 56)
+                 && (blocks64 >= (twsel == 8 ? 192 : 512));
       if (de && de[0] == 'b') big = true;
       if (de && de[0] == 's') big = false;
       #define CT_LAUNCH(TH_, TW_, IB_, WN_, MR_, NR_, HB_, GRID_, THTILES_, GN_)   \
@@ -1116,9 +1121,9 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                          GRID_, THTILES_, GN_);                                    \
         } while (0)
       if (big) {
-        if (g.Wd == 32) {
+        if (twsel == 32) {
           CT_BOTH(4, 32, 1, 2, 4, 2, (int64_t)g.B * tiles_h4 * gn64, tiles_h4, gn64);
-        } else if (g.Wd == 16) {
+        } else if (twsel == 16) {
           CT_BOTH(8, 16, 1, 2, 4, 2, (int64_t)g.B * tiles_h * gn64, tiles_h, gn64);
         } else {
           // W=8 deep stages: M is tiny (B*64 px) and K is huge, so weight
@@ -1169,8 +1174,9 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
       } else {
         int gn = (g.Cout + 31) / 32;
         int64_t grid32 = (int64_t)(g.B / ib) * tiles_h * gn;
-        if (g.Wd == 32) CT_BOTH(8, 32, 1, 1, 4, 2, grid32, tiles_h, gn);
-        else if (g.Wd == 16) CT_BOTH(8, 16, 1, 1, 2, 2, grid32, tiles_h, gn);
+        if (twsel == 56) CT_BOTH(8, 56, 1, 1, 7, 2, grid32, tiles_h, gn);
+        else if (twsel == 32) CT_BOTH(8, 32, 1, 1, 4, 2, grid32, tiles_h, gn);
+        else if (twsel == 16) CT_BOTH(8, 16, 1, 1, 2, 2, grid32, tiles_h, gn);
         else CT_BOTH(8, 8, 2, 1, 2, 2, grid32, tiles_h, gn);
       }
       #undef CT_BOTH

@@ -35,6 +35,11 @@ SHAPES = [
     (4, 256, 16, 256, 3, 1),
     (4, 320, 8, 320, 3, 1),
     (2, 640, 8, 640, 3, 1),
+    # ResNet-50 ImageNet spatials (masked direct tiles: 56/28/14/7)
+    (2, 64, 56, 64, 3, 1),
+    (2, 128, 28, 128, 3, 1),
+    (2, 256, 14, 256, 3, 1),
+    (2, 512, 7, 512, 3, 1),
 ]
 
 
