@@ -307,16 +307,24 @@ def _eligible(m: torch.nn.Conv2d) -> bool:
 
 
 def _runtime_faa_ok(m, x) -> bool:
-    """Per-call fwd dispatch: measured small/mid shapes always; big-channel
-    shapes only where the direct kernel fires and wins (call3/call4 logs)."""
-    if m.in_channels <= _FAA_MAX_CH and m.out_channels <= _FAA_MAX_CH:
-        return True
+    """Per-call fwd dispatch from per-shape measurements (call3/4/35 logs):
+    small/mid channels in-house except the ImageNet spatials MIOpen wins;
+    big channels only where the direct kernel fires and wins."""
     h, w = x.size(2), x.size(3)
-    if not (m.kernel_size[0] == 3 and m.stride[0] == 1
-            and w in (8, 16, 32) and h % 8 == 0):
+    k3s1 = m.kernel_size[0] == 3 and m.stride[0] == 1
+    if m.in_channels <= _FAA_MAX_CH and m.out_channels <= _FAA_MAX_CH:
+        # measured exceptions (call35, b128): MIOpen wins the 28^2 and 7^2
+        # 3x3s (76.6 vs 90.1 at 128ch@28; 107.9 vs 114.6 at 512...)
+        if k3s1 and w == h and w in (28, 7):
+            return False
+        return True
+    if not k3s1:
         return False
-    # 640ch @8px: MIOpen still ahead (call4 D8: 150 vs 159 close; fwd table
-    # 640^2@8 in-house loses in CONVBENCH) -> keep torch there
+    if w == h and w in (56, 14):
+        return True   # masked direct tiles win (call35: 87.7 vs 125.4 @14)
+    if not (w in (8, 16, 32) and h % 8 == 0):
+        return False
+    # 640ch @8px: MIOpen still ahead (call4 D8 / CONVBENCH)
     return not (max(m.in_channels, m.out_channels) >= 640 and w <= 8)
 
 

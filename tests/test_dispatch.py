@@ -76,3 +76,14 @@ def test_dbias_cpu_fallback():
     dy = torch.randn(4, 12, 8, 8)   # C % 8 != 0 -> always torch sum
     ref = dy.sum(dim=(0, 2, 3))
     assert torch.allclose(C._dbias(dy), ref)
+
+
+def test_runtime_fwd_dispatch_imagenet_spatials():
+    # measured (gpurun_out/call35.log): direct wins 56^2 and 14^2; MIOpen
+    # keeps 28^2 and 7^2
+    assert C._runtime_faa_ok(_conv(64, 64, 3, 1), torch.zeros(2, 64, 56, 56))
+    assert C._runtime_faa_ok(_conv(256, 256, 3, 1), torch.zeros(2, 256, 14, 14))
+    assert not C._runtime_faa_ok(_conv(128, 128, 3, 1), torch.zeros(2, 128, 28, 28))
+    assert not C._runtime_faa_ok(_conv(512, 512, 3, 1), torch.zeros(2, 512, 7, 7))
+    # 1x1s keep the small-channel blanket rule
+    assert C._runtime_faa_ok(_conv(64, 64, 1, 1), torch.zeros(2, 64, 28, 28))
