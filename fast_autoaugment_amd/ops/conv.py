@@ -192,6 +192,9 @@ class FaaConv2dFn(torch.autograd.Function):
             ch = max(weight.size(0), weight.size(1))
             faa_ok = (ctx.stride == 1 and ch <= 640
                       and not (ch >= 512 and x.size(2) <= 8)
+                      # ImageNet 28^2/7^2: MIOpen wins these 3x3 shapes
+                      # fwd-side (call35) — mirror for the flipped conv
+                      and not (x.size(2) == x.size(3) and x.size(2) in (28, 7))
                       and _os.environ.get("FAA_BWD_DATA", "faa") != "torch")
             s2_ok = (ctx.stride == 2 and weight.size(2) == 3 and ctx.padding == 1
                      and x.size(3) in (16, 32) and x.size(2) % 8 == 0
