@@ -275,8 +275,8 @@ def main():
     fixed_data = os.environ.get("FAA_BENCH_FIXED_DATA") == "1"
     fixed = {}
 
-    def gpu_fwd_bwd():
-        """aug + forward + loss + backward on static inputs (capturable)."""
+    def gpu_fwd_loss():
+        """aug + forward + loss on static inputs (capturable)."""
         if conv_flip_all is not None and os.environ.get("FAA_FLIP_BATCH") == "1":
             conv_flip_all()
         if fixed_data:
@@ -294,7 +294,10 @@ def main():
                                        mean_t, std_t, bf16)
             label = store.labels.index_select(0, sel_s)
         preds = model(data)
-        loss = crit(preds, label)
+        return crit(preds, label)
+
+    def gpu_fwd_bwd():
+        loss = gpu_fwd_loss()
         loss.backward()
         return loss
 
@@ -376,23 +379,8 @@ def main():
             ag_mode = os.environ.get("FAA_BENCH_AG", "1") == "1"
             ag_grads = None
             if ag_mode:
-                def fwd_loss_only():
-                    if conv_flip_all is not None and os.environ.get("FAA_FLIP_BATCH") == "1":
-                        conv_flip_all()
-                    if fixed_data:
-                        data, label = fixed["data"], fixed["label"]
-                    else:
-                        if imagenet:
-                            data = CX.aug_pipeline_imagenet(store.images, sel_s, prog_s,
-                                                            post_s, mean_t, std_t,
-                                                            out_size, out_size, bf16)
-                        else:
-                            data = CX.aug_pipeline(store.images, sel_s, prog_s, post_s,
-                                                   mean_t, std_t, bf16)
-                        label = store.labels.index_select(0, sel_s)
-                    return crit(model(data), label)
                 with torch.cuda.graph(graph):
-                    loss_ag = fwd_loss_only()
+                    loss_ag = gpu_fwd_loss()
                     ag_grads = torch.autograd.grad(loss_ag, flat.params,
                                                    allow_unused=True)
             else:

@@ -7,11 +7,12 @@ CDNA4 kernels:
     csrc/depthwise.hip streaming kernels (EfficientNet's MBConv);
   * dense bf16 NHWC, square kernel 1/3, groups=1, dilation=1 ->
     csrc/conv_mfma.hip implicit-GEMM MFMA kernels.
-The dense dispatch thresholds below are MEASURED per shape
-(profiles/conv_bench_r01.txt); everything else — including stride-2
-backward-data — keeps the torch/MIOpen path where it currently wins.
-Env knobs: FAA_NO_PATCH, FAA_WRW, FAA_BWD_DATA, FAA_CONV_SPLITK,
-FAA_CONV_TILE, FAA_DW_TPL5.
+All dispatch thresholds are MEASURED per shape (profiles/ +
+gpurun_out/call*.log sweeps, pinned by tests/test_dispatch.py); shapes
+where torch/MIOpen still wins keep the fallback. Env knobs:
+FAA_NO_PATCH, FAA_WRW, FAA_WRW_V3, FAA_BWD_DATA, FAA_CONV_DIRECT,
+FAA_CONV_D8, FAA_CONV_SPLITK, FAA_CONV_TILE, FAA_DBIAS, FAA_FLIP_BATCH,
+FAA_DW_V3, FAA_DW_TPL5.
 """
 from __future__ import annotations
 
@@ -22,13 +23,9 @@ import torch.nn.functional as F
 from . import ext
 
 
-# Measured dispatch rules (profiles/conv_bench_r01.txt + wrw v2 re-measure,
-# MI355X b128): the MFMA fwd kernel beats MIOpen up to 160 channels
-# (fwd 160x32x32->160: 249us vs 377us); bwd-data wins to 128 (loses at
-# 160: 226 vs 171); the wrw v2 kernel wins on the stem (tiny Cin) and the
-# deep stages (Cin>=128), loses 1.2-2x in between.
-_FAA_MAX_CH = 160          # fwd (gates module patching)
-_FAA_BWD_DATA_MAX = 128
+# Small/mid-channel blanket threshold (round-1 measurement, still the
+# round-2 cutover between "always in-house" and "per-shape rules"):
+_FAA_MAX_CH = 160
 
 
 import os as _os
