@@ -203,3 +203,21 @@ def test_fused_rmsprop_tf_cpu_matches_reference():
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         # bf16 grads vs fp32 grads: loose tolerance
         assert (p1.float() - p2).abs().max().item() < 5e-2
+
+
+def test_sample_pairing_compiles_to_pairing_slots():
+    """SamplePairing policies compile to OP_PAIRING slots with an in-range
+    batch-slot partner in BOTH compilers (scalar + vectorized)."""
+    import numpy as np
+    from fast_autoaugment_amd.aug import ops as aug_ops
+    policy = [[("SamplePairing", 1.0, 0.5)]]
+    for compiler in (aug_ops.compile_program, aug_ops.compile_program_fast):
+        rng = np.random.default_rng(5)
+        prog = compiler(policy, 16, 32, 32, rng)
+        codes = prog[:, 0, 0]
+        assert (codes == float(aug_ops.OpCode.PAIRING)).all(), compiler.__name__
+        alphas = prog[:, 0, 1]
+        # level 0.5 in range [0, 0.4] -> alpha 0.2
+        assert np.allclose(alphas, 0.2), compiler.__name__
+        partners = prog[:, 0, 2]
+        assert ((partners >= 0) & (partners < 16)).all(), compiler.__name__
