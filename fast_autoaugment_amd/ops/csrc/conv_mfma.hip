@@ -1088,8 +1088,8 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     // tile width covering Wd (exact for the CIFAR sizes; masked tails for
     // the ResNet-50 ImageNet spatials 56/28/14/7 — epilogue guards skip
     // out-of-range rows/cols, staging zero-fills them)
-    int twsel = (g.Wd <= 8) ? 8 : (g.Wd <= 16) ? 16 : (g.Wd <= 32) ? 32
-                : (g.Wd <= 56) ? 56 : 0;
+    int twsel = (g.Wd <= 8) ? 8 : (g.Wd <= 16) ? 16 : (g.Wd <= 28) ? 28
+                : (g.Wd <= 32) ? 32 : (g.Wd <= 56) ? 56 : 0;
     bool geom_ok = (g.KH == 3 && g.KW == 3 && g.stride == 1 && g.pad == 1
                     && g.Cin >= 16 && twsel > 0
                     && (twsel != 8 || g.B % 2 == 0));
@@ -1103,7 +1103,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
       int64_t blocks64 = (twsel == 32)
           ? (int64_t)g.B * tiles_h4 * gn64
           : (int64_t)(g.B / ib) * tiles_h * gn64;
-      bool big = (g.Cin > 64) && (twsel != 56)
+      bool big = (g.Cin > 64) && (twsel != 56) && (twsel != 28)
                  && (blocks64 >= (twsel == 8 ? 192 : 512));
       if (de && de[0] == 'b') big = true;
       if (de && de[0] == 's') big = false;
@@ -1175,6 +1175,12 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
         int gn = (g.Cout + 31) / 32;
         int64_t grid32 = (int64_t)(g.B / ib) * tiles_h * gn;
         if (twsel == 56) CT_BOTH(8, 56, 1, 1, 7, 2, grid32, tiles_h, gn);
+        else if (twsel == 28) {
+          // TH=16 x TW=28 (no column waste for the ResNet 28^2 stage)
+          int th16 = (g.H + 15) / 16;
+          int64_t grid28 = (int64_t)g.B * th16 * gn;
+          CT_BOTH(16, 28, 1, 1, 7, 2, grid28, th16, gn);
+        }
         else if (twsel == 32) CT_BOTH(8, 32, 1, 1, 4, 2, grid32, tiles_h, gn);
         else if (twsel == 16) CT_BOTH(8, 16, 1, 1, 2, 2, grid32, tiles_h, gn);
         else CT_BOTH(8, 8, 2, 1, 2, 2, grid32, tiles_h, gn);
