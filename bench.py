@@ -171,7 +171,6 @@ def main():
                    weight_decay=conf["optimizer"]["decay"], grad_clip=5.0)
     sched = build_scheduler(conf, opt, lr0)
     crit = CrossEntropyLabelSmooth(nc, 0.0)
-    amp_dtype = None   # bf16 runs natively through the flat bf16 weights
 
     model.train()
     steps_per_epoch = max(len(loader), 1) if loader is not None else 390
